@@ -1,0 +1,194 @@
+"""Algorithm-core tests: train-step mechanics, HER, checkpoint/resume,
+SharedAdam vs torch Adam, soft updates."""
+
+import copy
+
+import numpy as np
+import pytest
+import torch
+
+from d4pg_amd.algo.d4pg import DDPG
+from d4pg_amd.algo.shared_adam import SharedAdam
+from d4pg_amd.envs import make, obs_act_dims
+from d4pg_amd.her import add_experience, flat_obs, rollout_episode
+
+DIST = {"type": "categorical", "v_min": -300.0, "v_max": 0.0, "n_atoms": 51}
+
+
+def _agent(**kw):
+    env = make("Pendulum-v1", seed=0)
+    kw.setdefault("critic_dist_info", DIST)
+    kw.setdefault("memory_size", 2000)
+    kw.setdefault("seed", 0)
+    a = DDPG(3, 1, env=env, **kw)
+    return a, env
+
+
+def _fill(agent, env, episodes=3, n_steps=1):
+    rng = np.random.default_rng(0)
+    for _ in range(episodes):
+        ep, _, _ = rollout_episode(agent, env)
+        add_experience(agent.replayBuffer, env, ep, n_steps=n_steps,
+                       gamma=agent.gamma, rng=rng)
+
+
+def test_train_step_changes_params_and_targets():
+    agent, env = _agent(n_steps=5)
+    _fill(agent, env, n_steps=5)
+    w0 = agent.actor.fc1.weight.detach().clone()
+    t0 = agent.actor_target.fc1.weight.detach().clone()
+    c0 = agent.critic.fc3.weight.detach().clone()
+    cl, pl = agent.train()
+    assert np.isfinite(cl) and np.isfinite(pl)
+    assert not torch.allclose(w0, agent.actor.fc1.weight)
+    assert not torch.allclose(c0, agent.critic.fc3.weight)
+    # target lerp moved by tau-fraction
+    dt = (agent.actor_target.fc1.weight - t0).abs().max().item()
+    assert 0 < dt < 1e-2
+
+
+def test_train_updates_priorities():
+    agent, env = _agent()
+    _fill(agent, env)
+    before = agent.replayBuffer._it_sum.sum()
+    agent.train()
+    after = agent.replayBuffer._it_sum.sum()
+    assert before != pytest.approx(after)
+
+
+def test_uniform_replay_mode():
+    agent, env = _agent(prioritized_replay=False)
+    _fill(agent, env)
+    cl, pl = agent.train()
+    assert np.isfinite(cl)
+
+
+def test_soft_update_lerp_formula():
+    agent, _ = _agent(tau=0.5)
+    with torch.no_grad():
+        agent.actor.fc1.weight.fill_(2.0)
+        agent.actor_target.fc1.weight.fill_(0.0)
+    agent.update_target_parameters()
+    assert torch.allclose(agent.actor_target.fc1.weight,
+                          torch.ones_like(agent.actor_target.fc1.weight))
+
+
+def test_hogwild_grad_aliasing():
+    """copy_gradients aliases local .grad into the global params one-shot
+    (reference ddpg.py:104-108 semantics)."""
+    agent, env = _agent()
+    glob, _ = _agent()
+    _fill(agent, env)
+    opt_a = SharedAdam(glob.actor.parameters(), lr=1e-3)
+    opt_c = SharedAdam(glob.critic.parameters(), lr=1e-3)
+    agent.assign_global_optimizer(opt_a, opt_c)
+    g0 = glob.critic.fc1.weight.detach().clone()
+    agent.train(global_model=glob)
+    assert not torch.allclose(g0, glob.critic.fc1.weight)
+    # local was synced back from global
+    assert torch.allclose(agent.critic.fc1.weight, glob.critic.fc1.weight)
+    # aliasing: global grad IS the local grad tensor
+    assert glob.critic.fc1.weight.grad.data_ptr() \
+        == agent.critic.fc1.weight.grad.data_ptr()
+
+
+def test_shared_adam_matches_torch_adam():
+    """Bitwise-class parity vs torch.optim.Adam at identical betas
+    ('Adam bitwise vs torch.optim.Adam single-process')."""
+    torch.manual_seed(0)
+    net1 = torch.nn.Linear(4, 4)
+    net2 = copy.deepcopy(net1)
+    o1 = SharedAdam(net1.parameters(), lr=1e-3, betas=(0.9, 0.999))
+    o2 = torch.optim.Adam(net2.parameters(), lr=1e-3, betas=(0.9, 0.999))
+    for _ in range(5):
+        x = torch.randn(8, 4)
+        for net, o in ((net1, o1), (net2, o2)):
+            o.zero_grad()
+            net(x).pow(2).sum().backward()
+            o.step()
+    assert torch.allclose(net1.weight, net2.weight, atol=1e-7)
+
+
+def test_shared_adam_default_betas_quirk():
+    o = SharedAdam(torch.nn.Linear(2, 2).parameters())
+    assert o.param_groups[0]["betas"] == (0.9, 0.9)
+
+
+def test_her_relabeling():
+    env = make("GoalReach-v0", seed=0)
+    obs_dim, act_dim = obs_act_dims(env, her=True)
+    agent = DDPG(obs_dim, act_dim, env=env, memory_size=1000,
+                 critic_dist_info=DIST, seed=0)
+    rng = np.random.default_rng(0)
+    ep, _, _ = rollout_episode(agent, env)
+    n = add_experience(agent.replayBuffer, env, ep, her=True, her_ratio=1.0,
+                       rng=rng)
+    T = len(ep)
+    assert n == 2 * T                        # every step stored + relabeled
+    # relabeled tuples must carry the timestep's own action (bug fix vs
+    # reference main.py:184): verify actions in buffer match episode actions
+    st = agent.replayBuffer._store
+    stored_actions = st.actions[:n]
+    ep_actions = np.array([a for (_, a, *_rest) in ep], dtype=np.float32)
+    for t in range(T):
+        assert any(np.allclose(stored_actions[i], ep_actions[t])
+                   for i in range(n))
+
+
+def test_her_final_goal_success_relabel():
+    """Relabeling to the episode's own achieved goals must create some
+    reward-0 (success) transitions."""
+    env = make("GoalReach-v0", seed=1)
+    obs_dim, act_dim = obs_act_dims(env, her=True)
+    agent = DDPG(obs_dim, act_dim, env=env, memory_size=1000,
+                 critic_dist_info=DIST, seed=1)
+    rng = np.random.default_rng(1)
+    total, succ = 0, 0
+    for k in range(5):
+        ep, _, _ = rollout_episode(agent, env)
+        add_experience(agent.replayBuffer, env, ep, her=True, her_ratio=1.0,
+                       rng=rng)
+    st = agent.replayBuffer._store
+    assert (st.rewards[:len(agent.replayBuffer)] == 0).any()
+
+
+def test_full_resume_roundtrip(tmp_path):
+    agent, env = _agent()
+    _fill(agent, env)
+    for _ in range(3):
+        agent.train()
+    st = agent.state_dict()
+    torch.save(st, tmp_path / "ckpt.pth")
+    agent2, _ = _agent()
+    agent2.load_state_dict(torch.load(tmp_path / "ckpt.pth",
+                                      weights_only=False))
+    assert agent2.train_steps_done == agent.train_steps_done
+    assert agent2.beta_schedule.t == agent.beta_schedule.t
+    x = torch.randn(4, 3)
+    assert torch.allclose(agent.actor(x), agent2.actor(x))
+    # both continue identically on the same sampled batch
+    assert len(agent2.replayBuffer) == len(agent.replayBuffer)
+
+
+def test_reference_checkpoint_format(tmp_path):
+    agent, _ = _agent()
+    agent.save(str(tmp_path))
+    sd = torch.load(tmp_path / "actor.pth", weights_only=True)
+    assert sorted(sd.keys()) == sorted([
+        "fc1.weight", "fc1.bias", "fc2.weight", "fc2.bias",
+        "fc2_2.weight", "fc2_2.bias", "fc3.weight", "fc3.bias"])
+
+
+def test_select_action_clipped():
+    agent, _ = _agent()
+    for _ in range(10):
+        a = agent.select_action(np.array([1.0, 0.0, 0.0]), explore=True)
+        assert a.shape == (1,)
+        assert -1.0 <= a[0] <= 1.0
+
+
+def test_flat_obs_dict_and_array():
+    assert flat_obs(np.array([1.0, 2.0])).tolist() == [1.0, 2.0]
+    d = {"observation": np.array([1.0]), "desired_goal": np.array([2.0]),
+         "achieved_goal": np.array([3.0])}
+    assert flat_obs(d).tolist() == [1.0, 2.0]
