@@ -87,6 +87,8 @@ class NormalizeAction:
         self.wrapped = env
 
     def __getattr__(self, name):
+        if name == "wrapped":      # guard: unpickling calls __getattr__
+            raise AttributeError(name)   # before __dict__ is populated
         return getattr(self.wrapped, name)
 
     def _action(self, action):
