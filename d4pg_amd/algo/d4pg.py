@@ -167,6 +167,8 @@ class DDPG:
     def select_action(self, obs: np.ndarray, explore: bool = True) -> np.ndarray:
         """B=1 policy inference for env stepping (+clipped Gaussian noise,
         reference main.py:145-146 semantics)."""
+        if self._fused is not None:
+            self._fused.sync_params_if_dirty()
         x = torch.as_tensor(np.asarray(obs, dtype=np.float32),
                             device=self.device).reshape(1, -1)
         a = self.actor(x).cpu().numpy().reshape(-1)
@@ -212,9 +214,9 @@ class DDPG:
         optimizers step (reference ddpg.py:200-255); without it, the local
         optimizers step (the MI355X learner topology, where distribution is
         handled by RCCL all-reduce in parallel/learner.py instead)."""
-        batch = self.sample(self.batch_size)
         if self.backend == "hip":
-            return self._train_step_hip(batch)
+            return self._train_step_hip()
+        batch = self.sample(self.batch_size)
         return self._train_step_eager(batch, global_model)
 
     def _train_step_eager(self, batch, global_model=None):
@@ -285,11 +287,11 @@ class DDPG:
         self.train_steps_done += 1
         return float(qdist_loss.detach()), float(policy_loss.detach())
 
-    def _train_step_hip(self, batch):
+    def _train_step_hip(self):
         if self._fused is None:
             from ..ops import build_fused_engine
             self._fused = build_fused_engine(self)
-        return self._fused.step(batch)
+        return self._fused.step()
 
     # ------------------------------------------------------------------
     # checkpointing
@@ -298,6 +300,8 @@ class DDPG:
         """Reference-format checkpoint: actor.pth / critic.pth state_dicts
         with fc1/fc2/fc2_2/fc3 keys (main.py:367-368)."""
         import os
+        if self._fused is not None:
+            self._fused.sync_params_if_dirty()
         torch.save(self.actor.state_dict(),
                    os.path.join(run_dir, "actor.pth"))
         torch.save(self.critic.state_dict(),

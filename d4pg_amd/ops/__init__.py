@@ -1,0 +1,285 @@
+"""HIP extension loader + the fused-engine wrapper.
+
+The compiled extension (`_d4pg_hip.so`, built in-tree by setup.py /
+__graft_entry__.build()) owns all GPU state: parameter slabs (weights stored
+transposed [in][out]), Adam moments, target slabs, the on-HBM PER sum/min
+trees and SoA replay store, device RNG and schedule counters.  Python only
+packs/unpacks torch state_dicts into flat slabs and drives
+step/capture/replay.
+
+Policy: on a GPU machine the HIP path is mandatory — a missing or broken
+extension raises ImportError loudly (no silent eager fallback; eager exists
+for CPU tests only).
+"""
+
+from __future__ import annotations
+
+import glob
+import importlib.util
+import os
+
+import numpy as np
+import torch
+
+_EXT = None
+_EXT_ERR: Exception | None = None
+
+
+def _find_so():
+    here = os.path.dirname(os.path.abspath(__file__))
+    cands = sorted(glob.glob(os.path.join(here, "_d4pg_hip*.so")))
+    return cands[0] if cands else None
+
+
+def load_extension():
+    """Import the in-tree compiled extension (no JIT — the .so must have
+    been built by setup.py / __graft_entry__.build())."""
+    global _EXT, _EXT_ERR
+    if _EXT is not None:
+        return _EXT
+    so = _find_so()
+    if so is None:
+        _EXT_ERR = ImportError(
+            "_d4pg_hip.so not found — build it with "
+            "`python setup.py build_ext --inplace` (or __graft_entry__.build())")
+        raise _EXT_ERR
+    spec = importlib.util.spec_from_file_location("_d4pg_hip", so)
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    _EXT = mod
+    return mod
+
+
+def extension_available() -> bool:
+    try:
+        load_extension()
+        return True
+    except Exception:
+        return False
+
+
+# ---------------------------------------------------------------------------
+# slab packing: torch module (fc1/fc2/fc2_2/fc3) <-> flat engine slab
+# ---------------------------------------------------------------------------
+
+_LAYERS = ["fc1", "fc2", "fc2_2", "fc3"]
+
+
+def pack_net(module: torch.nn.Module) -> torch.Tensor:
+    """Flatten a 4-layer MLP into the engine slab layout:
+    per layer, transposed weight [in][out] then bias [out]."""
+    parts = []
+    for name in _LAYERS:
+        lin = getattr(module, name)
+        parts.append(lin.weight.detach().t().contiguous().reshape(-1))
+        parts.append(lin.bias.detach().reshape(-1))
+    return torch.cat(parts).to(torch.float32).cpu()
+
+
+def unpack_net(module: torch.nn.Module, flat: torch.Tensor) -> None:
+    off = 0
+    with torch.no_grad():
+        for name in _LAYERS:
+            lin = getattr(module, name)
+            o, i = lin.weight.shape
+            w = flat[off:off + i * o].reshape(i, o).t()
+            off += i * o
+            b = flat[off:off + o]
+            off += o
+            lin.weight.copy_(w)
+            lin.bias.copy_(b)
+    assert off == flat.numel(), "slab size mismatch on unpack"
+
+
+# ---------------------------------------------------------------------------
+# GPU replay adapter (host-side buffering -> batched device ingestion)
+# ---------------------------------------------------------------------------
+
+class GPUReplayAdapter:
+    """Drop-in for the learner's replayBuffer when the replay lives on-HBM.
+    add() buffers host-side; flush() ships one batched ingest (H2D + a
+    single level-synced tree-insert kernel)."""
+
+    def __init__(self, engine: "FusedEngine"):
+        self.engine = engine
+        self._pending = []
+
+    def add(self, state, action, reward, next_state, done):
+        self._pending.append((
+            np.asarray(state, np.float32).ravel(),
+            np.asarray(action, np.float32).ravel(),
+            np.float32(reward),
+            np.asarray(next_state, np.float32).ravel(),
+            np.float32(done)))
+        if len(self._pending) >= 32768:
+            self.flush()
+
+    def flush(self):
+        if not self._pending:
+            return
+        s = torch.from_numpy(np.stack([p[0] for p in self._pending]))
+        a = torch.from_numpy(np.stack([p[1] for p in self._pending]))
+        r = torch.from_numpy(np.asarray([p[2] for p in self._pending]))
+        s2 = torch.from_numpy(np.stack([p[3] for p in self._pending]))
+        d = torch.from_numpy(np.asarray([p[4] for p in self._pending]))
+        self._pending.clear()
+        self.engine.ingest(s, a, r, s2, d)
+
+    def __len__(self):
+        return int(self.engine.counters()["size"]) + len(self._pending)
+
+
+# ---------------------------------------------------------------------------
+# the fused engine wrapper
+# ---------------------------------------------------------------------------
+
+class FusedEngine:
+    """Owns one device-side D4PG learner (see ops/hip/engine.hip)."""
+
+    def __init__(self, obs_dim, act_dim, hidden, n_atoms, batch, capacity,
+                 v_min, v_max, gamma_n, tau, lr_actor, lr_critic,
+                 per_alpha=0.6, per_beta0=0.4, per_beta_iters=100000,
+                 per_eps=1e-6, seed=0, is_weighting=False):
+        self.ext = load_extension()
+        self.h = self.ext.create(
+            obs_dim, act_dim, hidden, n_atoms, batch, capacity,
+            float(v_min), float(v_max), float(gamma_n), float(tau),
+            float(lr_actor), float(lr_critic), float(per_alpha),
+            float(per_beta0), int(per_beta_iters), float(per_eps),
+            int(seed), bool(is_weighting))
+        self.batch = batch
+        self._captured = 0
+
+    def __del__(self):
+        try:
+            self.ext.destroy(self.h)
+        except Exception:
+            pass
+
+    # -- params --
+    SLABS = {"actor": 0, "actor_target": 1, "critic": 2, "critic_target": 3,
+             "g_actor": 4, "g_critic": 5, "m_actor": 6, "v_actor": 7,
+             "m_critic": 8, "v_critic": 9}
+
+    def load_from_modules(self, actor, actor_target, critic, critic_target):
+        self.ext.load_slab(self.h, 0, pack_net(actor))
+        self.ext.load_slab(self.h, 1, pack_net(actor_target))
+        self.ext.load_slab(self.h, 2, pack_net(critic))
+        self.ext.load_slab(self.h, 3, pack_net(critic_target))
+
+    def store_to_modules(self, actor, actor_target, critic, critic_target):
+        unpack_net(actor, self.ext.store_slab(self.h, 0))
+        unpack_net(actor_target, self.ext.store_slab(self.h, 1))
+        unpack_net(critic, self.ext.store_slab(self.h, 2))
+        unpack_net(critic_target, self.ext.store_slab(self.h, 3))
+
+    def load_slab(self, name, flat):
+        self.ext.load_slab(self.h, self.SLABS[name], flat)
+
+    def store_slab(self, name):
+        return self.ext.store_slab(self.h, self.SLABS[name])
+
+    # -- replay --
+    def synth_fill(self, n, seed=1234):
+        self.ext.synth_fill(self.h, int(n), int(seed))
+
+    def ingest(self, s, a, r, s2, d):
+        self.ext.ingest(self.h, s, a, r, s2, d)
+
+    # -- stepping --
+    def step(self, n=1):
+        """Uncaptured (eager-launch) steps — used by parity tests."""
+        self.ext.step(self.h, int(n))
+
+    def train_steps(self, n=1, steps_per_graph=8):
+        """Graph-replayed steps: captures once (steps_per_graph per replay),
+        then replays; the remainder runs uncaptured."""
+        if self._captured != steps_per_graph:
+            self.ext.capture(self.h, steps_per_graph)
+            self._captured = steps_per_graph
+        full, rem = divmod(int(n), steps_per_graph)
+        if full:
+            self.ext.replay(self.h, full)
+        if rem:
+            self.ext.step(self.h, rem)
+
+    def train_steps_async(self, n, steps_per_graph=8):
+        if self._captured != steps_per_graph:
+            self.ext.capture(self.h, steps_per_graph)
+            self._captured = steps_per_graph
+        assert n % steps_per_graph == 0
+        self.ext.replay_async(self.h, n // steps_per_graph)
+
+    def sync(self):
+        self.ext.sync(self.h)
+
+    # -- introspection --
+    def counters(self):
+        return self.ext.counters(self.h)
+
+    def read(self, name):
+        return self.ext.read_buffer(self.h, name)
+
+    def actor_forward(self, x):
+        return self.ext.actor_forward(self.h, x)
+
+    def info(self):
+        return self.ext.info(self.h)
+
+
+class FusedDDPGBridge:
+    """Glue between the Python DDPG object and the device engine for the
+    'hip' backend: replaces the replay with a GPUReplayAdapter, runs train
+    steps on-device, and lazily syncs parameters back to the torch modules
+    for acting/eval/checkpointing."""
+
+    def __init__(self, ddpg):
+        self.ddpg = ddpg
+        self.engine = FusedEngine(
+            obs_dim=ddpg.obs_dim, act_dim=ddpg.act_dim,
+            hidden=ddpg.actor.hidden, n_atoms=ddpg.n_atoms,
+            batch=ddpg.batch_size, capacity=ddpg.memory_size,
+            v_min=ddpg.v_min, v_max=ddpg.v_max, gamma_n=ddpg.n_step_gamma,
+            tau=ddpg.tau,
+            lr_actor=ddpg.optimizer_actor.param_groups[0]["lr"],
+            lr_critic=ddpg.optimizer_critic.param_groups[0]["lr"],
+            is_weighting=ddpg.is_weighting,
+            seed=0 if ddpg.rng is None else int(ddpg.rng.integers(1 << 31)))
+        self.engine.load_from_modules(ddpg.actor, ddpg.actor_target,
+                                      ddpg.critic, ddpg.critic_target)
+        # migrate any CPU replay contents, then swap in the GPU adapter
+        adapter = GPUReplayAdapter(self.engine)
+        old = getattr(ddpg, "replayBuffer", None)
+        if old is not None and len(old) > 0:
+            st = getattr(old, "_store", None) or getattr(old, "store", None)
+            if st is not None and st.size > 0:
+                n = st.size
+                self.engine.ingest(
+                    torch.from_numpy(st.states[:n]),
+                    torch.from_numpy(st.actions[:n]),
+                    torch.from_numpy(st.rewards[:n]),
+                    torch.from_numpy(st.next_states[:n]),
+                    torch.from_numpy(st.dones[:n]))
+        ddpg.replayBuffer = adapter
+        self._params_dirty = False
+
+    def step(self, batch=None, n=1):
+        self.ddpg.replayBuffer.flush()
+        self.engine.train_steps(n)
+        self._params_dirty = True
+        self.ddpg.train_steps_done += n
+        return float("nan"), float("nan")
+
+    def sync_params_if_dirty(self):
+        if self._params_dirty:
+            self.engine.store_to_modules(
+                self.ddpg.actor, self.ddpg.actor_target,
+                self.ddpg.critic, self.ddpg.critic_target)
+            self._params_dirty = False
+
+
+def build_fused_engine(ddpg):
+    if not torch.cuda.is_available():
+        raise RuntimeError(
+            "backend='hip' requires a GPU (use backend='eager' on CPU)")
+    return FusedDDPGBridge(ddpg)

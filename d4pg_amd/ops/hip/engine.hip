@@ -1,0 +1,1130 @@
+// d4pg_amd fused D4PG learner engine — CDNA4 (gfx950 / MI355X) native.
+//
+// The entire D4PG train step of the reference (/root/reference/ddpg.py:200-255
+// — PER sample, target forwards, C51 projection, critic CE backward, Adam,
+// policy backward, Adam, target soft-update, priority write-back) runs here
+// as a fixed sequence of ~34 HIP kernels on one stream, captured into a
+// hipGraph so N steps replay with no host dispatch.  All state is
+// device-resident: parameters in a flat slab (weights stored TRANSPOSED,
+// [in][out], for coalesced forward reads), Adam moments, target slabs, the
+// SoA replay store and the prioritized sum/min segment trees in HBM
+// (SURVEY.md §2c K12), philox4x32 RNG, and the beta/Adam step counters.
+//
+// Regime note (why this shape): at B=64, H=256 every GEMM is tiny, so the
+// step is kernel-boundary-bound (~1.45us per dependent boundary on MI355X);
+// the design minimizes launches (fused dW+dX+db backward kernels,
+// multi-job forward kernels batching independent layers) rather than
+// chasing per-GEMM MFMA peaks.  The wide-batch benchmark config
+// (B=4096, H=1024) reuses the same kernels with larger grids.
+//
+// Numerics: fp32 end-to-end, matching the reference's CPU fp32.
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+#include <cstdio>
+#include <cstring>
+#include <cmath>
+#include <vector>
+#include <stdexcept>
+
+#define HIP_CHECK(cmd) do { \
+    hipError_t e_ = (cmd); \
+    if (e_ != hipSuccess) { \
+        char buf[256]; \
+        snprintf(buf, sizeof(buf), "HIP error %s at %s:%d", \
+                 hipGetErrorString(e_), __FILE__, __LINE__); \
+        throw std::runtime_error(buf); \
+    } \
+} while (0)
+
+namespace d4pg {
+
+// ---------------------------------------------------------------------------
+// philox4x32-10 counter-based RNG (device)
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ uint32_t mulhi32(uint32_t a, uint32_t b) {
+    return (uint32_t)(((uint64_t)a * b) >> 32);
+}
+
+struct Philox4 { uint32_t v[4]; };
+
+__device__ inline Philox4 philox4(uint64_t seed, uint64_t ctr_hi,
+                                  uint64_t ctr_lo) {
+    uint32_t c0 = (uint32_t)ctr_lo, c1 = (uint32_t)(ctr_lo >> 32);
+    uint32_t c2 = (uint32_t)ctr_hi, c3 = (uint32_t)(ctr_hi >> 32);
+    uint32_t k0 = (uint32_t)seed, k1 = (uint32_t)(seed >> 32);
+    const uint32_t M0 = 0xD2511F53u, M1 = 0xCD9E8D57u;
+    const uint32_t W0 = 0x9E3779B9u, W1 = 0xBB67AE85u;
+#pragma unroll
+    for (int r = 0; r < 10; ++r) {
+        uint32_t hi0 = mulhi32(M0, c0), lo0 = M0 * c0;
+        uint32_t hi1 = mulhi32(M1, c2), lo1 = M1 * c2;
+        uint32_t n0 = hi1 ^ c1 ^ k0, n1 = lo1;
+        uint32_t n2 = hi0 ^ c3 ^ k1, n3 = lo0;
+        c0 = n0; c1 = n1; c2 = n2; c3 = n3;
+        k0 += W0; k1 += W1;
+    }
+    return {c0, c1, c2, c3};
+}
+
+__device__ inline float u01(uint32_t x) {           // (0,1]
+    return ((float)x + 1.0f) * 2.3283064e-10f;
+}
+
+// ---------------------------------------------------------------------------
+// Engine state blocks
+// ---------------------------------------------------------------------------
+
+// One linear layer's geometry inside a parameter slab.
+struct LayerDesc {
+    int in1, in2, out;       // in2 > 0 => concat second input
+    long w_off, b_off;       // offsets into the net's slab (floats)
+};
+
+enum Act { ACT_NONE = 0, ACT_RELU = 1, ACT_TANH = 2, ACT_SOFTMAX = 3 };
+
+// A forward job: y = act(x1 [. x2] @ Wt + b).  Grid-sliced when several
+// independent jobs share a launch.
+struct FwdJob {
+    const float* x1; const float* x2;
+    const float* wt; const float* bias;
+    float* y;
+    int B, in1, in2, out, act;
+    int wg0, nwg_b, nwg_o;   // wg slice: blocks [wg0, wg0+nwg_b*nwg_o)
+};
+
+// Backward job for one layer: given dz (grad wrt pre-activation, [B,out]):
+//   dWt[i][o] += sum_b x[b][i] dz[b][o];  db[o] += sum_b dz[b][o]
+//   dx[b][i]   = sum_o dz[b][o] Wt[i][o], then * act'(h_prev)
+struct BwdJob {
+    const float* dz;
+    const float* x1; const float* x2;       // layer inputs (concat aware)
+    const float* wt;
+    float* dwt; float* dbias;               // null => skip dW part
+    float* dx1; float* dx2;                 // null => skip dX part(s)
+    const float* h1;                        // prev activation for mask (dx1)
+    int B, in1, in2, out;
+    int prev_act;                           // act of PREVIOUS layer (mask)
+    int wg0_dw, nwg_dw_i, nwg_dw_o;         // dW tile grid slice
+    int wg0_dx, nwg_dx_b, nwg_dx_i;         // dX tile grid slice
+};
+
+struct EngineCfg {
+    int obs, act, hidden, atoms, batch;
+    long capacity;             // replay capacity (will be rounded to pow2 tree)
+    float v_min, v_max, gamma_n, tau, lr_actor, lr_critic;
+    float per_alpha, per_beta0, per_eps;
+    long per_beta_iters;
+    uint64_t seed;
+    int is_weighting;          // apply IS weights to the CE loss
+};
+
+// device-side counters (one small block)
+struct Counters {
+    long long beta_t;          // PER beta schedule position (stateful)
+    long long adam_t_actor;
+    long long adam_t_critic;
+    long long rng_epoch;       // bumped per step for fresh philox streams
+    long long size;            // replay occupancy
+    long long pos;             // replay ring position
+    float max_priority;
+    float loss_critic;         // per-step scalars (overwritten each step)
+    float loss_actor;
+};
+
+// ===========================================================================
+// Kernels
+// ===========================================================================
+
+// ---- tick: advance counters at the start of every step --------------------
+__global__ void k_tick(Counters* c) {
+    if (threadIdx.x == 0) {
+        c->beta_t += 1;
+        c->adam_t_actor += 1;
+        c->adam_t_critic += 1;
+        c->rng_epoch += 1;
+        c->loss_critic = 0.f;
+        c->loss_actor = 0.f;
+    }
+}
+
+// ---- PER sample + gather ---------------------------------------------------
+// One wave per probe: lane 0 walks the sum tree (double, exact), all lanes
+// then cooperatively gather the transition row into the batch SoA.
+// Also computes IS weights w = ((p/total)*N)^-beta / max_w.
+__global__ void k_per_sample(
+        const double* __restrict__ sum_tree, const double* __restrict__ min_tree,
+        long tree_cap,
+        const float* __restrict__ rs, const float* __restrict__ ra,
+        const float* __restrict__ rr, const float* __restrict__ rs2,
+        const float* __restrict__ rd,
+        int obs, int act,
+        float* __restrict__ bs, float* __restrict__ ba, float* __restrict__ br,
+        float* __restrict__ bs2, float* __restrict__ bd,
+        float* __restrict__ bw, long* __restrict__ bidx,
+        int B, const Counters* __restrict__ cnt,
+        float beta0, float beta_iters, uint64_t seed) {
+    int probe = blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
+    int lane = threadIdx.x & 63;
+    if (probe >= B) return;
+
+    long long n = cnt->size;
+    double total = sum_tree[1];
+    float frac = fminf((float)((double)cnt->beta_t / beta_iters), 1.0f);
+    float beta = beta0 + frac * (1.0f - beta0);
+
+    long idx;
+    if (lane == 0) {
+        Philox4 r = philox4(seed, (uint64_t)cnt->rng_epoch, (uint64_t)probe);
+        double mass = (double)u01(r.v[0]) * total;
+        long node = 1;
+        while (node < tree_cap) {
+            double ls = sum_tree[2 * node];
+            if (mass > ls) { mass -= ls; node = 2 * node + 1; }
+            else           { node = 2 * node; }
+        }
+        idx = node - tree_cap;
+        if (idx >= n) idx = n - 1;          // fp-roundoff guard at range top
+        bidx[probe] = idx;
+        // IS weight
+        double p = sum_tree[tree_cap + idx] / total;
+        double p_min = min_tree[1] / total;
+        double max_w = pow(p_min * (double)n, (double)-beta);
+        bw[probe] = (float)(pow(p * (double)n, (double)-beta) / max_w);
+        br[probe] = rr[idx];
+        bd[probe] = rd[idx];
+    }
+    idx = __shfl(idx, 0, 64);
+    for (int k = lane; k < obs; k += 64) {
+        bs[(long)probe * obs + k] = rs[idx * obs + k];
+        bs2[(long)probe * obs + k] = rs2[idx * obs + k];
+    }
+    for (int k = lane; k < act; k += 64)
+        ba[(long)probe * act + k] = ra[idx * act + k];
+}
+
+// ---- generic fused forward -------------------------------------------------
+// Tile: TB rows x TO cols per workgroup; 256 threads; LDS-staged x and Wt
+// chunks.  ACT_SOFTMAX requires out <= 64 and uses one wave per row.
+#define TB 4
+#define TO 64
+#define TK 32
+
+__device__ inline void fwd_one(const FwdJob& j, int wg) {
+    int rel = wg - j.wg0;
+    int bt = rel / j.nwg_o;            // which row tile
+    int ot = rel % j.nwg_o;            // which col tile
+    int b0 = bt * TB, o0 = ot * TO;
+    int in_total = j.in1 + j.in2;
+
+    __shared__ float xs[TB][TK];
+    __shared__ float ws[TK][TO + 1];
+
+    int tid = threadIdx.x;
+    int tb = tid / TO;                 // 0..TB-1
+    int to = tid % TO;                 // 0..TO-1
+    int b = b0 + tb, o = o0 + to;
+
+    float acc = 0.f;
+    for (int k0 = 0; k0 < in_total; k0 += TK) {
+        // stage x chunk: TB*TK elems by 256 threads
+        for (int t = tid; t < TB * TK; t += 256) {
+            int bb = t / TK, kk = t % TK;
+            int gb = b0 + bb, gk = k0 + kk;
+            float v = 0.f;
+            if (gb < j.B && gk < in_total)
+                v = (gk < j.in1) ? j.x1[(long)gb * j.in1 + gk]
+                                 : j.x2[(long)gb * j.in2 + (gk - j.in1)];
+            xs[bb][kk] = v;
+        }
+        // stage Wt chunk: TK*TO elems
+        for (int t = tid; t < TK * TO; t += 256) {
+            int kk = t / TO, oo = t % TO;
+            int gk = k0 + kk, go = o0 + oo;
+            ws[kk][oo] = (gk < in_total && go < j.out)
+                ? j.wt[(long)gk * j.out + go] : 0.f;
+        }
+        __syncthreads();
+#pragma unroll 8
+        for (int kk = 0; kk < TK; ++kk)
+            acc += xs[tb][kk] * ws[kk][to];
+        __syncthreads();
+    }
+
+    if (b < j.B && o < j.out) {
+        acc += j.bias[o];
+        if (j.act == ACT_RELU) acc = fmaxf(acc, 0.f);
+        else if (j.act == ACT_TANH) acc = tanhf(acc);
+    }
+    if (j.act == ACT_SOFTMAX) {
+        // one wave handles one row (TO==64 lanes over out<=64 columns)
+        float v = (b < j.B && o < j.out) ? acc : -INFINITY;
+        float mx = v;
+        for (int s = 32; s > 0; s >>= 1) mx = fmaxf(mx, __shfl_xor(mx, s, 64));
+        float e = (b < j.B && o < j.out) ? __expf(v - mx) : 0.f;
+        float sum = e;
+        for (int s = 32; s > 0; s >>= 1) sum += __shfl_xor(sum, s, 64);
+        if (b < j.B && o < j.out) j.y[(long)b * j.out + o] = e / sum;
+    } else if (b < j.B && o < j.out) {
+        j.y[(long)b * j.out + o] = acc;
+    }
+}
+
+__global__ void k_fwd3(FwdJob j0, FwdJob j1, FwdJob j2, int njobs) {
+    int wg = blockIdx.x;
+    if (njobs > 0 && wg >= j0.wg0 && wg < j0.wg0 + j0.nwg_b * j0.nwg_o)
+        { fwd_one(j0, wg); return; }
+    if (njobs > 1 && wg >= j1.wg0 && wg < j1.wg0 + j1.nwg_b * j1.nwg_o)
+        { fwd_one(j1, wg); return; }
+    if (njobs > 2 && wg >= j2.wg0 && wg < j2.wg0 + j2.nwg_b * j2.nwg_o)
+        { fwd_one(j2, wg); return; }
+}
+
+// ---- C51 categorical projection (K3) ---------------------------------------
+// One wave per batch row; lane j < K owns atom j.  Mass split accumulated in
+// an LDS row via LDS atomics (no global scatter races).  Equal-bin handling
+// identical to algo/projection.py (index-adjust so weights become (0,1)).
+__global__ void k_project(const float* __restrict__ p_t,
+                          const float* __restrict__ r,
+                          const float* __restrict__ d,
+                          float* __restrict__ m,
+                          int B, int K, float v_min, float v_max,
+                          float gamma_n) {
+    int row = blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
+    int lane = threadIdx.x & 63;
+    int wrow = threadIdx.x / 64;
+    extern __shared__ float lm[];                 // [waves][K]
+    float* mrow = lm + wrow * K;
+    for (int k = lane; k < K; k += 64) mrow[k] = 0.f;
+    __builtin_amdgcn_wave_barrier();
+    if (row < B && lane < K) {
+        float delta = (v_max - v_min) / (K - 1);
+        float z = v_min + lane * delta;
+        float tz = r[row] + gamma_n * (1.f - d[row]) * z;
+        tz = fminf(v_max, fmaxf(v_min, tz));
+        float b = (tz - v_min) / delta;
+        int l = (int)floorf(b);
+        int u = (int)ceilf(b);
+        if (l == u) { if (u > 0) l -= 1; else u += 1; }
+        float p = p_t[(long)row * K + lane];
+        atomicAdd(&mrow[l], p * ((float)u - b));
+        atomicAdd(&mrow[u], p * (b - (float)l));
+    }
+    __builtin_amdgcn_wave_barrier();
+    if (row < B)
+        for (int k = lane; k < K; k += 64)
+            m[(long)row * K + k] = mrow[k];
+}
+
+// ---- critic CE gradient + PER priorities (K4+K5) ---------------------------
+// dlogits = (q - m) * scale / B   (softmax+CE fused gradient)
+// priority = |-(sum m*q)| + eps = sum(m*q) + eps (reference proxy,
+// ddpg.py:220-222).  scale = IS weight when is_weighting.
+__global__ void k_ce_grad(const float* __restrict__ q,
+                          const float* __restrict__ m,
+                          const float* __restrict__ w,
+                          float* __restrict__ dlogits,
+                          float* __restrict__ pri,
+                          Counters* cnt,
+                          int B, int K, float per_eps, int is_weighting) {
+    int row = blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
+    int lane = threadIdx.x & 63;
+    if (row >= B) return;
+    float qv = 0.f, mv = 0.f;
+    if (lane < K) {
+        qv = q[(long)row * K + lane];
+        mv = m[(long)row * K + lane];
+    }
+    float dot = mv * qv;                    // for priority
+    float ce = -mv * __logf(qv + 1e-10f);   // for loss scalar
+    for (int s = 32; s > 0; s >>= 1) {
+        dot += __shfl_xor(dot, s, 64);
+        ce  += __shfl_xor(ce, s, 64);
+    }
+    float scale = (is_weighting && w) ? w[row] : 1.f;
+    if (lane < K)
+        dlogits[(long)row * K + lane] = scale * (qv - mv) / (float)B;
+    if (lane == 0) {
+        pri[row] = dot + per_eps;
+        atomicAdd(&cnt->loss_critic, scale * ce / (float)B);
+    }
+}
+
+// ---- policy gradient through the softmax head (K6 seed) ---------------------
+// L = -(1/B) sum_b sum_k q_k z_k  =>  dlogits_j = -q_j (z_j - E_q[z]) / B
+__global__ void k_policy_grad(const float* __restrict__ q,
+                              float* __restrict__ dlogits,
+                              Counters* cnt,
+                              int B, int K, float v_min, float v_max) {
+    int row = blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
+    int lane = threadIdx.x & 63;
+    if (row >= B) return;
+    float delta = (v_max - v_min) / (K - 1);
+    float z = v_min + lane * delta;
+    float qv = (lane < K) ? q[(long)row * K + lane] : 0.f;
+    float e = qv * z;
+    for (int s = 32; s > 0; s >>= 1) e += __shfl_xor(e, s, 64);
+    if (lane < K)
+        dlogits[(long)row * K + lane] = -qv * (z - e) / (float)B;
+    if (lane == 0) atomicAdd(&cnt->loss_actor, -e / (float)B);
+}
+
+// ---- generic fused backward (dW + db + dX in one launch) --------------------
+#define BWT 16          // dW tile: BWT_i x BWT_o, thread per (i,o)
+#define BXB 16          // dX tile rows
+#define BXI 16          // dX tile cols
+#define BXO 64          // dX o-chunk
+
+__device__ inline float act_mask(int act, float h) {
+    if (act == ACT_RELU) return h > 0.f ? 1.f : 0.f;
+    if (act == ACT_TANH) return 1.f - h * h;
+    return 1.f;
+}
+
+__device__ inline void bwd_one(const BwdJob& j, int wg) {
+    int in_total = j.in1 + j.in2;
+    int tid = threadIdx.x;
+
+    if (j.dwt && wg >= j.wg0_dw && wg < j.wg0_dw + j.nwg_dw_i * j.nwg_dw_o) {
+        // ---- dW part: tile [BWT i x BWT o], loop over batch in LDS chunks
+        int rel = wg - j.wg0_dw;
+        int it = rel / j.nwg_dw_o, ot = rel % j.nwg_dw_o;
+        int i0 = it * BWT, o0 = ot * BWT;
+        __shared__ float xs[32][BWT + 1];            // [bchunk][i]
+        __shared__ float zs[32][BWT + 1];            // [bchunk][o]
+        int ti = tid / BWT, to = tid % BWT;          // thread -> (i, o)
+        float acc = 0.f, accb = 0.f;
+        for (int bc = 0; bc < j.B; bc += 32) {
+            for (int t = tid; t < 32 * BWT; t += 256) {
+                int bb = t / BWT, ii = t % BWT;
+                int gb = bc + bb, gi = i0 + ii;
+                float xv = 0.f;
+                if (gb < j.B && gi < in_total)
+                    xv = (gi < j.in1) ? j.x1[(long)gb * j.in1 + gi]
+                                      : j.x2[(long)gb * j.in2 + (gi - j.in1)];
+                xs[bb][ii] = xv;
+                int oo = ii, go = o0 + oo;
+                zs[bb][oo] = (gb < j.B && go < j.out)
+                    ? j.dz[(long)gb * j.out + go] : 0.f;
+            }
+            __syncthreads();
+#pragma unroll 8
+            for (int bb = 0; bb < 32; ++bb) {
+                acc += xs[bb][ti] * zs[bb][to];
+                if (ti == 0) accb += zs[bb][to];
+            }
+            __syncthreads();
+        }
+        int gi = i0 + ti, go = o0 + to;
+        if (gi < in_total && go < j.out)
+            j.dwt[(long)gi * j.out + go] = acc;
+        if (ti == 0 && go < j.out && i0 == 0 && j.dbias)
+            j.dbias[go] = accb;
+        return;
+    }
+
+    if (wg >= j.wg0_dx && wg < j.wg0_dx + j.nwg_dx_b * j.nwg_dx_i) {
+        // ---- dX part: tile [BXB b x BXI i], o-chunked LDS staging
+        int rel = wg - j.wg0_dx;
+        int bt = rel / j.nwg_dx_i, it = rel % j.nwg_dx_i;
+        int b0 = bt * BXB, i0 = it * BXI;
+        __shared__ float zsx[BXB][BXO + 1];
+        __shared__ float wsx[BXI][BXO + 1];
+        int tb = tid / BXI, ti = tid % BXI;
+        int b = b0 + tb, i = i0 + ti;
+        float acc = 0.f;
+        for (int oc = 0; oc < j.out; oc += BXO) {
+            for (int t = tid; t < BXB * BXO; t += 256) {
+                int bb = t / BXO, oo = t % BXO;
+                int gb = b0 + bb, go = oc + oo;
+                zsx[bb][oo] = (gb < j.B && go < j.out)
+                    ? j.dz[(long)gb * j.out + go] : 0.f;
+                int ii = bb;          // reuse iteration space: BXI == BXB
+                int gi2 = i0 + ii;
+                wsx[ii][oo] = (gi2 < in_total && go < j.out)
+                    ? j.wt[(long)gi2 * j.out + go] : 0.f;
+            }
+            __syncthreads();
+#pragma unroll 8
+            for (int oo = 0; oo < BXO; ++oo)
+                acc += zsx[tb][oo] * wsx[ti][oo];
+            __syncthreads();
+        }
+        if (b < j.B && i < in_total) {
+            if (i < j.in1) {
+                if (j.dx1) {
+                    float h = j.h1 ? j.h1[(long)b * j.in1 + i] : 0.f;
+                    j.dx1[(long)b * j.in1 + i] =
+                        acc * act_mask(j.prev_act, h);
+                }
+            } else if (j.dx2) {
+                j.dx2[(long)b * j.in2 + (i - j.in1)] = acc;  // raw (concat 2nd)
+            }
+        }
+    }
+}
+
+__global__ void k_bwd(BwdJob j) { bwd_one(j, blockIdx.x); }
+
+// tanh backward at the actor output: dz4 = da * (1 - a_out^2)
+__global__ void k_tanh_bwd(const float* __restrict__ da,
+                           const float* __restrict__ a_out,
+                           float* __restrict__ dz, long n) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < n) {
+        float y = a_out[i];
+        dz[i] = da[i] * (1.f - y * y);
+    }
+}
+
+// ---- fused Adam over a flat slab (K8) --------------------------------------
+__global__ void k_adam(float* __restrict__ p, const float* __restrict__ g,
+                       float* __restrict__ m, float* __restrict__ v,
+                       long n, float lr, float b1, float b2, float eps,
+                       const Counters* cnt, int is_actor) {
+    long long t = is_actor ? cnt->adam_t_actor : cnt->adam_t_critic;
+    float bc1 = 1.f - __powf(b1, (float)t);
+    float bc2 = 1.f - __powf(b2, (float)t);
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += (long)gridDim.x * blockDim.x) {
+        float gi = g[i];
+        float mi = b1 * m[i] + (1.f - b1) * gi;
+        float vi = b2 * v[i] + (1.f - b2) * gi * gi;
+        m[i] = mi; v[i] = vi;
+        p[i] -= lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+    }
+}
+
+// ---- target soft update (K9): one kernel, two slabs ------------------------
+__global__ void k_soft_update(float* __restrict__ ta,
+                              const float* __restrict__ sa, long na,
+                              float* __restrict__ tc,
+                              const float* __restrict__ sc, long nc,
+                              float tau) {
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < na + nc;
+         i += (long)gridDim.x * blockDim.x) {
+        if (i < na) ta[i] += tau * (sa[i] - ta[i]);
+        else { long k = i - na; tc[k] += tau * (sc[k] - tc[k]); }
+    }
+}
+
+// ---- PER priority write-back (K12 update path) ------------------------------
+// Exact level-synchronized tree repair: ONE workgroup; thread i owns probe i;
+// per level all touched parents are recomputed from their (already final)
+// children.  Duplicate parents write identical values (benign).
+__global__ void k_per_update(double* __restrict__ sum_tree,
+                             double* __restrict__ min_tree,
+                             long tree_cap,
+                             const long* __restrict__ idx,
+                             const float* __restrict__ pri,
+                             int B, float alpha, Counters* cnt) {
+    int tid = threadIdx.x;
+    // leaves
+    float local_max = 0.f;
+    for (int i = tid; i < B; i += blockDim.x) {
+        float p = pri[i];
+        double pa = pow((double)p, (double)alpha);
+        long leaf = tree_cap + idx[i];
+        sum_tree[leaf] = pa;
+        min_tree[leaf] = pa;
+        local_max = fmaxf(local_max, p);
+    }
+    // wg-reduce max priority
+    __shared__ float smax[256];
+    smax[tid] = local_max;
+    __syncthreads();
+    for (int s = blockDim.x / 2; s > 0; s >>= 1) {
+        if (tid < s) smax[tid] = fmaxf(smax[tid], smax[tid + s]);
+        __syncthreads();
+    }
+    if (tid == 0)
+        cnt->max_priority = fmaxf(cnt->max_priority, smax[0]);
+    __syncthreads();
+    // level-synchronized repair
+    long levels = 0;
+    for (long c = tree_cap; c > 1; c >>= 1) ++levels;
+    for (long lv = 0; lv < levels; ++lv) {
+        for (int i = tid; i < B; i += blockDim.x) {
+            long node = (tree_cap + idx[i]) >> (lv + 1);
+            if (node >= 1) {
+                sum_tree[node] = sum_tree[2 * node] + sum_tree[2 * node + 1];
+                min_tree[node] = fmin(min_tree[2 * node],
+                                      min_tree[2 * node + 1]);
+            }
+        }
+        __syncthreads();
+    }
+}
+
+// ---- replay ingestion (batched add) -----------------------------------------
+// T transitions appended at the ring position with priority max_priority^alpha,
+// then one level-synced repair pass.  Single workgroup (T can exceed threads).
+__global__ void k_replay_add(float* __restrict__ rs, float* __restrict__ ra,
+                             float* __restrict__ rr, float* __restrict__ rs2,
+                             float* __restrict__ rd,
+                             double* __restrict__ sum_tree,
+                             double* __restrict__ min_tree,
+                             long tree_cap, long capacity,
+                             const float* __restrict__ ts,
+                             const float* __restrict__ ta,
+                             const float* __restrict__ tr,
+                             const float* __restrict__ ts2,
+                             const float* __restrict__ td,
+                             int T, int obs, int act, float alpha,
+                             Counters* cnt) {
+    int tid = threadIdx.x;
+    long pos0 = cnt->pos;
+    double pa = pow((double)cnt->max_priority, (double)alpha);
+    // copy rows (grid-stride inside the WG over T*max(obs,act) elems)
+    for (long e = tid; e < (long)T * obs; e += blockDim.x) {
+        long t = e / obs, k = e % obs;
+        long slot = (pos0 + t) % capacity;
+        rs[slot * obs + k] = ts[t * obs + k];
+        rs2[slot * obs + k] = ts2[t * obs + k];
+    }
+    for (long e = tid; e < (long)T * act; e += blockDim.x) {
+        long t = e / act, k = e % act;
+        long slot = (pos0 + t) % capacity;
+        ra[slot * act + k] = ta[t * act + k];
+    }
+    for (int t = tid; t < T; t += blockDim.x) {
+        long slot = (pos0 + t) % capacity;
+        rr[slot] = tr[t];
+        rd[slot] = td[t];
+        sum_tree[tree_cap + slot] = pa;
+        min_tree[tree_cap + slot] = pa;
+    }
+    __syncthreads();
+    long levels = 0;
+    for (long c = tree_cap; c > 1; c >>= 1) ++levels;
+    for (long lv = 0; lv < levels; ++lv) {
+        for (int t = tid; t < T; t += blockDim.x) {
+            long slot = (pos0 + t) % capacity;
+            long node = (tree_cap + slot) >> (lv + 1);
+            if (node >= 1) {
+                sum_tree[node] = sum_tree[2 * node] + sum_tree[2 * node + 1];
+                min_tree[node] = fmin(min_tree[2 * node],
+                                      min_tree[2 * node + 1]);
+            }
+        }
+        __syncthreads();
+    }
+    if (tid == 0) {
+        cnt->pos = (pos0 + T) % capacity;
+        cnt->size = cnt->size + T > capacity ? capacity : cnt->size + T;
+    }
+}
+
+// ---- synthetic replay fill (bench path: no H2D needed) ----------------------
+__global__ void k_synth_fill(float* rs, float* ra, float* rr, float* rs2,
+                             float* rd, double* sum_tree, double* min_tree,
+                             long tree_cap, long capacity, long n,
+                             int obs, int act, uint64_t seed, float alpha) {
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += (long)gridDim.x * blockDim.x) {
+        Philox4 r = philox4(seed, 0x5EEDull, (uint64_t)i);
+        uint32_t st = r.v[0];
+        for (int k = 0; k < obs; ++k) {
+            Philox4 q = philox4(seed, 1 + (uint64_t)k, (uint64_t)i);
+            rs[i * obs + k] = 2.f * u01(q.v[0]) - 1.f;
+            rs2[i * obs + k] = 2.f * u01(q.v[1]) - 1.f;
+        }
+        for (int k = 0; k < act; ++k) {
+            Philox4 q = philox4(seed, 1000 + (uint64_t)k, (uint64_t)i);
+            ra[i * act + k] = 2.f * u01(q.v[2]) - 1.f;
+        }
+        rr[i] = -u01(r.v[1]) * 10.f;
+        rd[i] = (u01(r.v[2]) < 0.01f) ? 1.f : 0.f;
+        sum_tree[tree_cap + i] = 1.0;     // max_priority(1)^alpha
+        min_tree[tree_cap + i] = 1.0;
+    }
+}
+
+// full-tree rebuild after bulk fill: one level at a time, many WGs
+__global__ void k_tree_build_level(double* sum_tree, double* min_tree,
+                                   long lo, long hi) {
+    for (long n = lo + (long)blockIdx.x * blockDim.x + threadIdx.x; n < hi;
+         n += (long)gridDim.x * blockDim.x) {
+        sum_tree[n] = sum_tree[2 * n] + sum_tree[2 * n + 1];
+        min_tree[n] = fmin(min_tree[2 * n], min_tree[2 * n + 1]);
+    }
+}
+
+// ===========================================================================
+// Host engine
+// ===========================================================================
+
+static long next_pow2(long n) { long c = 1; while (c < n) c <<= 1; return c; }
+static int ceil_div(long a, long b) { return (int)((a + b - 1) / b); }
+
+struct Net {
+    LayerDesc l[4];
+    long n_params;
+};
+
+static Net make_actor_net(int O, int A, int H) {
+    Net n{};
+    long off = 0;
+    auto add = [&](int i1, int i2, int o) {
+        LayerDesc d{i1, i2, o, off, off + (long)(i1 + i2) * o};
+        off += (long)(i1 + i2) * o + o;
+        return d;
+    };
+    n.l[0] = add(O, 0, H);
+    n.l[1] = add(H, 0, H);
+    n.l[2] = add(H, 0, H);
+    n.l[3] = add(H, 0, A);
+    n.n_params = off;
+    return n;
+}
+
+static Net make_critic_net(int O, int A, int H, int K) {
+    Net n{};
+    long off = 0;
+    auto add = [&](int i1, int i2, int o) {
+        LayerDesc d{i1, i2, o, off, off + (long)(i1 + i2) * o};
+        off += (long)(i1 + i2) * o + o;
+        return d;
+    };
+    n.l[0] = add(O, 0, H);
+    n.l[1] = add(H, A, H);
+    n.l[2] = add(H, 0, H);
+    n.l[3] = add(H, 0, K);
+    n.n_params = off;
+    return n;
+}
+
+class Engine {
+public:
+    EngineCfg cfg;
+    Net anet, cnet;
+    long tree_cap;
+    hipStream_t stream;
+    hipGraph_t graph = nullptr;
+    hipGraphExec_t graph_exec = nullptr;
+    int graph_steps = 0;
+
+    // device buffers
+    float *p_actor, *p_actor_t, *p_critic, *p_critic_t;
+    float *g_actor, *g_critic;
+    float *m_actor, *v_actor, *m_critic, *v_critic;
+    float *rs, *ra, *rr, *rs2, *rd;                 // replay SoA
+    double *sum_tree, *min_tree;
+    Counters* cnt;
+    // batch + activations workspace
+    float *bs, *ba, *br, *bs2, *bd, *bw, *pri;
+    long *bidx;
+    float *at_h1, *at_h2, *at_h3, *a2;              // actor_target path
+    float *ct_h1, *ct_h2, *ct_h3, *p_t, *m_proj;    // critic_target path
+    float *c_h1, *c_h2, *c_h3, *q;                  // critic path
+    float *dlog, *d3, *d2, *d1, *da;                // critic backward deltas
+    float *pa_h1, *pa_h2, *pa_h3, *a_out;           // actor (policy) path
+    float *pc_h1, *pc_h2, *pc_h3, *pq;              // critic(s, actor(s))
+    float *pd3, *pd2, *pdh1, *pda, *adz;            // policy backward deltas
+    float *ing_s, *ing_a, *ing_r, *ing_s2, *ing_d;  // ingestion staging
+    int ing_cap;
+
+    Engine(const EngineCfg& c) : cfg(c) {
+        anet = make_actor_net(c.obs, c.act, c.hidden);
+        cnet = make_critic_net(c.obs, c.act, c.hidden, c.atoms);
+        tree_cap = next_pow2(c.capacity);
+        HIP_CHECK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
+        alloc();
+    }
+
+    ~Engine() {
+        if (graph_exec) hipGraphExecDestroy(graph_exec);
+        if (graph) hipGraphDestroy(graph);
+        hipStreamDestroy(stream);
+        hipFree(pool_);
+    }
+
+    void* pool_ = nullptr;
+
+    template <typename T>
+    T* carve(long nelem, long& off) {
+        long bytes = nelem * (long)sizeof(T);
+        bytes = (bytes + 255) & ~255L;
+        T* p = reinterpret_cast<T*>(static_cast<char*>(pool_) + off);
+        off += bytes;
+        return p;
+    }
+
+    // One carve sequence, executed twice (count pass with null base, then
+    // assign pass) so sizes and pointers can never drift apart.
+    long layout() {
+        const int B = cfg.batch, O = cfg.obs, A = cfg.act, H = cfg.hidden,
+                  K = cfg.atoms;
+        const long C = cfg.capacity;
+        const long pa = anet.n_params, pc = cnet.n_params;
+        ing_cap = 65536;
+        long off = 0;
+        p_actor = carve<float>(pa, off);  p_actor_t = carve<float>(pa, off);
+        g_actor = carve<float>(pa, off);
+        p_critic = carve<float>(pc, off); p_critic_t = carve<float>(pc, off);
+        g_critic = carve<float>(pc, off);
+        m_actor = carve<float>(pa, off);  v_actor = carve<float>(pa, off);
+        m_critic = carve<float>(pc, off); v_critic = carve<float>(pc, off);
+        rs = carve<float>(C * O, off); ra = carve<float>(C * A, off);
+        rr = carve<float>(C, off); rs2 = carve<float>(C * O, off);
+        rd = carve<float>(C, off);
+        sum_tree = carve<double>(2 * tree_cap, off);
+        min_tree = carve<double>(2 * tree_cap, off);
+        cnt = carve<Counters>(1, off);
+        bs = carve<float>((long)B * O, off); ba = carve<float>((long)B * A, off);
+        br = carve<float>(B, off); bs2 = carve<float>((long)B * O, off);
+        bd = carve<float>(B, off); bw = carve<float>(B, off);
+        pri = carve<float>(B, off);
+        bidx = carve<long>(B, off);
+        at_h1 = carve<float>((long)B * H, off);
+        at_h2 = carve<float>((long)B * H, off);
+        at_h3 = carve<float>((long)B * H, off);
+        a2 = carve<float>((long)B * A, off);
+        ct_h1 = carve<float>((long)B * H, off);
+        ct_h2 = carve<float>((long)B * H, off);
+        ct_h3 = carve<float>((long)B * H, off);
+        p_t = carve<float>((long)B * K, off);
+        m_proj = carve<float>((long)B * K, off);
+        c_h1 = carve<float>((long)B * H, off);
+        c_h2 = carve<float>((long)B * H, off);
+        c_h3 = carve<float>((long)B * H, off);
+        q = carve<float>((long)B * K, off);
+        dlog = carve<float>((long)B * K, off);
+        d3 = carve<float>((long)B * H, off);
+        d2 = carve<float>((long)B * H, off);
+        d1 = carve<float>((long)B * H, off);
+        da = carve<float>((long)B * A, off);
+        pa_h1 = carve<float>((long)B * H, off);
+        pa_h2 = carve<float>((long)B * H, off);
+        pa_h3 = carve<float>((long)B * H, off);
+        a_out = carve<float>((long)B * A, off);
+        pc_h1 = carve<float>((long)B * H, off);
+        pc_h2 = carve<float>((long)B * H, off);
+        pc_h3 = carve<float>((long)B * H, off);
+        pq = carve<float>((long)B * K, off);
+        pd3 = carve<float>((long)B * K, off);
+        pd2 = carve<float>((long)B * H, off);
+        pdh1 = carve<float>((long)B * H, off);
+        pda = carve<float>((long)B * A, off);
+        adz = carve<float>((long)B * A, off);
+        ing_s = carve<float>((long)ing_cap * O, off);
+        ing_a = carve<float>((long)ing_cap * A, off);
+        ing_r = carve<float>(ing_cap, off);
+        ing_s2 = carve<float>((long)ing_cap * O, off);
+        ing_d = carve<float>(ing_cap, off);
+        return off;
+    }
+
+    void alloc() {
+        pool_ = nullptr;
+        long total = layout();               // count pass (null base)
+        HIP_CHECK(hipMalloc(&pool_, total));
+        HIP_CHECK(hipMemset(pool_, 0, total));
+        layout();                            // assign pass
+        Counters h{};
+        h.max_priority = 1.0f;
+        HIP_CHECK(hipMemcpy(cnt, &h, sizeof(Counters), hipMemcpyHostToDevice));
+    }
+
+    // ---------------- job builders ----------------
+    FwdJob fwd_job(const float* x1, const float* x2, const float* slab,
+                   const LayerDesc& l, float* y, int act_kind, int& wg) {
+        FwdJob j{};
+        j.x1 = x1; j.x2 = x2;
+        j.wt = slab + l.w_off; j.bias = slab + l.b_off;
+        j.y = y;
+        j.B = cfg.batch; j.in1 = l.in1; j.in2 = l.in2; j.out = l.out;
+        j.act = act_kind;
+        j.wg0 = wg;
+        j.nwg_b = ceil_div(cfg.batch, TB);
+        j.nwg_o = ceil_div(l.out, TO);
+        wg += j.nwg_b * j.nwg_o;
+        return j;
+    }
+
+    void launch_fwd(std::initializer_list<FwdJob> jobs) {
+        FwdJob a[3] = {};
+        int n = 0, wgs = 0;
+        for (auto& j : jobs) {
+            a[n++] = j;
+            wgs = j.wg0 + j.nwg_b * j.nwg_o;
+        }
+        hipLaunchKernelGGL(k_fwd3, dim3(wgs), dim3(256), 0, stream,
+                           a[0], a[1], a[2], n);
+    }
+
+    void launch_bwd(const float* dz, const float* x1, const float* x2,
+                    const float* slab, float* gslab, const LayerDesc& l,
+                    float* dx1, float* dx2, const float* h1, int prev_act,
+                    bool want_dw) {
+        BwdJob j{};
+        j.dz = dz; j.x1 = x1; j.x2 = x2;
+        j.wt = slab + l.w_off;
+        j.dwt = want_dw ? gslab + l.w_off : nullptr;
+        j.dbias = want_dw ? gslab + l.b_off : nullptr;
+        j.dx1 = dx1; j.dx2 = dx2; j.h1 = h1;
+        j.B = cfg.batch; j.in1 = l.in1; j.in2 = l.in2; j.out = l.out;
+        j.prev_act = prev_act;
+        int wg = 0;
+        if (want_dw) {
+            j.wg0_dw = 0;
+            j.nwg_dw_i = ceil_div(l.in1 + l.in2, BWT);
+            j.nwg_dw_o = ceil_div(l.out, BWT);
+            wg += j.nwg_dw_i * j.nwg_dw_o;
+        }
+        j.wg0_dx = wg;
+        if (dx1 || dx2) {
+            j.nwg_dx_b = ceil_div(cfg.batch, BXB);
+            j.nwg_dx_i = ceil_div(l.in1 + l.in2, BXI);
+            wg += j.nwg_dx_b * j.nwg_dx_i;
+        } else { j.nwg_dx_b = j.nwg_dx_i = 0; }
+        hipLaunchKernelGGL(k_bwd, dim3(wg), dim3(256), 0, stream, j);
+    }
+
+    // ---------------- the train step (one launch sequence) ----------------
+    void enqueue_step() {
+        const int B = cfg.batch, K = cfg.atoms, H = cfg.hidden;
+        const int waves_per_wg = 4;
+        int row_wgs = ceil_div(B, waves_per_wg);
+
+        hipLaunchKernelGGL(k_tick, dim3(1), dim3(64), 0, stream, cnt);
+
+        hipLaunchKernelGGL(k_per_sample, dim3(row_wgs), dim3(256), 0, stream,
+            sum_tree, min_tree, tree_cap, rs, ra, rr, rs2, rd,
+            cfg.obs, cfg.act, bs, ba, br, bs2, bd, bw, bidx, B, cnt,
+            cfg.per_beta0, (float)cfg.per_beta_iters, cfg.seed);
+
+        // P2: a_t.L1(s2) | c_t.L1(s2) | c.L1(s)
+        {
+            int wg = 0;
+            auto j0 = fwd_job(bs2, nullptr, p_actor_t, anet.l[0], at_h1,
+                              ACT_RELU, wg);
+            auto j1 = fwd_job(bs2, nullptr, p_critic_t, cnet.l[0], ct_h1,
+                              ACT_RELU, wg);
+            auto j2 = fwd_job(bs, nullptr, p_critic, cnet.l[0], c_h1,
+                              ACT_RELU, wg);
+            launch_fwd({j0, j1, j2});
+        }
+        // P3: a_t.L2 | c.L2(cat c_h1, a)
+        {
+            int wg = 0;
+            auto j0 = fwd_job(at_h1, nullptr, p_actor_t, anet.l[1], at_h2,
+                              ACT_NONE, wg);
+            auto j1 = fwd_job(c_h1, ba, p_critic, cnet.l[1], c_h2,
+                              ACT_RELU, wg);
+            launch_fwd({j0, j1});
+        }
+        // P4: a_t.L3 | c.L3
+        {
+            int wg = 0;
+            auto j0 = fwd_job(at_h2, nullptr, p_actor_t, anet.l[2], at_h3,
+                              ACT_RELU, wg);
+            auto j1 = fwd_job(c_h2, nullptr, p_critic, cnet.l[2], c_h3,
+                              ACT_RELU, wg);
+            launch_fwd({j0, j1});
+        }
+        // P5: a_t.L4 tanh | c.L4 softmax -> q
+        {
+            int wg = 0;
+            auto j0 = fwd_job(at_h3, nullptr, p_actor_t, anet.l[3], a2,
+                              ACT_TANH, wg);
+            auto j1 = fwd_job(c_h3, nullptr, p_critic, cnet.l[3], q,
+                              ACT_SOFTMAX, wg);
+            launch_fwd({j0, j1});
+        }
+        // P6-8: c_t.L2(cat ct_h1, a2), c_t.L3, c_t.L4 softmax -> p_t
+        { int wg = 0; launch_fwd({fwd_job(ct_h1, a2, p_critic_t, cnet.l[1],
+                                          ct_h2, ACT_RELU, wg)}); }
+        { int wg = 0; launch_fwd({fwd_job(ct_h2, nullptr, p_critic_t,
+                                          cnet.l[2], ct_h3, ACT_RELU, wg)}); }
+        { int wg = 0; launch_fwd({fwd_job(ct_h3, nullptr, p_critic_t,
+                                          cnet.l[3], p_t, ACT_SOFTMAX, wg)}); }
+        // P9: projection
+        hipLaunchKernelGGL(k_project, dim3(row_wgs), dim3(256),
+                           waves_per_wg * K * sizeof(float), stream,
+                           p_t, br, bd, m_proj, B, K, cfg.v_min, cfg.v_max,
+                           cfg.gamma_n);
+        // P10: CE grad + priorities
+        hipLaunchKernelGGL(k_ce_grad, dim3(row_wgs), dim3(256), 0, stream,
+                           q, m_proj, bw, dlog, pri, cnt, B, K, cfg.per_eps,
+                           cfg.is_weighting);
+        // P11-14: critic backward L4..L1
+        launch_bwd(dlog, c_h3, nullptr, p_critic, g_critic, cnet.l[3],
+                   d3, nullptr, c_h3, ACT_RELU, true);
+        launch_bwd(d3, c_h2, nullptr, p_critic, g_critic, cnet.l[2],
+                   d2, nullptr, c_h2, ACT_RELU, true);
+        launch_bwd(d2, c_h1, ba, p_critic, g_critic, cnet.l[1],
+                   d1, nullptr, c_h1, ACT_RELU, true);
+        launch_bwd(d1, bs, nullptr, p_critic, g_critic, cnet.l[0],
+                   nullptr, nullptr, nullptr, ACT_NONE, true);
+        // P15: Adam critic
+        hipLaunchKernelGGL(k_adam, dim3(256), dim3(256), 0, stream,
+                           p_critic, g_critic, m_critic, v_critic,
+                           cnet.n_params, cfg.lr_critic, 0.9f, 0.999f, 1e-8f,
+                           cnt, 0);
+        // P16: actor.L1(s) | critic'.L1(s)
+        {
+            int wg = 0;
+            auto j0 = fwd_job(bs, nullptr, p_actor, anet.l[0], pa_h1,
+                              ACT_RELU, wg);
+            auto j1 = fwd_job(bs, nullptr, p_critic, cnet.l[0], pc_h1,
+                              ACT_RELU, wg);
+            launch_fwd({j0, j1});
+        }
+        // P17-19: actor L2, L3, L4
+        { int wg = 0; launch_fwd({fwd_job(pa_h1, nullptr, p_actor, anet.l[1],
+                                          pa_h2, ACT_NONE, wg)}); }
+        { int wg = 0; launch_fwd({fwd_job(pa_h2, nullptr, p_actor, anet.l[2],
+                                          pa_h3, ACT_RELU, wg)}); }
+        { int wg = 0; launch_fwd({fwd_job(pa_h3, nullptr, p_actor, anet.l[3],
+                                          a_out, ACT_TANH, wg)}); }
+        // P20-22: critic'(s, a_out)
+        { int wg = 0; launch_fwd({fwd_job(pc_h1, a_out, p_critic, cnet.l[1],
+                                          pc_h2, ACT_RELU, wg)}); }
+        { int wg = 0; launch_fwd({fwd_job(pc_h2, nullptr, p_critic,
+                                          cnet.l[2], pc_h3, ACT_RELU, wg)}); }
+        { int wg = 0; launch_fwd({fwd_job(pc_h3, nullptr, p_critic,
+                                          cnet.l[3], pq, ACT_SOFTMAX, wg)}); }
+        // P23: policy head gradient
+        hipLaunchKernelGGL(k_policy_grad, dim3(row_wgs), dim3(256), 0, stream,
+                           pq, pd3, cnt, B, K, cfg.v_min, cfg.v_max);
+        // P24-26: dX back through critic' (no dW), ending at da
+        launch_bwd(pd3, pc_h3, nullptr, p_critic, nullptr, cnet.l[3],
+                   pd2, nullptr, pc_h3, ACT_RELU, false);
+        launch_bwd(pd2, pc_h2, nullptr, p_critic, nullptr, cnet.l[2],
+                   pdh1, nullptr, pc_h2, ACT_RELU, false);
+        launch_bwd(pdh1, pc_h1, a_out, p_critic, nullptr, cnet.l[1],
+                   nullptr, pda, pc_h1, ACT_RELU, false);
+        // P27: tanh backward at actor output
+        hipLaunchKernelGGL(k_tanh_bwd, dim3(ceil_div((long)B * cfg.act, 256)),
+                           dim3(256), 0, stream, pda, a_out, adz,
+                           (long)B * cfg.act);
+        // P28-31: actor backward L4..L1 (with dW)
+        launch_bwd(adz, pa_h3, nullptr, p_actor, g_actor, anet.l[3],
+                   pd2 /*reuse*/, nullptr, pa_h3, ACT_RELU, true);
+        launch_bwd(pd2, pa_h2, nullptr, p_actor, g_actor, anet.l[2],
+                   pdh1, nullptr, pa_h2, ACT_NONE, true);
+        launch_bwd(pdh1, pa_h1, nullptr, p_actor, g_actor, anet.l[1],
+                   pd2, nullptr, pa_h1, ACT_RELU, true);
+        launch_bwd(pd2, bs, nullptr, p_actor, g_actor, anet.l[0],
+                   nullptr, nullptr, nullptr, ACT_NONE, true);
+        // P32: Adam actor
+        hipLaunchKernelGGL(k_adam, dim3(256), dim3(256), 0, stream,
+                           p_actor, g_actor, m_actor, v_actor,
+                           anet.n_params, cfg.lr_actor, 0.9f, 0.999f, 1e-8f,
+                           cnt, 1);
+        // P33: soft updates (both nets)
+        hipLaunchKernelGGL(k_soft_update, dim3(256), dim3(256), 0, stream,
+                           p_actor_t, p_actor, anet.n_params,
+                           p_critic_t, p_critic, cnet.n_params, cfg.tau);
+        // P34: PER priority write-back
+        hipLaunchKernelGGL(k_per_update, dim3(1), dim3(256), 0, stream,
+                           sum_tree, min_tree, tree_cap, bidx, pri, B,
+                           cfg.per_alpha, cnt);
+    }
+
+    void step(int n) {
+        for (int i = 0; i < n; ++i) enqueue_step();
+        HIP_CHECK(hipStreamSynchronize(stream));
+    }
+
+    void capture(int steps_per_graph) {
+        if (graph_exec) { hipGraphExecDestroy(graph_exec); graph_exec = nullptr; }
+        if (graph) { hipGraphDestroy(graph); graph = nullptr; }
+        HIP_CHECK(hipStreamBeginCapture(stream, hipStreamCaptureModeThreadLocal));
+        for (int i = 0; i < steps_per_graph; ++i) enqueue_step();
+        HIP_CHECK(hipStreamEndCapture(stream, &graph));
+        HIP_CHECK(hipGraphInstantiate(&graph_exec, graph, nullptr, nullptr, 0));
+        graph_steps = steps_per_graph;
+    }
+
+    void replay(int iters) {
+        if (!graph_exec) throw std::runtime_error("no graph captured");
+        for (int i = 0; i < iters; ++i)
+            HIP_CHECK(hipGraphLaunch(graph_exec, stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
+    }
+
+    // non-blocking variants for overlap with host work
+    void replay_async(int iters) {
+        for (int i = 0; i < iters; ++i)
+            HIP_CHECK(hipGraphLaunch(graph_exec, stream));
+    }
+    void sync() { HIP_CHECK(hipStreamSynchronize(stream)); }
+
+    // ------------- replay ops -------------
+    void synth_fill(long n, uint64_t seed) {
+        if (n > cfg.capacity) n = cfg.capacity;
+        hipLaunchKernelGGL(k_synth_fill, dim3(1024), dim3(256), 0, stream,
+                           rs, ra, rr, rs2, rd, sum_tree, min_tree, tree_cap,
+                           cfg.capacity, n, cfg.obs, cfg.act, seed,
+                           cfg.per_alpha);
+        for (long lo = tree_cap / 2; lo >= 1; lo /= 2)
+            hipLaunchKernelGGL(k_tree_build_level, dim3(256), dim3(256), 0,
+                               stream, sum_tree, min_tree, lo,
+                               lo == 0 ? 1 : 2 * lo);
+        Counters h{};
+        HIP_CHECK(hipMemcpyAsync(&h, cnt, sizeof(h), hipMemcpyDeviceToHost,
+                                 stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
+        h.size = n; h.pos = n % cfg.capacity; h.max_priority = 1.0f;
+        HIP_CHECK(hipMemcpy(cnt, &h, sizeof(h), hipMemcpyHostToDevice));
+    }
+
+    void ingest(const float* hs, const float* ha, const float* hr,
+                const float* hs2, const float* hd, int T) {
+        if (T > ing_cap) throw std::runtime_error("ingest batch too large");
+        const int O = cfg.obs, A = cfg.act;
+        HIP_CHECK(hipMemcpyAsync(ing_s, hs, (long)T * O * 4,
+                                 hipMemcpyHostToDevice, stream));
+        HIP_CHECK(hipMemcpyAsync(ing_a, ha, (long)T * A * 4,
+                                 hipMemcpyHostToDevice, stream));
+        HIP_CHECK(hipMemcpyAsync(ing_r, hr, (long)T * 4,
+                                 hipMemcpyHostToDevice, stream));
+        HIP_CHECK(hipMemcpyAsync(ing_s2, hs2, (long)T * O * 4,
+                                 hipMemcpyHostToDevice, stream));
+        HIP_CHECK(hipMemcpyAsync(ing_d, hd, (long)T * 4,
+                                 hipMemcpyHostToDevice, stream));
+        hipLaunchKernelGGL(k_replay_add, dim3(1), dim3(1024), 0, stream,
+                           rs, ra, rr, rs2, rd, sum_tree, min_tree, tree_cap,
+                           cfg.capacity, ing_s, ing_a, ing_r, ing_s2, ing_d,
+                           T, O, A, cfg.per_alpha, cnt);
+        HIP_CHECK(hipStreamSynchronize(stream));
+    }
+
+    // ------------- param I/O (host pointers, already transposed [in][out]) --
+    void load_slab(float* dst, const float* src, long n) {
+        HIP_CHECK(hipMemcpy(dst, src, n * 4, hipMemcpyHostToDevice));
+    }
+    void store_slab(float* dst, const float* src, long n) {
+        HIP_CHECK(hipMemcpy(dst, src, n * 4, hipMemcpyDeviceToHost));
+    }
+
+    Counters read_counters() {
+        Counters h{};
+        HIP_CHECK(hipMemcpy(&h, cnt, sizeof(h), hipMemcpyDeviceToHost));
+        return h;
+    }
+
+    // actor forward for eval/smoke: x[B,obs] (device via staging) -> a[B,act]
+    void actor_forward(const float* hx, float* hy, int n) {
+        // reuse batch buffers (requires n <= batch)
+        if (n > cfg.batch) throw std::runtime_error("actor_forward: n > batch");
+        HIP_CHECK(hipMemcpyAsync(bs, hx, (long)n * cfg.obs * 4,
+                                 hipMemcpyHostToDevice, stream));
+        int wg;
+        wg = 0; launch_fwd({fwd_job(bs, nullptr, p_actor, anet.l[0], pa_h1,
+                                    ACT_RELU, wg)});
+        wg = 0; launch_fwd({fwd_job(pa_h1, nullptr, p_actor, anet.l[1], pa_h2,
+                                    ACT_NONE, wg)});
+        wg = 0; launch_fwd({fwd_job(pa_h2, nullptr, p_actor, anet.l[2], pa_h3,
+                                    ACT_RELU, wg)});
+        wg = 0; launch_fwd({fwd_job(pa_h3, nullptr, p_actor, anet.l[3], a_out,
+                                    ACT_TANH, wg)});
+        HIP_CHECK(hipMemcpyAsync(hy, a_out, (long)n * cfg.act * 4,
+                                 hipMemcpyDeviceToHost, stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
+    }
+};
+
+}  // namespace d4pg
