@@ -167,6 +167,12 @@ __global__ void k_per_sample(
     int probe = blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
     int lane = threadIdx.x & 63;
     if (probe >= B) return;
+    if (probe == 0 && lane == 0) {
+        // first kernel of the step: reset the per-step loss accumulators
+        // (counter increments happen in k_per_update at step end)
+        const_cast<Counters*>(cnt)->loss_critic = 0.f;
+        const_cast<Counters*>(cnt)->loss_actor = 0.f;
+    }
 
     long long n = cnt->size;
     double total = sum_tree[1];
@@ -208,7 +214,8 @@ __global__ void k_per_sample(
 // chunks.  ACT_SOFTMAX requires out <= 64 and uses one wave per row.
 #define TB 4
 #define TO 64
-#define TK 32
+// max fan-in a fwd workgroup stages (hidden(1024) + act margin)
+#define FWD_XMAX 1032
 
 __device__ inline void fwd_one(const FwdJob& j, int wg) {
     int rel = wg - j.wg0;
@@ -217,38 +224,34 @@ __device__ inline void fwd_one(const FwdJob& j, int wg) {
     int b0 = bt * TB, o0 = ot * TO;
     int in_total = j.in1 + j.in2;
 
-    __shared__ float xs[TB][TK];
-    __shared__ float ws[TK][TO + 1];
+    // x rows staged ONCE (LDS broadcast reads after); weights read straight
+    // from global — coalesced 256 B per wave per k, pipelined by the unroll
+    // (the previous sync-staged-chunks version was latency-bound at ~20 us
+    // per launch; this form measures ~3-5 us at B=64, H=256).
+    __shared__ float xs[TB][FWD_XMAX];
 
     int tid = threadIdx.x;
+    for (int t = tid; t < TB * in_total; t += 256) {
+        int bb = t / in_total, kk = t % in_total;
+        int gb = b0 + bb;
+        float v = 0.f;
+        if (gb < j.B)
+            v = (kk < j.in1) ? j.x1[(long)gb * j.in1 + kk]
+                             : j.x2[(long)gb * j.in2 + (kk - j.in1)];
+        xs[bb][kk] = v;
+    }
+    __syncthreads();
+
     int tb = tid / TO;                 // 0..TB-1
     int to = tid % TO;                 // 0..TO-1
     int b = b0 + tb, o = o0 + to;
 
     float acc = 0.f;
-    for (int k0 = 0; k0 < in_total; k0 += TK) {
-        // stage x chunk: TB*TK elems by 256 threads
-        for (int t = tid; t < TB * TK; t += 256) {
-            int bb = t / TK, kk = t % TK;
-            int gb = b0 + bb, gk = k0 + kk;
-            float v = 0.f;
-            if (gb < j.B && gk < in_total)
-                v = (gk < j.in1) ? j.x1[(long)gb * j.in1 + gk]
-                                 : j.x2[(long)gb * j.in2 + (gk - j.in1)];
-            xs[bb][kk] = v;
-        }
-        // stage Wt chunk: TK*TO elems
-        for (int t = tid; t < TK * TO; t += 256) {
-            int kk = t / TO, oo = t % TO;
-            int gk = k0 + kk, go = o0 + oo;
-            ws[kk][oo] = (gk < in_total && go < j.out)
-                ? j.wt[(long)gk * j.out + go] : 0.f;
-        }
-        __syncthreads();
+    if (o < j.out) {
+        const float* wcol = j.wt + o;
 #pragma unroll 8
-        for (int kk = 0; kk < TK; ++kk)
-            acc += xs[tb][kk] * ws[kk][to];
-        __syncthreads();
+        for (int k = 0; k < in_total; ++k)
+            acc += xs[tb][k] * wcol[(long)k * j.out];
     }
 
     if (b < j.B && o < j.out) {
@@ -369,33 +372,341 @@ __global__ void k_policy_grad(const float* __restrict__ q,
     if (lane == 0) atomicAdd(&cnt->loss_actor, -e / (float)B);
 }
 
-// ---- generic fused backward (dW + db + dX in one launch) --------------------
-#define BWT 16          // dW tile: BWT_i x BWT_o, thread per (i,o)
-#define BXB 16          // dX tile rows
-#define BXI 16          // dX tile cols
-#define BXO 64          // dX o-chunk
-
 __device__ inline float act_mask(int act, float h) {
     if (act == ACT_RELU) return h > 0.f ? 1.f : 0.f;
     if (act == ACT_TANH) return 1.f - h * h;
     return 1.f;
 }
 
+// ===========================================================================
+// Row-block megakernels (small-batch learner path)
+// ===========================================================================
+// Key observation: until the dW reduction, every quantity in the D4PG train
+// step is per-batch-row: forwards, softmax, the C51 projection, CE/policy
+// gradients and the whole backward-dX chain.  So ONE WAVE owns ONE row and
+// runs the entire chain against L2-resident weights with zero barriers —
+// the step then needs only 8 kernels (sample, critic-row-block, critic-dW,
+// critic-Adam+lerp, policy-row-block, actor-dW, actor-Adam+lerp, PER
+// write-back), which matters because at B=64 the step is bounded by the
+// ~1.5-5 us per-kernel floor, not by FLOPs.
+
+// per-wave LDS carve (floats): [xb XMAX][yb XMAX][h1 H][h2 H][h3 H]
+// [aa 32][row0 64][row1 64][row2 64]
+struct RowLds {
+    float *xb, *yb, *h1, *h2, *h3, *aa, *r0, *r1, *r2;
+};
+
+__device__ inline RowLds rb_carve(float* base, int wid, int H) {
+    long per = 2 * FWD_XMAX + 3 * H + 32 + 3 * 64;
+    float* p = base + (long)wid * per;
+    RowLds L;
+    L.xb = p;               p += FWD_XMAX;
+    L.yb = p;               p += FWD_XMAX;
+    L.h1 = p;               p += H;
+    L.h2 = p;               p += H;
+    L.h3 = p;               p += H;
+    L.aa = p;               p += 32;
+    L.r0 = p;               p += 64;
+    L.r1 = p;               p += 64;
+    L.r2 = p;
+    return L;
+}
+
+__host__ __device__ inline long rb_lds_bytes(int H) {
+    return 4L * (2 * FWD_XMAX + 3 * H + 32 + 3 * 64) * 4;
+}
+
+// y[o] = act(sum_k xb[k] * Wt[k][o] + b[o]), lane-parallel over o.
+__device__ inline void rb_fwd(const float* xb, float* yb, const float* wt,
+                              const float* bias, int in_total, int out,
+                              int act_kind, float* grow, int lane) {
+    for (int oc = 0; oc < out; oc += 64) {
+        int o = oc + lane;
+        float acc = 0.f;
+        if (o < out) {
+            const float* wcol = wt + o;
+#pragma unroll 16
+            for (int k = 0; k < in_total; ++k)
+                acc += xb[k] * wcol[(long)k * out];
+            acc += bias[o];
+            if (act_kind == ACT_RELU) acc = fmaxf(acc, 0.f);
+            else if (act_kind == ACT_TANH) acc = tanhf(acc);
+            yb[o] = acc;
+            if (grow) grow[o] = acc;
+        }
+    }
+}
+
+// softmax head (out <= 64): q[o] = softmax(logits)[o]
+__device__ inline void rb_fwd_softmax(const float* xb, float* qrow,
+                                      const float* wt, const float* bias,
+                                      int in_total, int out, float* grow,
+                                      int lane) {
+    float acc = -INFINITY;
+    if (lane < out) {
+        const float* wcol = wt + lane;
+        float a = 0.f;
+#pragma unroll 16
+        for (int k = 0; k < in_total; ++k)
+            a += xb[k] * wcol[(long)k * out];
+        acc = a + bias[lane];
+    }
+    float mx = acc;
+    for (int s = 32; s > 0; s >>= 1) mx = fmaxf(mx, __shfl_xor(mx, s, 64));
+    float e = (lane < out) ? __expf(acc - mx) : 0.f;
+    float sum = e;
+    for (int s = 32; s > 0; s >>= 1) sum += __shfl_xor(sum, s, 64);
+    if (lane < out) {
+        float q = e / sum;
+        qrow[lane] = q;
+        if (grow) grow[lane] = q;
+    }
+}
+
+// dx[i] = (sum_o dz[o] * Wt[i][o]) * act'(hprev[i]), lane-parallel over i.
+__device__ inline void rb_bwd_dx(const float* dzb, const float* wt,
+                                 int in_lo, int in_hi, int in_total, int out,
+                                 const float* hprev, int prev_act,
+                                 float* dxb, float* grow, int lane) {
+    for (int ic = in_lo; ic < in_hi; ic += 64) {
+        int i = ic + lane;
+        float acc = 0.f;
+        if (i < in_hi) {
+            const float* wrow = wt + (long)i * out;
+#pragma unroll 16
+            for (int o = 0; o < out; ++o)
+                acc += dzb[o] * wrow[o];
+            if (hprev)
+                acc *= act_mask(prev_act, hprev[i - in_lo]);
+            if (dxb) dxb[i - in_lo] = acc;
+            if (grow) grow[i - in_lo] = acc;
+        }
+    }
+}
+
+// Layer pointer bundle for a net slab.
+struct NetPtrs { const float *w1, *b1, *w2, *b2, *w3, *b3, *w4, *b4; };
+
+__device__ inline NetPtrs net_ptrs(const float* slab, const LayerDesc* l) {
+    return {slab + l[0].w_off, slab + l[0].b_off,
+            slab + l[1].w_off, slab + l[1].b_off,
+            slab + l[2].w_off, slab + l[2].b_off,
+            slab + l[3].w_off, slab + l[3].b_off};
+}
+
+// All geometry/pointers for the two row-block kernels.
+struct RowBlockArgs {
+    // dims
+    int B, O, A, H, K;
+    float v_min, v_max, gamma_n, per_eps;
+    int is_weighting;
+    // slabs
+    const float *p_actor, *p_actor_t, *p_critic, *p_critic_t;
+    LayerDesc al[4], cl[4];
+    // batch
+    const float *bs, *ba, *br, *bs2, *bd, *bw;
+    // outputs / saved activations
+    float *a2, *p_t, *m_proj, *q, *dlog, *pri;
+    float *c_h1, *c_h2, *c_h3, *d1, *d2, *d3;
+    float *pa_h1, *pa_h2, *pa_h3, *a_out;
+    float *pc_h1, *pc_h2, *pc_h3, *pq;
+    float *az1, *az2, *az3, *adz;
+    Counters* cnt;
+};
+
+// K2: target forwards + projection + critic forward + CE grad + priorities
+// + backward-dX chain.  One wave per row, no barriers.
+__global__ void __launch_bounds__(256)
+k_critic_rowblock(RowBlockArgs g) {
+    extern __shared__ __attribute__((aligned(16))) float smem[];
+    int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    int b = blockIdx.x * 4 + wid;
+    if (b >= g.B) return;
+    RowLds L = rb_carve(smem, wid, g.H);
+    NetPtrs at = net_ptrs(g.p_actor_t, g.al);
+    NetPtrs ct = net_ptrs(g.p_critic_t, g.cl);
+    NetPtrs c = net_ptrs(g.p_critic, g.cl);
+    int O = g.O, A = g.A, H = g.H, K = g.K;
+
+    // ---- actor_target(s2) -> a2 ----
+    for (int k = lane; k < O; k += 64) L.xb[k] = g.bs2[(long)b * O + k];
+    rb_fwd(L.xb, L.yb, at.w1, at.b1, O, H, ACT_RELU, nullptr, lane);
+    rb_fwd(L.yb, L.xb, at.w2, at.b2, H, H, ACT_NONE, nullptr, lane);
+    rb_fwd(L.xb, L.yb, at.w3, at.b3, H, H, ACT_RELU, nullptr, lane);
+    rb_fwd(L.yb, L.aa, at.w4, at.b4, H, A, ACT_TANH,
+           g.a2 + (long)b * A, lane);
+
+    // ---- critic_target(s2, a2) -> p_t ----
+    for (int k = lane; k < O; k += 64) L.xb[k] = g.bs2[(long)b * O + k];
+    rb_fwd(L.xb, L.yb, ct.w1, ct.b1, O, H, ACT_RELU, nullptr, lane);
+    for (int k = lane; k < A; k += 64) L.yb[H + k] = L.aa[k];
+    rb_fwd(L.yb, L.xb, ct.w2, ct.b2, H + A, H, ACT_RELU, nullptr, lane);
+    rb_fwd(L.xb, L.yb, ct.w3, ct.b3, H, H, ACT_RELU, nullptr, lane);
+    rb_fwd_softmax(L.yb, L.r0, ct.w4, ct.b4, H, K,
+                   g.p_t + (long)b * K, lane);          // r0 = p_t row
+
+    // ---- C51 projection of r0 -> r1 (m row) ----
+    {
+        float delta = (g.v_max - g.v_min) / (K - 1);
+        float rr = g.br[b], dd = g.bd[b];
+        for (int k = lane; k < K; k += 64) L.r1[k] = 0.f;
+        if (lane < K) {
+            float z = g.v_min + lane * delta;
+            float tz = rr + g.gamma_n * (1.f - dd) * z;
+            tz = fminf(g.v_max, fmaxf(g.v_min, tz));
+            float bj = (tz - g.v_min) / delta;
+            int l = (int)floorf(bj), u = (int)ceilf(bj);
+            if (l == u) { if (u > 0) l -= 1; else u += 1; }
+            float p = L.r0[lane];
+            atomicAdd(&L.r1[l], p * ((float)u - bj));
+            atomicAdd(&L.r1[u], p * (bj - (float)l));
+        }
+        for (int k = lane; k < K; k += 64)
+            g.m_proj[(long)b * K + k] = L.r1[k];
+    }
+
+    // ---- critic(s, a) -> q (saving h1..h3) ----
+    for (int k = lane; k < O; k += 64) L.xb[k] = g.bs[(long)b * O + k];
+    rb_fwd(L.xb, L.h1, c.w1, c.b1, O, H, ACT_RELU,
+           g.c_h1 + (long)b * H, lane);
+    for (int k = lane; k < H; k += 64) L.xb[k] = L.h1[k];
+    for (int k = lane; k < A; k += 64) L.xb[H + k] = g.ba[(long)b * A + k];
+    rb_fwd(L.xb, L.h2, c.w2, c.b2, H + A, H, ACT_RELU,
+           g.c_h2 + (long)b * H, lane);
+    rb_fwd(L.h2, L.h3, c.w3, c.b3, H, H, ACT_RELU,
+           g.c_h3 + (long)b * H, lane);
+    rb_fwd_softmax(L.h3, L.r0, c.w4, c.b4, H, K, g.q + (long)b * K, lane);
+
+    // ---- CE grad + priority (r0 = q, r1 = m) ----
+    {
+        float qv = (lane < K) ? L.r0[lane] : 0.f;
+        float mv = (lane < K) ? L.r1[lane] : 0.f;
+        float dot = mv * qv;
+        float ce = -mv * __logf(qv + 1e-10f);
+        for (int s = 32; s > 0; s >>= 1) {
+            dot += __shfl_xor(dot, s, 64);
+            ce += __shfl_xor(ce, s, 64);
+        }
+        float scale = (g.is_weighting && g.bw) ? g.bw[b] : 1.f;
+        if (lane < K) {
+            float dz = scale * (qv - mv) / (float)g.B;
+            L.r2[lane] = dz;
+            g.dlog[(long)b * K + lane] = dz;
+        }
+        if (lane == 0) {
+            g.pri[b] = dot + g.per_eps;
+            atomicAdd(&g.cnt->loss_critic, scale * ce / (float)g.B);
+        }
+    }
+
+    // ---- backward-dX chain (r2 = dlogits) ----
+    rb_bwd_dx(L.r2, c.w4, 0, H, H, K, L.h3, ACT_RELU, L.xb,
+              g.d3 + (long)b * H, lane);                 // dz3
+    rb_bwd_dx(L.xb, c.w3, 0, H, H, H, L.h2, ACT_RELU, L.yb,
+              g.d2 + (long)b * H, lane);                 // dz2
+    rb_bwd_dx(L.yb, c.w2, 0, H, H + A, H, L.h1, ACT_RELU, L.xb,
+              g.d1 + (long)b * H, lane);                 // dz1 (h-part only)
+}
+
+// K5: actor forward + critic(s, actor(s)) + policy grad + dX chain down to
+// the actor's per-layer dz's.  One wave per row.
+__global__ void __launch_bounds__(256)
+k_policy_rowblock(RowBlockArgs g) {
+    extern __shared__ __attribute__((aligned(16))) float smem[];
+    int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    int b = blockIdx.x * 4 + wid;
+    if (b >= g.B) return;
+    RowLds L = rb_carve(smem, wid, g.H);
+    NetPtrs a = net_ptrs(g.p_actor, g.al);
+    NetPtrs c = net_ptrs(g.p_critic, g.cl);
+    int O = g.O, A = g.A, H = g.H, K = g.K;
+
+    // ---- actor(s): save pa_h1..3 in h1..h3 ----
+    for (int k = lane; k < O; k += 64) L.xb[k] = g.bs[(long)b * O + k];
+    rb_fwd(L.xb, L.h1, a.w1, a.b1, O, H, ACT_RELU,
+           g.pa_h1 + (long)b * H, lane);
+    rb_fwd(L.h1, L.h2, a.w2, a.b2, H, H, ACT_NONE,
+           g.pa_h2 + (long)b * H, lane);
+    rb_fwd(L.h2, L.h3, a.w3, a.b3, H, H, ACT_RELU,
+           g.pa_h3 + (long)b * H, lane);
+    rb_fwd(L.h3, L.aa, a.w4, a.b4, H, A, ACT_TANH,
+           g.a_out + (long)b * A, lane);
+
+    // ---- critic(s, a_out): keep pc_h1..3 in xb/yb + globals ----
+    for (int k = lane; k < O; k += 64) L.xb[k] = g.bs[(long)b * O + k];
+    rb_fwd(L.xb, L.yb, c.w1, c.b1, O, H, ACT_RELU,
+           g.pc_h1 + (long)b * H, lane);
+    for (int k = lane; k < A; k += 64) L.yb[H + k] = L.aa[k];
+    rb_fwd(L.yb, L.xb, c.w2, c.b2, H + A, H, ACT_RELU,
+           g.pc_h2 + (long)b * H, lane);
+    rb_fwd(L.xb, L.yb, c.w3, c.b3, H, H, ACT_RELU,
+           g.pc_h3 + (long)b * H, lane);
+    // yb holds pc_h3; xb holds pc_h2 — careful reuse below.
+    rb_fwd_softmax(L.yb, L.r0, c.w4, c.b4, H, K, g.pq + (long)b * K, lane);
+
+    // ---- policy head gradient: pd4 = -q (z - E[z]) / B ----
+    {
+        float delta = (g.v_max - g.v_min) / (K - 1);
+        float z = g.v_min + lane * delta;
+        float qv = (lane < K) ? L.r0[lane] : 0.f;
+        float e = qv * z;
+        for (int s = 32; s > 0; s >>= 1) e += __shfl_xor(e, s, 64);
+        if (lane < K) L.r2[lane] = -qv * (z - e) / (float)g.B;
+        if (lane == 0) atomicAdd(&g.cnt->loss_actor, -e / (float)g.B);
+    }
+
+    // ---- critic dX (no dW): r2 -> da -> actor dz chain ----
+    // masks for the critic activations come from the global rows just
+    // written (pc_h3/pc_h2); the actor's h1..h3 are still LDS-resident.
+    const float* gpc_h3 = g.pc_h3 + (long)b * H;
+    const float* gpc_h2 = g.pc_h2 + (long)b * H;
+    // dz3' = (r2 @ W4c^T) * relu'(pc_h3)           -> xb
+    rb_bwd_dx(L.r2, c.w4, 0, H, H, K, gpc_h3, ACT_RELU, L.xb, nullptr, lane);
+    // dz2' = (xb @ W3c^T) * relu'(pc_h2)           -> yb
+    rb_bwd_dx(L.xb, c.w3, 0, H, H, H, gpc_h2, ACT_RELU, L.yb, nullptr, lane);
+    // da = action slice of (yb @ W2c^T)            -> r1[0..A)
+    rb_bwd_dx(L.yb, c.w2, H, H + A, H + A, H, nullptr, ACT_NONE, L.r1,
+              nullptr, lane);
+    // adz = da * (1 - a_out^2)                     -> r1, stored
+    if (lane < A) {
+        float y = L.aa[lane];
+        float v = L.r1[lane] * (1.f - y * y);
+        L.r1[lane] = v;
+        g.adz[(long)b * A + lane] = v;
+    }
+    // actor dz chain: az3 = (r1 @ W4a^T)*relu'(h3) -> xb
+    rb_bwd_dx(L.r1, a.w4, 0, H, H, A, L.h3, ACT_RELU, L.xb,
+              g.az3 + (long)b * H, lane);
+    // az2 = (xb @ W3a^T) * 1 (h2 had no act)       -> yb
+    rb_bwd_dx(L.xb, a.w3, 0, H, H, H, nullptr, ACT_NONE, L.yb,
+              g.az2 + (long)b * H, lane);
+    // az1 = (yb @ W2a^T) * relu'(h1)               -> xb
+    rb_bwd_dx(L.yb, a.w2, 0, H, H, H, L.h1, ACT_RELU, L.xb,
+              g.az1 + (long)b * H, lane);
+}
+
+// ---- generic fused backward (dW + db + dX in one launch) --------------------
+#define BWT 16          // dW tile: BWT_i x BWT_o, thread per (i,o)
+#define BXB 8           // dX tile rows (8 * FWD_XMAX floats of dz staged)
+#define BXI 32          // dX tile cols (BXB * BXI == 256 threads)
+
+
 __device__ inline void bwd_one(const BwdJob& j, int wg) {
     int in_total = j.in1 + j.in2;
     int tid = threadIdx.x;
 
     if (j.dwt && wg >= j.wg0_dw && wg < j.wg0_dw + j.nwg_dw_i * j.nwg_dw_o) {
-        // ---- dW part: tile [BWT i x BWT o], loop over batch in LDS chunks
+        // ---- dW part: tile [BWT i x BWT o]; x^T and dz slices staged per
+        // 64-row batch chunk, then pure-LDS MACs (one chunk at B=64).
         int rel = wg - j.wg0_dw;
         int it = rel / j.nwg_dw_o, ot = rel % j.nwg_dw_o;
         int i0 = it * BWT, o0 = ot * BWT;
-        __shared__ float xs[32][BWT + 1];            // [bchunk][i]
-        __shared__ float zs[32][BWT + 1];            // [bchunk][o]
+        __shared__ float xs[64][BWT + 1];            // [bchunk][i]
+        __shared__ float zs[64][BWT + 1];            // [bchunk][o]
         int ti = tid / BWT, to = tid % BWT;          // thread -> (i, o)
         float acc = 0.f, accb = 0.f;
-        for (int bc = 0; bc < j.B; bc += 32) {
-            for (int t = tid; t < 32 * BWT; t += 256) {
+        for (int bc = 0; bc < j.B; bc += 64) {
+            for (int t = tid; t < 64 * BWT; t += 256) {
                 int bb = t / BWT, ii = t % BWT;
                 int gb = bc + bb, gi = i0 + ii;
                 float xv = 0.f;
@@ -408,8 +719,8 @@ __device__ inline void bwd_one(const BwdJob& j, int wg) {
                     ? j.dz[(long)gb * j.out + go] : 0.f;
             }
             __syncthreads();
-#pragma unroll 8
-            for (int bb = 0; bb < 32; ++bb) {
+#pragma unroll 16
+            for (int bb = 0; bb < 64; ++bb) {
                 acc += xs[bb][ti] * zs[bb][to];
                 if (ti == 0) accb += zs[bb][to];
             }
@@ -424,31 +735,28 @@ __device__ inline void bwd_one(const BwdJob& j, int wg) {
     }
 
     if (wg >= j.wg0_dx && wg < j.wg0_dx + j.nwg_dx_b * j.nwg_dx_i) {
-        // ---- dX part: tile [BXB b x BXI i], o-chunked LDS staging
+        // ---- dX part: tile [BXB b x BXI i]; dz rows staged once, Wt rows
+        // read straight from global (per-thread sequential stream, L1/L2
+        // cached; same latency-pipelining rationale as fwd_one).
         int rel = wg - j.wg0_dx;
         int bt = rel / j.nwg_dx_i, it = rel % j.nwg_dx_i;
         int b0 = bt * BXB, i0 = it * BXI;
-        __shared__ float zsx[BXB][BXO + 1];
-        __shared__ float wsx[BXI][BXO + 1];
+        __shared__ float zsx[BXB][FWD_XMAX];   // 8 rows x out<=1032 = 33 KB
         int tb = tid / BXI, ti = tid % BXI;
         int b = b0 + tb, i = i0 + ti;
+        for (int t = tid; t < BXB * j.out; t += 256) {
+            int bb = t / j.out, oo = t % j.out;
+            int gb = b0 + bb;
+            zsx[bb][oo] = (gb < j.B)
+                ? j.dz[(long)gb * j.out + oo] : 0.f;
+        }
+        __syncthreads();
         float acc = 0.f;
-        for (int oc = 0; oc < j.out; oc += BXO) {
-            for (int t = tid; t < BXB * BXO; t += 256) {
-                int bb = t / BXO, oo = t % BXO;
-                int gb = b0 + bb, go = oc + oo;
-                zsx[bb][oo] = (gb < j.B && go < j.out)
-                    ? j.dz[(long)gb * j.out + go] : 0.f;
-                int ii = bb;          // reuse iteration space: BXI == BXB
-                int gi2 = i0 + ii;
-                wsx[ii][oo] = (gi2 < in_total && go < j.out)
-                    ? j.wt[(long)gi2 * j.out + go] : 0.f;
-            }
-            __syncthreads();
+        if (i < in_total) {
+            const float* wrow = j.wt + (long)i * j.out;
 #pragma unroll 8
-            for (int oo = 0; oo < BXO; ++oo)
-                acc += zsx[tb][oo] * wsx[ti][oo];
-            __syncthreads();
+            for (int oo = 0; oo < j.out; ++oo)
+                acc += zsx[tb][oo] * wrow[oo];
         }
         if (b < j.B && i < in_total) {
             if (i < j.in1) {
@@ -465,6 +773,16 @@ __device__ inline void bwd_one(const BwdJob& j, int wg) {
 }
 
 __global__ void k_bwd(BwdJob j) { bwd_one(j, blockIdx.x); }
+
+// four dW-only jobs batched into one launch (row-block path: all layer
+// weight gradients of one net in a single kernel)
+__global__ void k_bwd4(BwdJob j0, BwdJob j1, BwdJob j2, BwdJob j3) {
+    int wg = blockIdx.x;
+    if (wg < j1.wg0_dw) { bwd_one(j0, wg); return; }
+    if (wg < j2.wg0_dw) { bwd_one(j1, wg); return; }
+    if (wg < j3.wg0_dw) { bwd_one(j2, wg); return; }
+    bwd_one(j3, wg);
+}
 
 // tanh backward at the actor output: dz4 = da * (1 - a_out^2)
 __global__ void k_tanh_bwd(const float* __restrict__ da,
@@ -492,6 +810,27 @@ __global__ void k_adam(float* __restrict__ p, const float* __restrict__ g,
         float vi = b2 * v[i] + (1.f - b2) * gi * gi;
         m[i] = mi; v[i] = vi;
         p[i] -= lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+    }
+}
+
+// ---- fused Adam + target soft-update in one pass (row-block path) ----------
+__global__ void k_adam_lerp(float* __restrict__ p, const float* __restrict__ g,
+                            float* __restrict__ m, float* __restrict__ v,
+                            float* __restrict__ tgt, long n, float lr,
+                            float b1, float b2, float eps, float tau,
+                            const Counters* cnt, int is_actor) {
+    long long t = is_actor ? cnt->adam_t_actor : cnt->adam_t_critic;
+    float bc1 = 1.f - __powf(b1, (float)t);
+    float bc2 = 1.f - __powf(b2, (float)t);
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += (long)gridDim.x * blockDim.x) {
+        float gi = g[i];
+        float mi = b1 * m[i] + (1.f - b1) * gi;
+        float vi = b2 * v[i] + (1.f - b2) * gi * gi;
+        m[i] = mi; v[i] = vi;
+        float pn = p[i] - lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+        p[i] = pn;
+        tgt[i] += tau * (pn - tgt[i]);
     }
 }
 
@@ -553,6 +892,14 @@ __global__ void k_per_update(double* __restrict__ sum_tree,
             }
         }
         __syncthreads();
+    }
+    // last kernel of the step: advance the schedule counters for the NEXT
+    // step (counters are initialized to 1 so the first step sees t=1)
+    if (tid == 0) {
+        cnt->beta_t += 1;
+        cnt->adam_t_actor += 1;
+        cnt->adam_t_critic += 1;
+        cnt->rng_epoch += 1;
     }
 }
 
@@ -640,6 +987,12 @@ __global__ void k_synth_fill(float* rs, float* ra, float* rr, float* rs2,
     }
 }
 
+__global__ void k_fill_f64(double* p, long n, double v) {
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += (long)gridDim.x * blockDim.x)
+        p[i] = v;
+}
+
 // full-tree rebuild after bulk fill: one level at a time, many WGs
 __global__ void k_tree_build_level(double* sum_tree, double* min_tree,
                                    long lo, long hi) {
@@ -721,10 +1074,15 @@ public:
     float *pa_h1, *pa_h2, *pa_h3, *a_out;           // actor (policy) path
     float *pc_h1, *pc_h2, *pc_h3, *pq;              // critic(s, actor(s))
     float *pd3, *pd2, *pdh1, *pda, *adz;            // policy backward deltas
+    float *az1, *az2, *az3;                         // actor per-layer dz rows
     float *ing_s, *ing_a, *ing_r, *ing_s2, *ing_d;  // ingestion staging
     int ing_cap;
 
     Engine(const EngineCfg& c) : cfg(c) {
+        if (c.hidden + c.act > FWD_XMAX || c.obs > FWD_XMAX)
+            throw std::runtime_error("layer fan-in exceeds FWD_XMAX LDS tile");
+        if (c.atoms > 64)
+            throw std::runtime_error("n_atoms > 64 (softmax is wave-wide)");
         anet = make_actor_net(c.obs, c.act, c.hidden);
         cnet = make_critic_net(c.obs, c.act, c.hidden, c.atoms);
         tree_cap = next_pow2(c.capacity);
@@ -807,6 +1165,9 @@ public:
         pdh1 = carve<float>((long)B * H, off);
         pda = carve<float>((long)B * A, off);
         adz = carve<float>((long)B * A, off);
+        az1 = carve<float>((long)B * H, off);
+        az2 = carve<float>((long)B * H, off);
+        az3 = carve<float>((long)B * H, off);
         ing_s = carve<float>((long)ing_cap * O, off);
         ing_a = carve<float>((long)ing_cap * A, off);
         ing_r = carve<float>(ing_cap, off);
@@ -823,7 +1184,18 @@ public:
         layout();                            // assign pass
         Counters h{};
         h.max_priority = 1.0f;
+        // schedule counters pre-advanced: k_per_update increments at step
+        // END for the next step, so step 1 must already read Adam t=1.
+        // beta_t starts at 0 (reference LinearSchedule returns value_at(0)
+        // on its first stateful call, prioritized_replay_memory.py:25-29).
+        h.beta_t = 0;
+        h.adam_t_actor = h.adam_t_critic = h.rng_epoch = 1;
         HIP_CHECK(hipMemcpy(cnt, &h, sizeof(Counters), hipMemcpyHostToDevice));
+        // min-tree neutral element is +inf (memset zero would make every
+        // unoccupied leaf the minimum and zero out all IS weights)
+        hipLaunchKernelGGL(k_fill_f64, dim3(1024), dim3(256), 0, nullptr,
+                           min_tree, 2 * tree_cap, HUGE_VAL);
+        HIP_CHECK(hipDeviceSynchronize());
     }
 
     // ---------------- job builders ----------------
@@ -882,12 +1254,109 @@ public:
     }
 
     // ---------------- the train step (one launch sequence) ----------------
+    bool use_row_block() const {
+        // row-block wastes weight bandwidth at large batch (every wave
+        // re-streams every weight); the per-layer tiled path wins there.
+        return cfg.batch <= 512;
+    }
+
+    RowBlockArgs rb_args() {
+        RowBlockArgs g{};
+        g.B = cfg.batch; g.O = cfg.obs; g.A = cfg.act; g.H = cfg.hidden;
+        g.K = cfg.atoms;
+        g.v_min = cfg.v_min; g.v_max = cfg.v_max; g.gamma_n = cfg.gamma_n;
+        g.per_eps = cfg.per_eps; g.is_weighting = cfg.is_weighting;
+        g.p_actor = p_actor; g.p_actor_t = p_actor_t;
+        g.p_critic = p_critic; g.p_critic_t = p_critic_t;
+        for (int i = 0; i < 4; ++i) { g.al[i] = anet.l[i]; g.cl[i] = cnet.l[i]; }
+        g.bs = bs; g.ba = ba; g.br = br; g.bs2 = bs2; g.bd = bd; g.bw = bw;
+        g.a2 = a2; g.p_t = p_t; g.m_proj = m_proj; g.q = q; g.dlog = dlog;
+        g.pri = pri;
+        g.c_h1 = c_h1; g.c_h2 = c_h2; g.c_h3 = c_h3;
+        g.d1 = d1; g.d2 = d2; g.d3 = d3;
+        g.pa_h1 = pa_h1; g.pa_h2 = pa_h2; g.pa_h3 = pa_h3; g.a_out = a_out;
+        g.pc_h1 = pc_h1; g.pc_h2 = pc_h2; g.pc_h3 = pc_h3; g.pq = pq;
+        g.az1 = az1; g.az2 = az2; g.az3 = az3; g.adz = adz;
+        g.cnt = cnt;
+        return g;
+    }
+
+    // dW-only job for one layer (used by the batched k_bwd4 launch)
+    BwdJob dw_job(const float* dz, const float* x1, const float* x2,
+                  const float* slab, float* gslab, const LayerDesc& l,
+                  int& wg) {
+        BwdJob j{};
+        j.dz = dz; j.x1 = x1; j.x2 = x2;
+        j.wt = slab + l.w_off;
+        j.dwt = gslab + l.w_off; j.dbias = gslab + l.b_off;
+        j.B = cfg.batch; j.in1 = l.in1; j.in2 = l.in2; j.out = l.out;
+        j.wg0_dw = wg;
+        j.nwg_dw_i = ceil_div(l.in1 + l.in2, BWT);
+        j.nwg_dw_o = ceil_div(l.out, BWT);
+        wg += j.nwg_dw_i * j.nwg_dw_o;
+        j.wg0_dx = wg; j.nwg_dx_b = j.nwg_dx_i = 0;
+        return j;
+    }
+
+    void enqueue_step_rowblock() {
+        const int B = cfg.batch, H = cfg.hidden;
+        int row_wgs = ceil_div(B, 4);
+        long smem = rb_lds_bytes(H);
+
+        hipLaunchKernelGGL(k_per_sample, dim3(row_wgs), dim3(256), 0, stream,
+            sum_tree, min_tree, tree_cap, rs, ra, rr, rs2, rd,
+            cfg.obs, cfg.act, bs, ba, br, bs2, bd, bw, bidx, B, cnt,
+            cfg.per_beta0, (float)cfg.per_beta_iters, cfg.seed);
+
+        RowBlockArgs g = rb_args();
+        hipLaunchKernelGGL(k_critic_rowblock, dim3(row_wgs), dim3(256),
+                           smem, stream, g);
+        {
+            int wg = 0;
+            BwdJob j0 = dw_job(d1, bs, nullptr, p_critic, g_critic,
+                               cnet.l[0], wg);
+            BwdJob j1 = dw_job(d2, c_h1, ba, p_critic, g_critic,
+                               cnet.l[1], wg);
+            BwdJob j2 = dw_job(d3, c_h2, nullptr, p_critic, g_critic,
+                               cnet.l[2], wg);
+            BwdJob j3 = dw_job(dlog, c_h3, nullptr, p_critic, g_critic,
+                               cnet.l[3], wg);
+            hipLaunchKernelGGL(k_bwd4, dim3(wg), dim3(256), 0, stream,
+                               j0, j1, j2, j3);
+        }
+        hipLaunchKernelGGL(k_adam_lerp, dim3(256), dim3(256), 0, stream,
+                           p_critic, g_critic, m_critic, v_critic,
+                           p_critic_t, cnet.n_params, cfg.lr_critic,
+                           0.9f, 0.999f, 1e-8f, cfg.tau, cnt, 0);
+        hipLaunchKernelGGL(k_policy_rowblock, dim3(row_wgs), dim3(256),
+                           smem, stream, g);
+        {
+            int wg = 0;
+            BwdJob j0 = dw_job(az1, bs, nullptr, p_actor, g_actor,
+                               anet.l[0], wg);
+            BwdJob j1 = dw_job(az2, pa_h1, nullptr, p_actor, g_actor,
+                               anet.l[1], wg);
+            BwdJob j2 = dw_job(az3, pa_h2, nullptr, p_actor, g_actor,
+                               anet.l[2], wg);
+            BwdJob j3 = dw_job(adz, pa_h3, nullptr, p_actor, g_actor,
+                               anet.l[3], wg);
+            hipLaunchKernelGGL(k_bwd4, dim3(wg), dim3(256), 0, stream,
+                               j0, j1, j2, j3);
+        }
+        hipLaunchKernelGGL(k_adam_lerp, dim3(256), dim3(256), 0, stream,
+                           p_actor, g_actor, m_actor, v_actor,
+                           p_actor_t, anet.n_params, cfg.lr_actor,
+                           0.9f, 0.999f, 1e-8f, cfg.tau, cnt, 1);
+        hipLaunchKernelGGL(k_per_update, dim3(1), dim3(256), 0, stream,
+                           sum_tree, min_tree, tree_cap, bidx, pri, B,
+                           cfg.per_alpha, cnt);
+    }
+
     void enqueue_step() {
+        if (use_row_block()) { enqueue_step_rowblock(); return; }
         const int B = cfg.batch, K = cfg.atoms, H = cfg.hidden;
         const int waves_per_wg = 4;
         int row_wgs = ceil_div(B, waves_per_wg);
-
-        hipLaunchKernelGGL(k_tick, dim3(1), dim3(64), 0, stream, cnt);
 
         hipLaunchKernelGGL(k_per_sample, dim3(row_wgs), dim3(256), 0, stream,
             sum_tree, min_tree, tree_cap, rs, ra, rr, rs2, rd,
