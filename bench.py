@@ -1,0 +1,214 @@
+#!/usr/bin/env python3
+"""Driver benchmark contract.
+
+``python bench.py --gpus N --steps K --warmup W`` runs the flagship
+training step — the Pendulum-v1 D4PG learner (51-atom C51 critic,
+n-step=5, prioritized replay, batch 64) of BASELINE.json — on N GPUs of
+one node, one rank per GPU (launched via torch.distributed.run for N>1),
+on synthetic transitions and random-init weights.  Each rank owns a full
+fused HIP learner engine (weak scaling: per-GPU work fixed); the value
+reported is the WHOLE-JOB aggregate learner grad-steps/sec over all
+ranks, using the MAX elapsed time over ranks for EXACTLY K timed steps
+bracketed by barrier + torch.cuda.synchronize on both sides.
+
+Reference parity: the train step is the full D4PG update of
+/root/reference/ddpg.py:200-255 (PER sample + IS weights, target
+forwards, C51 categorical projection, cross-entropy critic loss +
+backward + Adam, policy loss + backward + Adam, target soft-update, PER
+priority writeback) — nothing is skipped inside the timed region.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import torch
+
+FLAGSHIP = dict(obs_dim=3, act_dim=1, hidden=256, n_atoms=51, batch=64,
+                capacity=1_000_000, v_min=-300.0, v_max=0.0,
+                gamma=0.99, n_steps=5, tau=0.001,
+                lr_actor=1e-4, lr_critic=1e-3)
+
+
+def _dist_init(args):
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local = int(os.environ.get("LOCAL_RANK", str(rank)))
+    if world > 1:
+        import torch.distributed as dist
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29511")
+        dist.init_process_group(backend=backend, rank=rank, world_size=world)
+        return dist, world, rank, local
+    return None, 1, 0, 0
+
+
+def _barrier(dist):
+    if dist is not None:
+        dist.barrier()
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+
+
+def _max_over_ranks(dist, x: float) -> float:
+    if dist is None:
+        return x
+    dev = "cuda" if torch.cuda.is_available() else "cpu"
+    t = torch.tensor([x], dtype=torch.float64, device=dev)
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    return float(t.item())
+
+
+def make_gpu_engine(seed: int):
+    from d4pg_amd.models import actor, critic
+    from d4pg_amd.ops import FusedEngine
+    cfg = FLAGSHIP
+    eng = FusedEngine(obs_dim=cfg["obs_dim"], act_dim=cfg["act_dim"],
+                      hidden=cfg["hidden"], n_atoms=cfg["n_atoms"],
+                      batch=cfg["batch"], capacity=cfg["capacity"],
+                      v_min=cfg["v_min"], v_max=cfg["v_max"],
+                      gamma_n=cfg["gamma"] ** cfg["n_steps"],
+                      tau=cfg["tau"], lr_actor=cfg["lr_actor"],
+                      lr_critic=cfg["lr_critic"], seed=seed)
+    torch.manual_seed(seed)
+    a = actor(cfg["obs_dim"], cfg["act_dim"], hidden=cfg["hidden"])
+    c = critic(cfg["obs_dim"], cfg["act_dim"],
+               {"type": "categorical", "v_min": cfg["v_min"],
+                "v_max": cfg["v_max"], "n_atoms": cfg["n_atoms"]},
+               hidden=cfg["hidden"])
+    eng.load_from_modules(a, a, c, c)
+    eng.synth_fill(cfg["capacity"], seed=seed + 7)
+    return eng
+
+
+def make_cpu_agent(seed: int):
+    """Eager CPU fallback so `python bench.py` runs in a GPU-less
+    container (numbers are then CPU numbers, not the MI355X headline)."""
+    import numpy as np
+    from d4pg_amd.algo.d4pg import DDPG
+    cfg = FLAGSHIP
+    agent = DDPG(cfg["obs_dim"], cfg["act_dim"],
+                 memory_size=100000, batch_size=cfg["batch"],
+                 gamma=cfg["gamma"], tau=cfg["tau"], prioritized_replay=True,
+                 critic_dist_info={"type": "categorical",
+                                   "v_min": cfg["v_min"],
+                                   "v_max": cfg["v_max"],
+                                   "n_atoms": cfg["n_atoms"]},
+                 n_steps=cfg["n_steps"], device="cpu", backend="eager",
+                 seed=seed)
+    rng = np.random.default_rng(seed)
+    for _ in range(5000):
+        agent.replayBuffer.add(
+            rng.standard_normal(cfg["obs_dim"]).astype("f"),
+            rng.uniform(-1, 1, cfg["act_dim"]).astype("f"),
+            -rng.random(), rng.standard_normal(cfg["obs_dim"]).astype("f"),
+            0.0)
+    return agent
+
+
+def measure_env_steps_per_sec(seed: int, n: int = 3000) -> float:
+    """Auxiliary (untimed-region) metric: actor-side env-steps/sec — native
+    Pendulum dynamics + B=1 actor inference + exploration noise, the per-env-
+    step path of /root/reference/main.py:142-152."""
+    import numpy as np
+    from d4pg_amd.envs import make
+    from d4pg_amd.models import actor
+    from d4pg_amd.noise import GaussianNoise
+    env = make("Pendulum-v1", seed=seed)
+    net = actor(3, 1)
+    net.eval()
+    noise = GaussianNoise(1, rng=np.random.default_rng(seed))
+    obs = env.reset()
+    with torch.no_grad():
+        t0 = time.perf_counter()
+        for _ in range(n):
+            a = net(torch.as_tensor(obs, dtype=torch.float32)[None])[0]
+            act = np.clip(a.numpy() + noise.sample(), -1.0, 1.0)
+            obs, r, done, _ = env.step(act)
+            if done:
+                obs = env.reset()
+        dt = time.perf_counter() - t0
+    return n / dt
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=4000)
+    ap.add_argument("--warmup", type=int, default=400)
+    ap.add_argument("--steps-per-graph", type=int, default=32)
+    args = ap.parse_args()
+
+    dist, world, rank, local = _dist_init(args)
+    n_gpus = max(world, args.gpus) if world > 1 else args.gpus
+    use_gpu = torch.cuda.is_available()
+    if use_gpu:
+        torch.cuda.set_device(local)
+
+    K, W = args.steps, args.warmup
+    if use_gpu:
+        eng = make_gpu_engine(seed=1000 + rank)
+        spg = args.steps_per_graph
+        eng.train_steps(max(W, spg), steps_per_graph=spg)  # warmup+capture
+        _barrier(dist)
+        t0 = time.perf_counter()
+        eng.train_steps(K, steps_per_graph=spg)
+        torch.cuda.synchronize()
+        elapsed = time.perf_counter() - t0
+        _barrier(dist)
+    else:
+        K = min(K, 400)
+        W = min(W, 20)
+        agent = make_cpu_agent(seed=1000 + rank)
+        for _ in range(W):
+            agent.train()
+        _barrier(dist)
+        t0 = time.perf_counter()
+        for _ in range(K):
+            agent.train()
+        elapsed = time.perf_counter() - t0
+        _barrier(dist)
+
+    elapsed = _max_over_ranks(dist, elapsed)
+    env_sps = measure_env_steps_per_sec(seed=1234 + rank) if rank == 0 else 0.0
+
+    if rank == 0:
+        out = {
+            "metric": "learner grad-steps/sec (Pendulum-v1 D4PG)",
+            "value": n_gpus * K / elapsed,
+            "unit": "grad_steps/s",
+            "n_gpus": n_gpus,
+            "steps": K,
+            "warmup": W,
+            "ms_per_step": elapsed / K * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic (random-init weights, synthetic replay "
+                    "transitions; reference publishes no numbers)",
+            "config": {"model": "D4PG Pendulum-v1 (obs3/act1, 4x256 MLP "
+                                "actor+critic, 51 atoms)",
+                       "global_batch": FLAGSHIP["batch"] * n_gpus,
+                       "seq_len": None,
+                       "parallelism": f"dp{n_gpus} (one learner per GPU)",
+                       "n_step": FLAGSHIP["n_steps"],
+                       "prioritized_replay": True,
+                       "replay_capacity": FLAGSHIP["capacity"],
+                       "device": "cuda" if use_gpu else "cpu-fallback"},
+            "env_steps_per_sec_1actor": env_sps,
+        }
+        print(json.dumps(out), flush=True)
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
