@@ -1572,6 +1572,12 @@ public:
     Counters read_counters() {
         Counters h{};
         HIP_CHECK(hipMemcpy(&h, cnt, sizeof(h), hipMemcpyDeviceToHost));
+        // device counters are pre-advanced for the NEXT step (init 1 so the
+        // first step sees t=1); report steps COMPLETED to the host.
+        h.adam_t_actor -= 1;
+        h.adam_t_critic -= 1;
+        h.beta_t -= 1;
+        h.rng_epoch -= 1;
         return h;
     }
 

@@ -1,0 +1,306 @@
+"""Distributed D4PG over torch.distributed (RCCL on ROCm, gloo on CPU).
+
+MI355X-native re-expression of the reference's shared-memory topology
+(/root/reference/main.py:382-405): instead of HogWild
+workers aliasing gradients into shared host memory, one process per GPU
+with an explicit role split —
+
+  rank 0            learner: owns params, Adam state, targets and the
+                    ENTIRE prioritized replay (on-HBM via the fused HIP
+                    engine when a GPU is present; eager-torch otherwise)
+  ranks 1..W-1      actor ranks: envs + exploration noise + n-step fold
+                    + HER relabel (reference main.py:137-184 semantics),
+                    pure inference clients
+  last rank         optional evaluator: greedy rollouts, EWMA return
+                    (reference main.py:103-134 semantics)
+
+Round-based schedule with a bounded staleness window (one round):
+
+  1. broadcast(actor params ++ global step) learner -> all.  The blob is
+     ~0.5 MB fp32 — latency-bound on xGMI, so one flat broadcast per
+     round, not per train step (reference pulled params EVERY step via
+     load_state_dict, ddpg.py:118-120 — pointless on a GPU fabric).
+  2. each actor rank collects `episodes_per_round` episodes.
+  3. fixed-size SoA all_gather (counts, then transition blocks)
+     actor -> learner.  all_gather (not gather) so the same code runs on
+     both RCCL and gloo.
+  4. learner ingests + runs `train_steps_per_cycle` grad steps while the
+     actors immediately start the next collection window (compute/env
+     overlap happens across the round boundary).
+
+The global step counter rides in the broadcast blob — replacing the
+reference's non-atomic shared `global_count += 1` (main.py:307) with a
+learner-owned monotonic counter.
+"""
+
+from __future__ import annotations
+
+import os
+import time
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+from ..algo.d4pg import DDPG
+from ..config import critic_dist_info, run_dir_name
+from ..envs import make, obs_act_dims
+from ..her import add_experience, rollout_episode
+from ..ops import pack_net, unpack_net
+from ..utils.logging import Meter, SummaryWriter
+
+
+class _ListBuffer:
+    """Replay-interface shim that records transitions into a list, so the
+    actor side can reuse add_experience (n-step fold + HER relabel) and
+    ship the result over the wire instead of into a local buffer."""
+
+    def __init__(self):
+        self.items = []
+
+    def add(self, state, action, reward, next_state, done):
+        self.items.append((np.asarray(state, np.float32).ravel(),
+                           np.asarray(action, np.float32).ravel(),
+                           np.float32(reward),
+                           np.asarray(next_state, np.float32).ravel(),
+                           np.float32(done)))
+
+    def __len__(self):
+        return len(self.items)
+
+
+def _row_width(obs_dim: int, act_dim: int) -> int:
+    return 2 * obs_dim + act_dim + 2
+
+
+def _encode(items, cap, obs_dim, act_dim, device):
+    buf = torch.zeros(cap, _row_width(obs_dim, act_dim), device=device)
+    n = min(len(items), cap)
+    if n:
+        o = obs_dim
+        a = act_dim
+        arr = np.zeros((n, _row_width(o, a)), np.float32)
+        for i, (s, ac, r, s2, d) in enumerate(items[:n]):
+            arr[i, :o] = s
+            arr[i, o:o + a] = ac
+            arr[i, o + a] = r
+            arr[i, o + a + 1:2 * o + a + 1] = s2
+            arr[i, 2 * o + a + 1] = d
+        buf[:n] = torch.from_numpy(arr).to(device)
+    return buf, n
+
+
+def _decode(buf, n, obs_dim, act_dim):
+    o, a = obs_dim, act_dim
+    arr = buf[:n].cpu().numpy()
+    return (arr[:, :o], arr[:, o:o + a], arr[:, o + a],
+            arr[:, o + a + 1:2 * o + a + 1], arr[:, 2 * o + a + 1])
+
+
+class DistributedD4PG:
+    """One rank of the distributed actor/learner topology.  Construct on
+    every rank with the same args, then call run(rounds)."""
+
+    def __init__(self, args, rank=None, world=None, device=None,
+                 with_evaluator=True, writer=None):
+        self.args = args
+        self.rank = int(os.environ.get("RANK", "0")) if rank is None else rank
+        self.world = (int(os.environ.get("WORLD_SIZE", "1"))
+                      if world is None else world)
+        if device is None:
+            if torch.cuda.is_available():
+                local = int(os.environ.get("LOCAL_RANK", str(self.rank)))
+                torch.cuda.set_device(local % torch.cuda.device_count())
+                device = "cuda"
+            else:
+                device = "cpu"
+        self.device = torch.device(device)
+        self.is_learner = self.rank == 0
+        self.eval_rank = (self.world - 1
+                          if (with_evaluator and self.world >= 3) else -1)
+        self.is_evaluator = self.rank == self.eval_rank
+
+        seed = (args.seed or 0) + 7919 * self.rank
+        self.env = make(args.env, seed=seed)
+        self.env._max_episode_steps = args.max_steps
+        self.obs_dim, self.act_dim = obs_act_dims(self.env,
+                                                  her=bool(args.her))
+        backend = "eager"
+        if self.is_learner and self.device.type == "cuda":
+            backend = "hip"
+        self.agent = DDPG(
+            self.obs_dim, self.act_dim, env=self.env,
+            memory_size=args.rmsize if self.is_learner else 1,
+            batch_size=args.bsize, gamma=args.gamma, tau=args.tau,
+            prioritized_replay=bool(args.p_replay) and self.is_learner,
+            critic_dist_info=critic_dist_info(args), n_steps=args.n_steps,
+            device=str(self.device) if self.is_learner else "cpu",
+            backend=backend, seed=seed)
+        self.rng = np.random.default_rng(seed)
+
+        # wire geometry: per-round per-rank transition cap.  HER can add up
+        # to 2x (real + hindsight) transitions per step.
+        self.episodes_per_round = max(
+            1, args.episodes_per_cycle // max(1, self.world - 1))
+        mult = 2 if args.her else 1
+        self.push_cap = self.episodes_per_round * args.max_steps * mult
+        self.blob_len = pack_net(self.agent.actor).numel() + 1
+
+        self.grad_meter = Meter()
+        self.env_meter = Meter()
+        self.global_step = 0
+        self.ewma = None
+        self.writer = writer
+        self.run_dir = run_dir_name(args)
+
+    # -- round phases -----------------------------------------------------
+
+    def _broadcast_params(self):
+        if self.is_learner:
+            if self.agent.backend == "hip":
+                self.agent.engine.sync_params_if_dirty()
+            blob = torch.cat([pack_net(self.agent.actor),
+                              torch.tensor([float(self.global_step)])])
+            blob = blob.to(self.device)
+        else:
+            blob = torch.zeros(self.blob_len, device=self.device)
+        dist.broadcast(blob, src=0)
+        if not self.is_learner:
+            blob = blob.cpu()
+            unpack_net(self.agent.actor, blob[:-1])
+            self.global_step = int(blob[-1].item())
+
+    def _collect(self):
+        lb = _ListBuffer()
+        if self.is_learner:
+            return lb
+        if self.is_evaluator:
+            _, R, _ = rollout_episode(self.agent, self.env, noise=False)
+            self.ewma = R if self.ewma is None else \
+                0.95 * self.ewma + 0.05 * R
+            print(f"[eval] step {self.global_step} return {R:.2f} "
+                  f"ewma {self.ewma:.2f}", flush=True)
+            return lb
+        for _ in range(self.episodes_per_round):
+            ep, _, _ = rollout_episode(self.agent, self.env, noise=True)
+            self.env_meter.add(len(ep))
+            add_experience(lb, self.env, ep, her=bool(self.args.her),
+                           n_steps=self.args.n_steps, gamma=self.args.gamma,
+                           rng=self.rng)
+        return lb
+
+    def _exchange(self, lb):
+        dev = self.device
+        cnt = torch.tensor([float(len(lb))], device=dev)
+        counts = [torch.zeros_like(cnt) for _ in range(self.world)]
+        dist.all_gather(counts, cnt)
+        buf, n = _encode(lb.items, self.push_cap, self.obs_dim,
+                         self.act_dim, dev)
+        blocks = [torch.zeros_like(buf) for _ in range(self.world)]
+        dist.all_gather(blocks, buf)
+        if not self.is_learner:
+            return 0
+        total = 0
+        for r in range(1, self.world):
+            n_r = int(counts[r].item())
+            if n_r == 0:
+                continue
+            s, a, rw, s2, d = _decode(blocks[r], n_r, self.obs_dim,
+                                      self.act_dim)
+            for i in range(n_r):
+                self.agent.replayBuffer.add(s[i], a[i], rw[i], s2[i], d[i])
+            total += n_r
+        return total
+
+    def _train(self):
+        if not self.is_learner:
+            return
+        floor = max(self.args.bsize,
+                    getattr(self.args, "warmup", 0) * self.args.max_steps)
+        if len(self.agent.replayBuffer) < floor:
+            return
+        n = self.args.train_steps_per_cycle
+        if self.agent.backend == "hip":
+            self.agent.engine.ddpg.replayBuffer.flush()
+            self.agent.engine.engine.train_steps(n)
+            self.agent.train_steps_done += n
+        else:
+            for _ in range(n):
+                self.agent.train()
+        self.grad_meter.add(n)
+        self.global_step += n
+
+    # -- main loop --------------------------------------------------------
+
+    def run(self, rounds: int, save: bool = False):
+        for rnd in range(rounds):
+            self._broadcast_params()
+            lb = self._collect()
+            ingested = self._exchange(lb)
+            self._train()
+            if self.is_learner:
+                if self.writer is not None:
+                    self.writer.add_scalar("grad_steps_per_sec",
+                                           self.grad_meter.rate(),
+                                           self.global_step)
+                    self.writer.add_scalar("replay_occupancy",
+                                           len(self.agent.replayBuffer),
+                                           self.global_step)
+                if self.args.debug:
+                    print(f"[learner] round {rnd} step {self.global_step} "
+                          f"ingested {ingested} "
+                          f"replay {len(self.agent.replayBuffer)}",
+                          flush=True)
+                if save and self.run_dir:
+                    os.makedirs(self.run_dir, exist_ok=True)
+                    self.agent.save(self.run_dir)
+        # final param sync so every rank ends with the trained policy
+        self._broadcast_params()
+        return self.global_step
+
+
+def init_process_group(backend: str | None = None):
+    if dist.is_initialized():
+        return
+    if backend is None:
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29521")
+    os.environ.setdefault("RANK", "0")
+    os.environ.setdefault("WORLD_SIZE", "1")
+    dist.init_process_group(backend=backend)
+
+
+def run_distributed(args, rounds: int | None = None, with_evaluator=True):
+    """Entry point for `torch.distributed.run ... -m d4pg_amd.parallel.learner`
+    or direct invocation from tests (with env RANK/WORLD_SIZE preset)."""
+    init_process_group()
+    rank = dist.get_rank()
+    writer = None
+    if rank == 0:
+        writer = SummaryWriter(run_dir_name(args))
+    node = DistributedD4PG(args, rank=rank, world=dist.get_world_size(),
+                           with_evaluator=with_evaluator, writer=writer)
+    t0 = time.perf_counter()
+    n_rounds = rounds if rounds is not None else args.n_eps * args.cycles_per_epoch
+    step = node.run(n_rounds, save=(rank == 0))
+    if rank == 0:
+        dt = time.perf_counter() - t0
+        print(f"[learner] done: {step} grad steps in {dt:.1f}s "
+              f"({step / max(dt, 1e-9):.1f} steps/s)", flush=True)
+    dist.barrier()
+    dist.destroy_process_group()
+    return step
+
+
+def main(argv=None):
+    from ..config import configure_env_params, make_parser
+    p = make_parser()
+    p.add_argument("--rounds", type=int, default=None)
+    args = p.parse_args(argv)
+    configure_env_params(args)
+    run_distributed(args, rounds=args.rounds)
+
+
+if __name__ == "__main__":
+    main()
