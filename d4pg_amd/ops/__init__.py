@@ -149,6 +149,7 @@ class FusedEngine:
             int(seed), bool(is_weighting))
         self.batch = batch
         self._captured = 0
+        self._persistent = bool(self.ext.info(self.h).get("persistent", False))
 
     def __del__(self):
         try:
@@ -193,7 +194,12 @@ class FusedEngine:
 
     def train_steps(self, n=1, steps_per_graph=8):
         """Graph-replayed steps: captures once (steps_per_graph per replay),
-        then replays; the remainder runs uncaptured."""
+        then replays; the remainder runs uncaptured.  On the persistent-
+        megakernel path this is a single multi-step launch instead (no graph
+        needed — the step loop lives inside the kernel)."""
+        if self._persistent:
+            self.ext.step(self.h, int(n))
+            return
         if self._captured != steps_per_graph:
             self.ext.capture(self.h, steps_per_graph)
             self._captured = steps_per_graph

@@ -1004,6 +1004,599 @@ __global__ void k_tree_build_level(double* sum_tree, double* min_tree,
 }
 
 // ===========================================================================
+// Persistent whole-step megakernel (flagship small-batch learner path)
+// ===========================================================================
+// Rationale (measured, profiles/step_kernel_stats round 1): the row-block
+// path spends 451 of 497 us/step in two kernels where each of B=64 waves
+// re-streams EVERY layer weight from L2 with only 16 workgroups in flight —
+// 64x redundant weight traffic and no latency hiding.  This kernel instead
+// runs the WHOLE train step (PER sample -> forwards -> C51 projection ->
+// CE/policy grads -> dX/dW backward -> Adam+soft-update -> PER write-back)
+// as ONE persistent kernel: PNWG=64 workgroups x 256 threads (guaranteed
+// co-resident on 256 CUs, so a software grid barrier is deadlock-free),
+// ~28 layer-parallel phases separated by grid barriers.  Each phase tiles
+// its GEMM over the grid (16-row x 64-col tiles, x rows staged in LDS,
+// weight columns read once per row-tile instead of once per row), so
+// weight traffic drops ~16x and every phase has 16-64 workgroups of
+// parallelism.  nsteps loop inside the kernel => zero host involvement
+// between steps (the hipGraph path is only needed for the wide-batch
+// configs that use the per-layer kernels above).
+#define PNWG 64
+#define PROWS 16
+#define PXMAX 512
+#define PLDS_FLOATS (PROWS * PXMAX + 256)
+
+struct PStepArgs {
+    int B, O, A, H, K;
+    float v_min, v_max, gamma_n, per_eps, tau, lr_actor, lr_critic;
+    float per_alpha, per_beta0, per_beta_iters;
+    int is_weighting;
+    uint64_t seed;
+    long tree_cap;
+    long n_actor, n_critic;
+    LayerDesc al[4], cl[4];
+    // params / grads / moments / targets
+    float *p_actor, *p_actor_t, *p_critic, *p_critic_t;
+    float *g_actor, *g_critic, *m_actor, *v_actor, *m_critic, *v_critic;
+    // replay + trees
+    const float *rs, *ra, *rr, *rs2, *rd;
+    double *sum_tree, *min_tree;
+    // batch
+    float *bs, *ba, *br, *bs2, *bd, *bw, *pri;
+    long *bidx;
+    // activations / deltas workspace
+    float *at_h1, *at_h2, *at_h3, *a2;
+    float *ct_h1, *ct_h2, *ct_h3, *p_t, *m_proj;
+    float *c_h1, *c_h2, *c_h3, *q, *dlog, *d3, *d2, *d1;
+    float *pa_h1, *pa_h2, *pa_h3, *a_out;
+    float *pc_h1, *pc_h2, *pc_h3, *pq;
+    float *pd3, *pd2, *pdh1, *adz, *az1, *az2, *az3;
+    Counters* cnt;
+    unsigned long long* gbar;        // [0] = arrival counter, [1] = base
+};
+
+// software grid barrier: monotonic arrival counter, per-thread local target.
+// Safe because all PNWG workgroups are co-resident (64 wgs of 256 threads /
+// 34 KB LDS on a 256-CU chip).  The base survives across launches in
+// gbar[1] (written by wg 0 after the final barrier; stream order makes the
+// next launch's read race-free).
+__device__ inline void p_bar(unsigned long long* ctr,
+                             unsigned long long& tgt) {
+    __syncthreads();
+    tgt += PNWG;
+    if (threadIdx.x == 0) {
+        __threadfence();                       // release: flush our writes
+        atomicAdd(ctr, 1ull);
+        volatile unsigned long long* vctr = ctr;
+        long spins = 0;
+        while (*vctr < tgt) {
+            __builtin_amdgcn_s_sleep(2);
+            // safety valve (~10 s): a barrier logic bug must never
+            // hard-hang the GPU — desynchronize and terminate instead
+            if (++spins > (1L << 28)) break;
+        }
+        __threadfence();                       // acquire: invalidate L1
+    }
+    __syncthreads();
+}
+
+// Tiled forward: 16-row x 64-col tiles; x rows staged in LDS (broadcast
+// reads), weight column read ONCE per tile (not once per row).  Thread
+// (rq = tid/64, c = tid%64) accumulates rows {r0+rq, +4, +8, +12} of column
+// c0+c — four independent FMA chains for ILP; softmax (out<=64) reduces
+// per-row across the wave.  Summation order over k is ascending, matching
+// the eager/rowblock paths bit-for-bit.
+// (__noinline__ is load-bearing: inlining 15 copies into the persistent
+// kernel segfaults ROCm 7.2's clang in ADCE; the call overhead is noise
+// next to the ~1 us phase time.)
+__device__ __noinline__ void p_fwd(float* lds, const float* x1, const float* x2,
+                             const float* wt, const float* bias, float* y,
+                             int B, int in1, int in2, int out, int act_kind,
+                             int wg_rel, int nwg) {
+    int in_total = in1 + in2;
+    int nrt = (B + PROWS - 1) / PROWS;
+    int nct = (out + 63) / 64;
+    int ntiles = nrt * nct;
+    int tid = threadIdx.x;
+    for (int t = wg_rel; t < ntiles; t += nwg) {
+        int r0 = (t / nct) * PROWS, c0 = (t % nct) * 64;
+        for (int e = tid; e < PROWS * in_total; e += 256) {
+            int rr_ = e / in_total, kk = e % in_total;
+            int gb = r0 + rr_;
+            float v = 0.f;
+            if (gb < B)
+                v = (kk < in1) ? x1[(long)gb * in1 + kk]
+                               : x2[(long)gb * in2 + (kk - in1)];
+            lds[rr_ * in_total + kk] = v;
+        }
+        __syncthreads();
+        int rq = tid >> 6, c = tid & 63;
+        int o = c0 + c;
+        float acc0 = 0.f, acc1 = 0.f, acc2 = 0.f, acc3 = 0.f;
+        if (o < out) {
+            const float* wcol = wt + o;
+            const float* xr0 = lds + (rq + 0) * in_total;
+            const float* xr1 = lds + (rq + 4) * in_total;
+            const float* xr2 = lds + (rq + 8) * in_total;
+            const float* xr3 = lds + (rq + 12) * in_total;
+#pragma unroll 8
+            for (int k = 0; k < in_total; ++k) {
+                float wv = wcol[(long)k * out];
+                acc0 += xr0[k] * wv; acc1 += xr1[k] * wv;
+                acc2 += xr2[k] * wv; acc3 += xr3[k] * wv;
+            }
+            float bv = bias[o];
+            acc0 += bv; acc1 += bv; acc2 += bv; acc3 += bv;
+        }
+        float accs[4] = {acc0, acc1, acc2, acc3};
+        if (act_kind == ACT_SOFTMAX) {
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                int r = r0 + rq + 4 * j;
+                float v = (o < out) ? accs[j] : -INFINITY;
+                float mx = v;
+                for (int s = 32; s > 0; s >>= 1)
+                    mx = fmaxf(mx, __shfl_xor(mx, s, 64));
+                float e = (o < out) ? __expf(v - mx) : 0.f;
+                float sum = e;
+                for (int s = 32; s > 0; s >>= 1)
+                    sum += __shfl_xor(sum, s, 64);
+                if (r < B && o < out) y[(long)r * out + o] = e / sum;
+            }
+        } else if (o < out) {
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                int r = r0 + rq + 4 * j;
+                float v = accs[j];
+                if (act_kind == ACT_RELU) v = fmaxf(v, 0.f);
+                else if (act_kind == ACT_TANH) v = tanhf(v);
+                if (r < B) y[(long)r * out + o] = v;
+            }
+        }
+        __syncthreads();
+    }
+}
+
+// Tiled backward-dX: dx[b][i-in_lo] = (sum_o dz[b][o] wt[i][o]) * mask.
+// dz rows staged in LDS; mask/dx arrays are [B][span] (span = in_hi-in_lo):
+// full-range h arrays and the concat action slice both fit this form.
+__device__ inline void p_bwd_dx(float* lds, const float* dz, const float* wt,
+                                int in_lo, int in_hi, int out, int B,
+                                const float* hprev, int prev_act, float* dx,
+                                int wg_rel, int nwg) {
+    int span = in_hi - in_lo;
+    int nrt = (B + PROWS - 1) / PROWS;
+    int nit = (span + 63) / 64;
+    int ntiles = nrt * nit;
+    int tid = threadIdx.x;
+    for (int t = wg_rel; t < ntiles; t += nwg) {
+        int r0 = (t / nit) * PROWS, i0 = in_lo + (t % nit) * 64;
+        for (int e = tid; e < PROWS * out; e += 256) {
+            int rr_ = e / out, oo = e % out;
+            int gb = r0 + rr_;
+            lds[rr_ * out + oo] = (gb < B) ? dz[(long)gb * out + oo] : 0.f;
+        }
+        __syncthreads();
+        int rq = tid >> 6, c = tid & 63;
+        int i = i0 + c;
+        float acc0 = 0.f, acc1 = 0.f, acc2 = 0.f, acc3 = 0.f;
+        if (i < in_hi) {
+            const float* wrow = wt + (long)i * out;
+            const float* z0 = lds + (rq + 0) * out;
+            const float* z1 = lds + (rq + 4) * out;
+            const float* z2 = lds + (rq + 8) * out;
+            const float* z3 = lds + (rq + 12) * out;
+#pragma unroll 8
+            for (int o = 0; o < out; ++o) {
+                float wv = wrow[o];
+                acc0 += z0[o] * wv; acc1 += z1[o] * wv;
+                acc2 += z2[o] * wv; acc3 += z3[o] * wv;
+            }
+        }
+        if (i < in_hi) {
+            float accs[4] = {acc0, acc1, acc2, acc3};
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                int r = r0 + rq + 4 * j;
+                if (r < B) {
+                    long rel = (long)r * span + (i - in_lo);
+                    float m_ = hprev ? act_mask(prev_act, hprev[rel]) : 1.f;
+                    dx[rel] = accs[j] * m_;
+                }
+            }
+        }
+        __syncthreads();
+    }
+}
+
+// Tiled dW (+db on the i0==0 tile row): 16x16 tiles, B-chunked LDS staging,
+// identical math/order to bwd_one's dW part.
+__device__ inline void p_dw(float* lds, const float* dz, const float* x1,
+                            const float* x2, float* dwt, float* dbias,
+                            int B, int in1, int in2, int out,
+                            int wg_rel, int nwg) {
+    int in_total = in1 + in2;
+    int nti = (in_total + BWT - 1) / BWT, nto = (out + BWT - 1) / BWT;
+    int ntiles = nti * nto;
+    float* xs = lds;                       // [64][BWT+1]
+    float* zs = lds + 64 * (BWT + 1);      // [64][BWT+1]
+    int tid = threadIdx.x;
+    int ti = tid / BWT, to = tid % BWT;
+    for (int t = wg_rel; t < ntiles; t += nwg) {
+        int i0 = (t / nto) * BWT, o0 = (t % nto) * BWT;
+        float acc = 0.f, accb = 0.f;
+        for (int bc = 0; bc < B; bc += 64) {
+            for (int e = tid; e < 64 * BWT; e += 256) {
+                int bb = e / BWT, ii = e % BWT;
+                int gb = bc + bb, gi = i0 + ii;
+                float xv = 0.f;
+                if (gb < B && gi < in_total)
+                    xv = (gi < in1) ? x1[(long)gb * in1 + gi]
+                                    : x2[(long)gb * in2 + (gi - in1)];
+                xs[bb * (BWT + 1) + ii] = xv;
+                int go = o0 + ii;
+                zs[bb * (BWT + 1) + ii] =
+                    (gb < B && go < out) ? dz[(long)gb * out + go] : 0.f;
+            }
+            __syncthreads();
+#pragma unroll 16
+            for (int bb = 0; bb < 64; ++bb) {
+                acc += xs[bb * (BWT + 1) + ti] * zs[bb * (BWT + 1) + to];
+                if (ti == 0) accb += zs[bb * (BWT + 1) + to];
+            }
+            __syncthreads();
+        }
+        int gi = i0 + ti, go = o0 + to;
+        if (gi < in_total && go < out) dwt[(long)gi * out + go] = acc;
+        if (ti == 0 && go < out && i0 == 0 && dbias) dbias[go] = accb;
+    }
+}
+
+__device__ inline void p_adam_lerp(float* p, const float* gr, float* m,
+                                   float* v, float* tgt_slab, long n,
+                                   float lr, float tau, long long t) {
+    const float b1 = 0.9f, b2 = 0.999f, eps = 1e-8f;
+    float bc1 = 1.f - __powf(b1, (float)t);
+    float bc2 = 1.f - __powf(b2, (float)t);
+    for (long i = (long)blockIdx.x * 256 + threadIdx.x; i < n;
+         i += (long)PNWG * 256) {
+        float gi = gr[i];
+        float mi = b1 * m[i] + (1.f - b1) * gi;
+        float vi = b2 * v[i] + (1.f - b2) * gi * gi;
+        m[i] = mi; v[i] = vi;
+        float pn = p[i] - lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+        p[i] = pn;
+        tgt_slab[i] += tau * (pn - tgt_slab[i]);
+    }
+}
+
+// PER sample + gather phase (k_per_sample semantics; one wave per probe)
+__device__ inline void p_sample(const PStepArgs& g) {
+    int probe = blockIdx.x * 4 + (threadIdx.x >> 6);
+    int lane = threadIdx.x & 63;
+    if (probe == 0 && lane == 0) {
+        g.cnt->loss_critic = 0.f;
+        g.cnt->loss_actor = 0.f;
+    }
+    if (probe >= g.B) return;
+    long long n = g.cnt->size;
+    double total = g.sum_tree[1];
+    float frac = fminf((float)((double)g.cnt->beta_t / g.per_beta_iters),
+                       1.0f);
+    float beta = g.per_beta0 + frac * (1.0f - g.per_beta0);
+    long idx;
+    if (lane == 0) {
+        Philox4 r = philox4(g.seed, (uint64_t)g.cnt->rng_epoch,
+                            (uint64_t)probe);
+        double mass = (double)u01(r.v[0]) * total;
+        long node = 1;
+        while (node < g.tree_cap) {
+            double ls = g.sum_tree[2 * node];
+            if (mass > ls) { mass -= ls; node = 2 * node + 1; }
+            else           { node = 2 * node; }
+        }
+        idx = node - g.tree_cap;
+        if (idx >= n) idx = n - 1;
+        g.bidx[probe] = idx;
+        double p = g.sum_tree[g.tree_cap + idx] / total;
+        double p_min = g.min_tree[1] / total;
+        double max_w = pow(p_min * (double)n, (double)-beta);
+        g.bw[probe] = (float)(pow(p * (double)n, (double)-beta) / max_w);
+        g.br[probe] = g.rr[idx];
+        g.bd[probe] = g.rd[idx];
+    }
+    idx = __shfl(idx, 0, 64);
+    for (int k = lane; k < g.O; k += 64) {
+        g.bs[(long)probe * g.O + k] = g.rs[idx * g.O + k];
+        g.bs2[(long)probe * g.O + k] = g.rs2[idx * g.O + k];
+    }
+    for (int k = lane; k < g.A; k += 64)
+        g.ba[(long)probe * g.A + k] = g.ra[idx * g.A + k];
+}
+
+// C51 projection phase (k_project semantics; one wave per row, LDS row)
+__device__ inline void p_project(const PStepArgs& g, float* lds) {
+    int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    int row = blockIdx.x * 4 + wid;
+    int K = g.K;
+    float* mrow = lds + wid * 64;
+    for (int k = lane; k < K; k += 64) mrow[k] = 0.f;
+    __builtin_amdgcn_wave_barrier();
+    if (row < g.B && lane < K) {
+        float delta = (g.v_max - g.v_min) / (K - 1);
+        float z = g.v_min + lane * delta;
+        float tz = g.br[row] + g.gamma_n * (1.f - g.bd[row]) * z;
+        tz = fminf(g.v_max, fmaxf(g.v_min, tz));
+        float b = (tz - g.v_min) / delta;
+        int l = (int)floorf(b), u = (int)ceilf(b);
+        if (l == u) { if (u > 0) l -= 1; else u += 1; }
+        float p = g.p_t[(long)row * K + lane];
+        atomicAdd(&mrow[l], p * ((float)u - b));
+        atomicAdd(&mrow[u], p * (b - (float)l));
+    }
+    __builtin_amdgcn_wave_barrier();
+    if (row < g.B)
+        for (int k = lane; k < K; k += 64)
+            g.m_proj[(long)row * K + k] = mrow[k];
+}
+
+__device__ inline void p_ce_grad(const PStepArgs& g) {
+    int row = blockIdx.x * 4 + (threadIdx.x >> 6);
+    int lane = threadIdx.x & 63;
+    if (row >= g.B) return;
+    int K = g.K;
+    float qv = 0.f, mv = 0.f;
+    if (lane < K) {
+        qv = g.q[(long)row * K + lane];
+        mv = g.m_proj[(long)row * K + lane];
+    }
+    float dot = mv * qv;
+    float ce = -mv * __logf(qv + 1e-10f);
+    for (int s = 32; s > 0; s >>= 1) {
+        dot += __shfl_xor(dot, s, 64);
+        ce += __shfl_xor(ce, s, 64);
+    }
+    float scale = (g.is_weighting && g.bw) ? g.bw[row] : 1.f;
+    if (lane < K)
+        g.dlog[(long)row * K + lane] = scale * (qv - mv) / (float)g.B;
+    if (lane == 0) {
+        g.pri[row] = dot + g.per_eps;
+        atomicAdd(&g.cnt->loss_critic, scale * ce / (float)g.B);
+    }
+}
+
+__device__ inline void p_policy_grad(const PStepArgs& g) {
+    int row = blockIdx.x * 4 + (threadIdx.x >> 6);
+    int lane = threadIdx.x & 63;
+    if (row >= g.B) return;
+    int K = g.K;
+    float delta = (g.v_max - g.v_min) / (K - 1);
+    float z = g.v_min + lane * delta;
+    float qv = (lane < K) ? g.pq[(long)row * K + lane] : 0.f;
+    float e = qv * z;
+    for (int s = 32; s > 0; s >>= 1) e += __shfl_xor(e, s, 64);
+    if (lane < K)
+        g.pd3[(long)row * K + lane] = -qv * (z - e) / (float)g.B;
+    if (lane == 0) atomicAdd(&g.cnt->loss_actor, -e / (float)g.B);
+}
+
+// PER write-back + schedule-counter tick (k_per_update semantics; wg 0 only)
+__device__ inline void p_per_update(const PStepArgs& g) {
+    if (blockIdx.x != 0) return;
+    int tid = threadIdx.x;
+    float local_max = 0.f;
+    for (int i = tid; i < g.B; i += 256) {
+        float p = g.pri[i];
+        double pa = pow((double)p, (double)g.per_alpha);
+        long leaf = g.tree_cap + g.bidx[i];
+        g.sum_tree[leaf] = pa;
+        g.min_tree[leaf] = pa;
+        local_max = fmaxf(local_max, p);
+    }
+    __shared__ float smax[256];
+    smax[tid] = local_max;
+    __syncthreads();
+    for (int s = 128; s > 0; s >>= 1) {
+        if (tid < s) smax[tid] = fmaxf(smax[tid], smax[tid + s]);
+        __syncthreads();
+    }
+    if (tid == 0)
+        g.cnt->max_priority = fmaxf(g.cnt->max_priority, smax[0]);
+    __syncthreads();
+    long levels = 0;
+    for (long c = g.tree_cap; c > 1; c >>= 1) ++levels;
+    for (long lv = 0; lv < levels; ++lv) {
+        for (int i = tid; i < g.B; i += 256) {
+            long node = (g.tree_cap + g.bidx[i]) >> (lv + 1);
+            if (node >= 1) {
+                g.sum_tree[node] =
+                    g.sum_tree[2 * node] + g.sum_tree[2 * node + 1];
+                g.min_tree[node] = fmin(g.min_tree[2 * node],
+                                        g.min_tree[2 * node + 1]);
+            }
+        }
+        __syncthreads();
+    }
+    if (tid == 0) {
+        g.cnt->beta_t += 1;
+        g.cnt->adam_t_actor += 1;
+        g.cnt->adam_t_critic += 1;
+        g.cnt->rng_epoch += 1;
+    }
+}
+
+__global__ void __launch_bounds__(256, 1)
+k_step_persistent(PStepArgs g, int nsteps) {
+    __shared__ float lds[PLDS_FLOATS];
+    unsigned long long tgt = g.gbar[1];
+    unsigned long long* ctr = g.gbar;
+    int wg = blockIdx.x;
+    NetPtrs a = net_ptrs(g.p_actor, g.al);
+    NetPtrs at = net_ptrs(g.p_actor_t, g.al);
+    NetPtrs c = net_ptrs(g.p_critic, g.cl);
+    NetPtrs ct = net_ptrs(g.p_critic_t, g.cl);
+    const int O = g.O, A = g.A, H = g.H, K = g.K, B = g.B;
+
+    for (int s = 0; s < nsteps; ++s) {
+        // PH0: PER sample + batch gather
+        p_sample(g);
+        p_bar(ctr, tgt);
+        // PH1: four independent L1s (16 wgs each)
+        if (wg < 16)
+            p_fwd(lds, g.bs2, nullptr, at.w1, at.b1, g.at_h1, B, O, 0, H,
+                  ACT_RELU, wg, 16);
+        else if (wg < 32)
+            p_fwd(lds, g.bs2, nullptr, ct.w1, ct.b1, g.ct_h1, B, O, 0, H,
+                  ACT_RELU, wg - 16, 16);
+        else if (wg < 48)
+            p_fwd(lds, g.bs, nullptr, c.w1, c.b1, g.c_h1, B, O, 0, H,
+                  ACT_RELU, wg - 32, 16);
+        else
+            p_fwd(lds, g.bs, nullptr, a.w1, a.b1, g.pa_h1, B, O, 0, H,
+                  ACT_RELU, wg - 48, 16);
+        p_bar(ctr, tgt);
+        // PH2: actor_t.L2 | critic.L2(cat h1, a) | actor.L2
+        if (wg < 21)
+            p_fwd(lds, g.at_h1, nullptr, at.w2, at.b2, g.at_h2, B, H, 0, H,
+                  ACT_NONE, wg, 21);
+        else if (wg < 42)
+            p_fwd(lds, g.c_h1, g.ba, c.w2, c.b2, g.c_h2, B, H, A, H,
+                  ACT_RELU, wg - 21, 21);
+        else
+            p_fwd(lds, g.pa_h1, nullptr, a.w2, a.b2, g.pa_h2, B, H, 0, H,
+                  ACT_NONE, wg - 42, 22);
+        p_bar(ctr, tgt);
+        // PH3: L3s
+        if (wg < 21)
+            p_fwd(lds, g.at_h2, nullptr, at.w3, at.b3, g.at_h3, B, H, 0, H,
+                  ACT_RELU, wg, 21);
+        else if (wg < 42)
+            p_fwd(lds, g.c_h2, nullptr, c.w3, c.b3, g.c_h3, B, H, 0, H,
+                  ACT_RELU, wg - 21, 21);
+        else
+            p_fwd(lds, g.pa_h2, nullptr, a.w3, a.b3, g.pa_h3, B, H, 0, H,
+                  ACT_RELU, wg - 42, 22);
+        p_bar(ctr, tgt);
+        // PH4: heads — actor_t tanh -> a2 | critic softmax -> q |
+        //      actor tanh -> a_out
+        if (wg < 21)
+            p_fwd(lds, g.at_h3, nullptr, at.w4, at.b4, g.a2, B, H, 0, A,
+                  ACT_TANH, wg, 21);
+        else if (wg < 42)
+            p_fwd(lds, g.c_h3, nullptr, c.w4, c.b4, g.q, B, H, 0, K,
+                  ACT_SOFTMAX, wg - 21, 21);
+        else
+            p_fwd(lds, g.pa_h3, nullptr, a.w4, a.b4, g.a_out, B, H, 0, A,
+                  ACT_TANH, wg - 42, 22);
+        p_bar(ctr, tgt);
+        // PH5-7: critic_target chain on (s2, a2)
+        p_fwd(lds, g.ct_h1, g.a2, ct.w2, ct.b2, g.ct_h2, B, H, A, H,
+              ACT_RELU, wg, PNWG);
+        p_bar(ctr, tgt);
+        p_fwd(lds, g.ct_h2, nullptr, ct.w3, ct.b3, g.ct_h3, B, H, 0, H,
+              ACT_RELU, wg, PNWG);
+        p_bar(ctr, tgt);
+        p_fwd(lds, g.ct_h3, nullptr, ct.w4, ct.b4, g.p_t, B, H, 0, K,
+              ACT_SOFTMAX, wg, PNWG);
+        p_bar(ctr, tgt);
+        // PH8: C51 projection; PH9: CE grad + priorities
+        p_project(g, lds);
+        p_bar(ctr, tgt);
+        p_ce_grad(g);
+        p_bar(ctr, tgt);
+        // PH10-12: critic dX chain (pre-update weights)
+        p_bwd_dx(lds, g.dlog, c.w4, 0, H, K, B, g.c_h3, ACT_RELU, g.d3,
+                 wg, PNWG);
+        p_bar(ctr, tgt);
+        p_bwd_dx(lds, g.d3, c.w3, 0, H, H, B, g.c_h2, ACT_RELU, g.d2,
+                 wg, PNWG);
+        p_bar(ctr, tgt);
+        p_bwd_dx(lds, g.d2, c.w2, 0, H, H, B, g.c_h1, ACT_RELU, g.d1,
+                 wg, PNWG);
+        p_bar(ctr, tgt);
+        // PH13: critic dW, 4 jobs split by tile count (l1 is the biggest)
+        if (wg < 4)
+            p_dw(lds, g.d1, g.bs, nullptr, g.g_critic + g.cl[0].w_off,
+                 g.g_critic + g.cl[0].b_off, B, O, 0, H, wg, 4);
+        else if (wg < 32)
+            p_dw(lds, g.d2, g.c_h1, g.ba, g.g_critic + g.cl[1].w_off,
+                 g.g_critic + g.cl[1].b_off, B, H, A, H, wg - 4, 28);
+        else if (wg < 56)
+            p_dw(lds, g.d3, g.c_h2, nullptr, g.g_critic + g.cl[2].w_off,
+                 g.g_critic + g.cl[2].b_off, B, H, 0, H, wg - 32, 24);
+        else
+            p_dw(lds, g.dlog, g.c_h3, nullptr, g.g_critic + g.cl[3].w_off,
+                 g.g_critic + g.cl[3].b_off, B, H, 0, K, wg - 56, 8);
+        p_bar(ctr, tgt);
+        // PH14: Adam + target soft-update, critic
+        p_adam_lerp(g.p_critic, g.g_critic, g.m_critic, g.v_critic,
+                    g.p_critic_t, g.n_critic, g.lr_critic, g.tau,
+                    g.cnt->adam_t_critic);
+        p_bar(ctr, tgt);
+        // PH15-18: critic(s, a_out) with UPDATED critic params
+        p_fwd(lds, g.bs, nullptr, c.w1, c.b1, g.pc_h1, B, O, 0, H,
+              ACT_RELU, wg, PNWG);
+        p_bar(ctr, tgt);
+        p_fwd(lds, g.pc_h1, g.a_out, c.w2, c.b2, g.pc_h2, B, H, A, H,
+              ACT_RELU, wg, PNWG);
+        p_bar(ctr, tgt);
+        p_fwd(lds, g.pc_h2, nullptr, c.w3, c.b3, g.pc_h3, B, H, 0, H,
+              ACT_RELU, wg, PNWG);
+        p_bar(ctr, tgt);
+        p_fwd(lds, g.pc_h3, nullptr, c.w4, c.b4, g.pq, B, H, 0, K,
+              ACT_SOFTMAX, wg, PNWG);
+        p_bar(ctr, tgt);
+        // PH19: policy head gradient
+        p_policy_grad(g);
+        p_bar(ctr, tgt);
+        // PH20-22: dX back through the critic, ending at the tanh-masked
+        // action slice (adz)
+        p_bwd_dx(lds, g.pd3, c.w4, 0, H, K, B, g.pc_h3, ACT_RELU, g.pd2,
+                 wg, PNWG);
+        p_bar(ctr, tgt);
+        p_bwd_dx(lds, g.pd2, c.w3, 0, H, H, B, g.pc_h2, ACT_RELU, g.pdh1,
+                 wg, PNWG);
+        p_bar(ctr, tgt);
+        p_bwd_dx(lds, g.pdh1, c.w2, H, H + A, H, B, g.a_out, ACT_TANH,
+                 g.adz, wg, PNWG);
+        p_bar(ctr, tgt);
+        // PH23-25: dX through the actor (az3, az2, az1)
+        p_bwd_dx(lds, g.adz, a.w4, 0, H, A, B, g.pa_h3, ACT_RELU, g.az3,
+                 wg, PNWG);
+        p_bar(ctr, tgt);
+        p_bwd_dx(lds, g.az3, a.w3, 0, H, H, B, nullptr, ACT_NONE, g.az2,
+                 wg, PNWG);
+        p_bar(ctr, tgt);
+        p_bwd_dx(lds, g.az2, a.w2, 0, H, H, B, g.pa_h1, ACT_RELU, g.az1,
+                 wg, PNWG);
+        p_bar(ctr, tgt);
+        // PH26: actor dW
+        if (wg < 4)
+            p_dw(lds, g.az1, g.bs, nullptr, g.g_actor + g.al[0].w_off,
+                 g.g_actor + g.al[0].b_off, B, O, 0, H, wg, 4);
+        else if (wg < 30)
+            p_dw(lds, g.az2, g.pa_h1, nullptr, g.g_actor + g.al[1].w_off,
+                 g.g_actor + g.al[1].b_off, B, H, 0, H, wg - 4, 26);
+        else if (wg < 56)
+            p_dw(lds, g.az3, g.pa_h2, nullptr, g.g_actor + g.al[2].w_off,
+                 g.g_actor + g.al[2].b_off, B, H, 0, H, wg - 30, 26);
+        else
+            p_dw(lds, g.adz, g.pa_h3, nullptr, g.g_actor + g.al[3].w_off,
+                 g.g_actor + g.al[3].b_off, B, H, 0, A, wg - 56, 8);
+        p_bar(ctr, tgt);
+        // PH27: Adam + soft-update, actor
+        p_adam_lerp(g.p_actor, g.g_actor, g.m_actor, g.v_actor,
+                    g.p_actor_t, g.n_actor, g.lr_actor, g.tau,
+                    g.cnt->adam_t_actor);
+        p_bar(ctr, tgt);
+        // PH28: PER priority write-back + counter tick
+        p_per_update(g);
+        p_bar(ctr, tgt);
+    }
+    if (wg == 0 && threadIdx.x == 0) g.gbar[1] = tgt;
+}
+
+// ===========================================================================
 // Host engine
 // ===========================================================================
 
@@ -1077,6 +1670,7 @@ public:
     float *az1, *az2, *az3;                         // actor per-layer dz rows
     float *ing_s, *ing_a, *ing_r, *ing_s2, *ing_d;  // ingestion staging
     int ing_cap;
+    unsigned long long* gbar;                       // persistent grid barrier
 
     Engine(const EngineCfg& c) : cfg(c) {
         if (c.hidden + c.act > FWD_XMAX || c.obs > FWD_XMAX)
@@ -1168,6 +1762,7 @@ public:
         az1 = carve<float>((long)B * H, off);
         az2 = carve<float>((long)B * H, off);
         az3 = carve<float>((long)B * H, off);
+        gbar = carve<unsigned long long>(2, off);
         ing_s = carve<float>((long)ing_cap * O, off);
         ing_a = carve<float>((long)ing_cap * A, off);
         ing_r = carve<float>(ing_cap, off);
@@ -1254,6 +1849,55 @@ public:
     }
 
     // ---------------- the train step (one launch sequence) ----------------
+    bool use_persistent() const {
+        // the persistent megakernel assumes: probe/row waves cover the batch
+        // (B <= PNWG*4), fan-ins fit the LDS x-stage, softmax <= one wave
+        return cfg.batch <= PNWG * 4 &&
+               cfg.obs <= PXMAX && cfg.hidden + cfg.act <= PXMAX &&
+               cfg.hidden <= PXMAX && cfg.atoms <= 64;
+    }
+
+    PStepArgs ps_args() {
+        PStepArgs g{};
+        g.B = cfg.batch; g.O = cfg.obs; g.A = cfg.act; g.H = cfg.hidden;
+        g.K = cfg.atoms;
+        g.v_min = cfg.v_min; g.v_max = cfg.v_max; g.gamma_n = cfg.gamma_n;
+        g.per_eps = cfg.per_eps; g.tau = cfg.tau;
+        g.lr_actor = cfg.lr_actor; g.lr_critic = cfg.lr_critic;
+        g.per_alpha = cfg.per_alpha; g.per_beta0 = cfg.per_beta0;
+        g.per_beta_iters = (float)cfg.per_beta_iters;
+        g.is_weighting = cfg.is_weighting;
+        g.seed = cfg.seed; g.tree_cap = tree_cap;
+        g.n_actor = anet.n_params; g.n_critic = cnet.n_params;
+        for (int i = 0; i < 4; ++i) { g.al[i] = anet.l[i]; g.cl[i] = cnet.l[i]; }
+        g.p_actor = p_actor; g.p_actor_t = p_actor_t;
+        g.p_critic = p_critic; g.p_critic_t = p_critic_t;
+        g.g_actor = g_actor; g.g_critic = g_critic;
+        g.m_actor = m_actor; g.v_actor = v_actor;
+        g.m_critic = m_critic; g.v_critic = v_critic;
+        g.rs = rs; g.ra = ra; g.rr = rr; g.rs2 = rs2; g.rd = rd;
+        g.sum_tree = sum_tree; g.min_tree = min_tree;
+        g.bs = bs; g.ba = ba; g.br = br; g.bs2 = bs2; g.bd = bd; g.bw = bw;
+        g.pri = pri; g.bidx = bidx;
+        g.at_h1 = at_h1; g.at_h2 = at_h2; g.at_h3 = at_h3; g.a2 = a2;
+        g.ct_h1 = ct_h1; g.ct_h2 = ct_h2; g.ct_h3 = ct_h3; g.p_t = p_t;
+        g.m_proj = m_proj;
+        g.c_h1 = c_h1; g.c_h2 = c_h2; g.c_h3 = c_h3; g.q = q;
+        g.dlog = dlog; g.d3 = d3; g.d2 = d2; g.d1 = d1;
+        g.pa_h1 = pa_h1; g.pa_h2 = pa_h2; g.pa_h3 = pa_h3; g.a_out = a_out;
+        g.pc_h1 = pc_h1; g.pc_h2 = pc_h2; g.pc_h3 = pc_h3; g.pq = pq;
+        g.pd3 = pd3; g.pd2 = pd2; g.pdh1 = pdh1; g.adz = adz;
+        g.az1 = az1; g.az2 = az2; g.az3 = az3;
+        g.cnt = cnt; g.gbar = gbar;
+        return g;
+    }
+
+    void enqueue_persistent(int nsteps) {
+        PStepArgs g = ps_args();
+        hipLaunchKernelGGL(k_step_persistent, dim3(PNWG), dim3(256), 0,
+                           stream, g, nsteps);
+    }
+
     bool use_row_block() const {
         // row-block wastes weight bandwidth at large batch (every wave
         // re-streams every weight); the per-layer tiled path wins there.
@@ -1353,6 +1997,7 @@ public:
     }
 
     void enqueue_step() {
+        if (use_persistent()) { enqueue_persistent(1); return; }
         if (use_row_block()) { enqueue_step_rowblock(); return; }
         const int B = cfg.batch, K = cfg.atoms, H = cfg.hidden;
         const int waves_per_wg = 4;
@@ -1493,7 +2138,13 @@ public:
     }
 
     void step(int n) {
-        for (int i = 0; i < n; ++i) enqueue_step();
+        if (use_persistent() && n > 0) {
+            // whole multi-step run in ONE launch — zero host dispatch
+            // between steps (numerically identical to n single launches)
+            enqueue_persistent(n);
+        } else {
+            for (int i = 0; i < n; ++i) enqueue_step();
+        }
         HIP_CHECK(hipStreamSynchronize(stream));
     }
 

@@ -49,6 +49,7 @@ static py::dict info(int64_t h) {
     d["n_params_actor"] = e.anet.n_params;
     d["n_params_critic"] = e.cnet.n_params;
     d["tree_cap"] = e.tree_cap;
+    d["persistent"] = e.use_persistent();
     py::list al, cl;
     for (int i = 0; i < 4; ++i) {
         py::dict ld;
