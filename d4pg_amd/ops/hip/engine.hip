@@ -1004,6 +1004,224 @@ __global__ void k_tree_build_level(double* sum_tree, double* min_tree,
 }
 
 // ===========================================================================
+// MFMA tiled GEMM kernels (wide-batch learner path, f32-in matrix cores)
+// ===========================================================================
+// For the large-batch config (BASELINE config 5: B=4096, H=1024) the GEMMs
+// are compute-bound and belong on the matrix cores.  gfx950's f32-in MFMA
+// (v_mfma_f32_32x32x2_f32) runs at the full fp32 rate (157 TF peak) with
+// EXACT f32 numerics — bitwise a k-ordered fmaf chain, i.e. the same
+// summation order as the per-thread loops of the small-batch kernels.
+// Tile: 128x128 per workgroup (4 waves, each 64x64 = 2x2 of 32x32 MFMA
+// tiles), K staged through LDS in 32-deep slices.
+#define MT_M 128
+#define MT_N 128
+#define MT_K 32
+
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+
+// C[B,out] = act(X[B,in1 (++ concat in2)] @ Wt[in,out] + bias)
+__global__ void __launch_bounds__(256, 1)
+k_mfma_fwd(const float* __restrict__ x1, const float* __restrict__ x2,
+           const float* __restrict__ wt, const float* __restrict__ bias,
+           float* __restrict__ y, int B, int in1, int in2, int out,
+           int act_kind) {
+    __shared__ float As[MT_K][MT_M + 4];
+    __shared__ float Bs[MT_K][MT_N + 4];
+    int in_total = in1 + in2;
+    int ntn = (out + MT_N - 1) / MT_N;
+    int m0 = (blockIdx.x / ntn) * MT_M, n0 = (blockIdx.x % ntn) * MT_N;
+    int tid = threadIdx.x, wid = tid >> 6, lane = tid & 63;
+    int wm0 = (wid >> 1) * 64, wn0 = (wid & 1) * 64;
+    int r = lane & 31, kk2 = lane >> 5;
+    f32x16 acc00 = {}, acc01 = {}, acc10 = {}, acc11 = {};
+    for (int k0 = 0; k0 < in_total; k0 += MT_K) {
+        for (int e = tid; e < MT_M * MT_K; e += 256) {
+            int mm = e / MT_K, kk = e % MT_K;
+            int gm = m0 + mm, gk = k0 + kk;
+            float v = 0.f;
+            if (gm < B && gk < in_total)
+                v = (gk < in1) ? x1[(long)gm * in1 + gk]
+                               : x2[(long)gm * in2 + (gk - in1)];
+            As[kk][mm] = v;
+        }
+        for (int e = tid; e < MT_K * MT_N; e += 256) {
+            int kk = e / MT_N, nn = e % MT_N;
+            int gk = k0 + kk, gn = n0 + nn;
+            Bs[kk][nn] = (gk < in_total && gn < out)
+                ? wt[(long)gk * out + gn] : 0.f;
+        }
+        __syncthreads();
+#pragma unroll
+        for (int ks = 0; ks < MT_K; ks += 2) {
+            float a0 = As[ks + kk2][wm0 + r];
+            float a1 = As[ks + kk2][wm0 + 32 + r];
+            float b0 = Bs[ks + kk2][wn0 + r];
+            float b1 = Bs[ks + kk2][wn0 + 32 + r];
+            acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
+            acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
+            acc10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
+            acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);
+        }
+        __syncthreads();
+    }
+    const f32x16* accs[4] = {&acc00, &acc01, &acc10, &acc11};
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+        int i = t >> 1, j = t & 1;
+#pragma unroll
+        for (int reg = 0; reg < 16; ++reg) {
+            int row = (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+            int col = lane & 31;
+            int gm = m0 + wm0 + i * 32 + row;
+            int gn = n0 + wn0 + j * 32 + col;
+            if (gm < B && gn < out) {
+                float v = (*accs[t])[reg] + bias[gn];
+                if (act_kind == ACT_RELU) v = fmaxf(v, 0.f);
+                else if (act_kind == ACT_TANH) v = tanhf(v);
+                y[(long)gm * out + gn] = v;
+            }
+        }
+    }
+}
+
+// dX[B, in_lo:in_hi] = (dz[B,out] @ Wt^T) * act'(hprev); hprev/dx are
+// [B][span] (span = in_hi - in_lo), hprev null => no mask.
+__global__ void __launch_bounds__(256, 1)
+k_mfma_dx(const float* __restrict__ dz, const float* __restrict__ wt,
+          const float* __restrict__ hprev, float* __restrict__ dx,
+          int B, int in_lo, int in_hi, int out, int prev_act) {
+    __shared__ float As[MT_K][MT_M + 4];
+    __shared__ float Bs[MT_K][MT_N + 4];
+    int span = in_hi - in_lo;
+    int ntn = (span + MT_N - 1) / MT_N;
+    int m0 = (blockIdx.x / ntn) * MT_M;
+    int n0 = (blockIdx.x % ntn) * MT_N;             // relative i tile
+    int tid = threadIdx.x, wid = tid >> 6, lane = tid & 63;
+    int wm0 = (wid >> 1) * 64, wn0 = (wid & 1) * 64;
+    int r = lane & 31, kk2 = lane >> 5;
+    f32x16 acc00 = {}, acc01 = {}, acc10 = {}, acc11 = {};
+    for (int k0 = 0; k0 < out; k0 += MT_K) {
+        for (int e = tid; e < MT_M * MT_K; e += 256) {
+            int mm = e / MT_K, kk = e % MT_K;
+            int gm = m0 + mm, gk = k0 + kk;
+            As[kk][mm] = (gm < B && gk < out)
+                ? dz[(long)gm * out + gk] : 0.f;
+        }
+        // B tile = Wt^T: Bs[k(=o)][n(=i rel)] = wt[i][o]; o fastest on the
+        // read side for coalescing
+        for (int e = tid; e < MT_N * MT_K; e += 256) {
+            int nn = e / MT_K, kk = e % MT_K;
+            int gi = in_lo + n0 + nn, gk = k0 + kk;
+            Bs[kk][nn] = (gi < in_hi && gk < out)
+                ? wt[(long)gi * out + gk] : 0.f;
+        }
+        __syncthreads();
+#pragma unroll
+        for (int ks = 0; ks < MT_K; ks += 2) {
+            float a0 = As[ks + kk2][wm0 + r];
+            float a1 = As[ks + kk2][wm0 + 32 + r];
+            float b0 = Bs[ks + kk2][wn0 + r];
+            float b1 = Bs[ks + kk2][wn0 + 32 + r];
+            acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
+            acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
+            acc10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
+            acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);
+        }
+        __syncthreads();
+    }
+    const f32x16* accs[4] = {&acc00, &acc01, &acc10, &acc11};
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+        int i = t >> 1, j = t & 1;
+#pragma unroll
+        for (int reg = 0; reg < 16; ++reg) {
+            int row = (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+            int col = lane & 31;
+            int gm = m0 + wm0 + i * 32 + row;
+            int gn = n0 + wn0 + j * 32 + col;          // relative i
+            if (gm < B && gn < span) {
+                long rel = (long)gm * span + gn;
+                float m_ = hprev ? act_mask(prev_act, hprev[rel]) : 1.f;
+                dx[rel] = (*accs[t])[reg] * m_;
+            }
+        }
+    }
+}
+
+// dWt[in,out] = X^T[in,B] @ dz[B,out]  (K dimension = batch)
+__global__ void __launch_bounds__(256, 1)
+k_mfma_dw(const float* __restrict__ dz, const float* __restrict__ x1,
+          const float* __restrict__ x2, float* __restrict__ dwt,
+          int B, int in1, int in2, int out) {
+    __shared__ float As[MT_K][MT_M + 4];
+    __shared__ float Bs[MT_K][MT_N + 4];
+    int in_total = in1 + in2;
+    int ntn = (out + MT_N - 1) / MT_N;
+    int m0 = (blockIdx.x / ntn) * MT_M;                // i tile
+    int n0 = (blockIdx.x % ntn) * MT_N;                // o tile
+    int tid = threadIdx.x, wid = tid >> 6, lane = tid & 63;
+    int wm0 = (wid >> 1) * 64, wn0 = (wid & 1) * 64;
+    int r = lane & 31, kk2 = lane >> 5;
+    f32x16 acc00 = {}, acc01 = {}, acc10 = {}, acc11 = {};
+    for (int k0 = 0; k0 < B; k0 += MT_K) {
+        // As[k(=b)][m(=i)] = X[b][i], i fastest on the read side
+        for (int e = tid; e < MT_K * MT_M; e += 256) {
+            int kk = e / MT_M, mm = e % MT_M;
+            int gb = k0 + kk, gi = m0 + mm;
+            float v = 0.f;
+            if (gb < B && gi < in_total)
+                v = (gi < in1) ? x1[(long)gb * in1 + gi]
+                               : x2[(long)gb * in2 + (gi - in1)];
+            As[kk][mm] = v;
+        }
+        for (int e = tid; e < MT_K * MT_N; e += 256) {
+            int kk = e / MT_N, nn = e % MT_N;
+            int gb = k0 + kk, go = n0 + nn;
+            Bs[kk][nn] = (gb < B && go < out)
+                ? dz[(long)gb * out + go] : 0.f;
+        }
+        __syncthreads();
+#pragma unroll
+        for (int ks = 0; ks < MT_K; ks += 2) {
+            float a0 = As[ks + kk2][wm0 + r];
+            float a1 = As[ks + kk2][wm0 + 32 + r];
+            float b0 = Bs[ks + kk2][wn0 + r];
+            float b1 = Bs[ks + kk2][wn0 + 32 + r];
+            acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
+            acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
+            acc10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
+            acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);
+        }
+        __syncthreads();
+    }
+    const f32x16* accs[4] = {&acc00, &acc01, &acc10, &acc11};
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+        int i = t >> 1, j = t & 1;
+#pragma unroll
+        for (int reg = 0; reg < 16; ++reg) {
+            int row = (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+            int col = lane & 31;
+            int gi = m0 + wm0 + i * 32 + row;
+            int go = n0 + wn0 + j * 32 + col;
+            if (gi < in_total && go < out)
+                dwt[(long)gi * out + go] = (*accs[t])[reg];
+        }
+    }
+}
+
+// db[o] = sum_b dz[b][o] — column reduce, one thread per column with a
+// grid-stride batch loop (coalesced across o).
+__global__ void k_bias_reduce(const float* __restrict__ dz,
+                              float* __restrict__ db, int B, int out) {
+    int o = blockIdx.x * blockDim.x + threadIdx.x;
+    if (o >= out) return;
+    float acc = 0.f;
+    for (int b = 0; b < B; ++b) acc += dz[(long)b * out + o];
+    db[o] = acc;
+}
+
+// ===========================================================================
 // Persistent whole-step megakernel (flagship small-batch learner path)
 // ===========================================================================
 // Rationale (measured, profiles/step_kernel_stats round 1): the row-block
@@ -2223,11 +2441,12 @@ public:
     Counters read_counters() {
         Counters h{};
         HIP_CHECK(hipMemcpy(&h, cnt, sizeof(h), hipMemcpyDeviceToHost));
-        // device counters are pre-advanced for the NEXT step (init 1 so the
-        // first step sees t=1); report steps COMPLETED to the host.
+        // adam/rng counters are pre-advanced for the NEXT step (init 1 so
+        // the first step sees t=1); report steps COMPLETED to the host.
+        // beta_t starts at 0 (LinearSchedule value_at(0) on the first
+        // sample) and ticks at step end, so its raw value IS steps done.
         h.adam_t_actor -= 1;
         h.adam_t_critic -= 1;
-        h.beta_t -= 1;
         h.rng_epoch -= 1;
         return h;
     }
