@@ -320,9 +320,21 @@ class DDPG:
             "train_steps_done": self.train_steps_done,
             "torch_rng": torch.get_rng_state(),
         }
-        if self.prioritized_replay:
+        if self.prioritized_replay and hasattr(self.replayBuffer,
+                                               "state_dict"):
             st["beta_schedule"] = self.beta_schedule.state_dict()
             st["replay"] = self.replayBuffer.state_dict()
+        if self._fused is not None:
+            # device-side optimizer state (Adam moments live in engine
+            # slabs, not in the torch optimizers) + schedule counters
+            eng = self._fused.engine
+            st["engine"] = {
+                "m_actor": eng.store_slab("m_actor"),
+                "v_actor": eng.store_slab("v_actor"),
+                "m_critic": eng.store_slab("m_critic"),
+                "v_critic": eng.store_slab("v_critic"),
+                "counters": eng.counters(),
+            }
         return st
 
     def load_state_dict(self, st: dict, load_replay: bool = True) -> None:
@@ -336,5 +348,12 @@ class DDPG:
         torch.set_rng_state(st["torch_rng"])
         if self.prioritized_replay and "beta_schedule" in st:
             self.beta_schedule.load_state_dict(st["beta_schedule"])
-            if load_replay and "replay" in st:
+            if (load_replay and "replay" in st
+                    and hasattr(self.replayBuffer, "load_state_dict")):
                 self.replayBuffer.load_state_dict(st["replay"])
+        if self._fused is not None and "engine" in st:
+            eng = self._fused.engine
+            eng.load_from_modules(self.actor, self.actor_target,
+                                  self.critic, self.critic_target)
+            for k in ("m_actor", "v_actor", "m_critic", "v_critic"):
+                eng.load_slab(k, st["engine"][k])

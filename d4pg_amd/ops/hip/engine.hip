@@ -1271,7 +1271,15 @@ struct PStepArgs {
     float *pd3, *pd2, *pdh1, *adz, *az1, *az2, *az3;
     Counters* cnt;
     unsigned long long* gbar;        // [0] = arrival counter, [1] = base
+    unsigned long long* tstamp;      // [64] per-phase s_memrealtime stamps
 };
+
+// phase timing probe: wg0/lane0 stamps the wall clock (100 MHz constant
+// clock) after each barrier of the FIRST step of a launch; read back via
+// read_buffer("tstamp") to locate slow phases.
+#define PTIME(g, s, i) \
+    do { if (blockIdx.x == 0 && threadIdx.x == 0 && (s) == 0) \
+        (g).tstamp[i] = __builtin_amdgcn_s_memrealtime(); } while (0)
 
 // software grid barrier: monotonic arrival counter, per-thread local target.
 // Safe because all PNWG workgroups are co-resident (64 wgs of 256 threads /
@@ -1656,9 +1664,10 @@ k_step_persistent(PStepArgs g, int nsteps) {
     const int O = g.O, A = g.A, H = g.H, K = g.K, B = g.B;
 
     for (int s = 0; s < nsteps; ++s) {
+        PTIME(g, s, 0);
         // PH0: PER sample + batch gather
         p_sample(g);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 1);
         // PH1: four independent L1s (16 wgs each)
         if (wg < 16)
             p_fwd(lds, g.bs2, nullptr, at.w1, at.b1, g.at_h1, B, O, 0, H,
@@ -1672,7 +1681,7 @@ k_step_persistent(PStepArgs g, int nsteps) {
         else
             p_fwd(lds, g.bs, nullptr, a.w1, a.b1, g.pa_h1, B, O, 0, H,
                   ACT_RELU, wg - 48, 16);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 2);
         // PH2: actor_t.L2 | critic.L2(cat h1, a) | actor.L2
         if (wg < 21)
             p_fwd(lds, g.at_h1, nullptr, at.w2, at.b2, g.at_h2, B, H, 0, H,
@@ -1683,7 +1692,7 @@ k_step_persistent(PStepArgs g, int nsteps) {
         else
             p_fwd(lds, g.pa_h1, nullptr, a.w2, a.b2, g.pa_h2, B, H, 0, H,
                   ACT_NONE, wg - 42, 22);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 3);
         // PH3: L3s
         if (wg < 21)
             p_fwd(lds, g.at_h2, nullptr, at.w3, at.b3, g.at_h3, B, H, 0, H,
@@ -1694,7 +1703,7 @@ k_step_persistent(PStepArgs g, int nsteps) {
         else
             p_fwd(lds, g.pa_h2, nullptr, a.w3, a.b3, g.pa_h3, B, H, 0, H,
                   ACT_RELU, wg - 42, 22);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 4);
         // PH4: heads — actor_t tanh -> a2 | critic softmax -> q |
         //      actor tanh -> a_out
         if (wg < 21)
@@ -1706,32 +1715,32 @@ k_step_persistent(PStepArgs g, int nsteps) {
         else
             p_fwd(lds, g.pa_h3, nullptr, a.w4, a.b4, g.a_out, B, H, 0, A,
                   ACT_TANH, wg - 42, 22);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 5);
         // PH5-7: critic_target chain on (s2, a2)
         p_fwd(lds, g.ct_h1, g.a2, ct.w2, ct.b2, g.ct_h2, B, H, A, H,
               ACT_RELU, wg, PNWG);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 6);
         p_fwd(lds, g.ct_h2, nullptr, ct.w3, ct.b3, g.ct_h3, B, H, 0, H,
               ACT_RELU, wg, PNWG);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 7);
         p_fwd(lds, g.ct_h3, nullptr, ct.w4, ct.b4, g.p_t, B, H, 0, K,
               ACT_SOFTMAX, wg, PNWG);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 8);
         // PH8: C51 projection; PH9: CE grad + priorities
         p_project(g, lds);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 9);
         p_ce_grad(g);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 10);
         // PH10-12: critic dX chain (pre-update weights)
         p_bwd_dx(lds, g.dlog, c.w4, 0, H, K, B, g.c_h3, ACT_RELU, g.d3,
                  wg, PNWG);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 11);
         p_bwd_dx(lds, g.d3, c.w3, 0, H, H, B, g.c_h2, ACT_RELU, g.d2,
                  wg, PNWG);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 12);
         p_bwd_dx(lds, g.d2, c.w2, 0, H, H, B, g.c_h1, ACT_RELU, g.d1,
                  wg, PNWG);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 13);
         // PH13: critic dW, 4 jobs split by tile count (l1 is the biggest)
         if (wg < 4)
             p_dw(lds, g.d1, g.bs, nullptr, g.g_critic + g.cl[0].w_off,
@@ -1745,49 +1754,49 @@ k_step_persistent(PStepArgs g, int nsteps) {
         else
             p_dw(lds, g.dlog, g.c_h3, nullptr, g.g_critic + g.cl[3].w_off,
                  g.g_critic + g.cl[3].b_off, B, H, 0, K, wg - 56, 8);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 14);
         // PH14: Adam + target soft-update, critic
         p_adam_lerp(g.p_critic, g.g_critic, g.m_critic, g.v_critic,
                     g.p_critic_t, g.n_critic, g.lr_critic, g.tau,
                     g.cnt->adam_t_critic);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 15);
         // PH15-18: critic(s, a_out) with UPDATED critic params
         p_fwd(lds, g.bs, nullptr, c.w1, c.b1, g.pc_h1, B, O, 0, H,
               ACT_RELU, wg, PNWG);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 16);
         p_fwd(lds, g.pc_h1, g.a_out, c.w2, c.b2, g.pc_h2, B, H, A, H,
               ACT_RELU, wg, PNWG);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 17);
         p_fwd(lds, g.pc_h2, nullptr, c.w3, c.b3, g.pc_h3, B, H, 0, H,
               ACT_RELU, wg, PNWG);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 18);
         p_fwd(lds, g.pc_h3, nullptr, c.w4, c.b4, g.pq, B, H, 0, K,
               ACT_SOFTMAX, wg, PNWG);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 19);
         // PH19: policy head gradient
         p_policy_grad(g);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 20);
         // PH20-22: dX back through the critic, ending at the tanh-masked
         // action slice (adz)
         p_bwd_dx(lds, g.pd3, c.w4, 0, H, K, B, g.pc_h3, ACT_RELU, g.pd2,
                  wg, PNWG);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 21);
         p_bwd_dx(lds, g.pd2, c.w3, 0, H, H, B, g.pc_h2, ACT_RELU, g.pdh1,
                  wg, PNWG);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 22);
         p_bwd_dx(lds, g.pdh1, c.w2, H, H + A, H, B, g.a_out, ACT_TANH,
                  g.adz, wg, PNWG);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 23);
         // PH23-25: dX through the actor (az3, az2, az1)
         p_bwd_dx(lds, g.adz, a.w4, 0, H, A, B, g.pa_h3, ACT_RELU, g.az3,
                  wg, PNWG);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 24);
         p_bwd_dx(lds, g.az3, a.w3, 0, H, H, B, nullptr, ACT_NONE, g.az2,
                  wg, PNWG);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 25);
         p_bwd_dx(lds, g.az2, a.w2, 0, H, H, B, g.pa_h1, ACT_RELU, g.az1,
                  wg, PNWG);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 26);
         // PH26: actor dW
         if (wg < 4)
             p_dw(lds, g.az1, g.bs, nullptr, g.g_actor + g.al[0].w_off,
@@ -1801,15 +1810,15 @@ k_step_persistent(PStepArgs g, int nsteps) {
         else
             p_dw(lds, g.adz, g.pa_h3, nullptr, g.g_actor + g.al[3].w_off,
                  g.g_actor + g.al[3].b_off, B, H, 0, A, wg - 56, 8);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 27);
         // PH27: Adam + soft-update, actor
         p_adam_lerp(g.p_actor, g.g_actor, g.m_actor, g.v_actor,
                     g.p_actor_t, g.n_actor, g.lr_actor, g.tau,
                     g.cnt->adam_t_actor);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 28);
         // PH28: PER priority write-back + counter tick
         p_per_update(g);
-        p_bar(ctr, tgt);
+        p_bar(ctr, tgt); PTIME(g, s, 29);
     }
     if (wg == 0 && threadIdx.x == 0) g.gbar[1] = tgt;
 }
@@ -1889,6 +1898,7 @@ public:
     float *ing_s, *ing_a, *ing_r, *ing_s2, *ing_d;  // ingestion staging
     int ing_cap;
     unsigned long long* gbar;                       // persistent grid barrier
+    unsigned long long* tstamp;                     // phase timing probe
 
     Engine(const EngineCfg& c) : cfg(c) {
         if (c.hidden + c.act > FWD_XMAX || c.obs > FWD_XMAX)
@@ -1981,6 +1991,7 @@ public:
         az2 = carve<float>((long)B * H, off);
         az3 = carve<float>((long)B * H, off);
         gbar = carve<unsigned long long>(2, off);
+        tstamp = carve<unsigned long long>(64, off);
         ing_s = carve<float>((long)ing_cap * O, off);
         ing_a = carve<float>((long)ing_cap * A, off);
         ing_r = carve<float>(ing_cap, off);
@@ -2027,9 +2038,28 @@ public:
         return j;
     }
 
+    bool mfma_eligible(int in_total, int out, int act_kind) const {
+        // wide-batch GEMMs go to the matrix cores; the tiny heads and the
+        // wave-wide softmax stay on the per-layer VALU kernels
+        return cfg.batch >= 512 && out >= 64 && in_total >= 32 &&
+               act_kind != ACT_SOFTMAX;
+    }
+
     void launch_fwd(std::initializer_list<FwdJob> jobs) {
         FwdJob a[3] = {};
         int n = 0, wgs = 0;
+        bool all_mfma = cfg.batch >= 512;
+        for (auto& j : jobs)
+            all_mfma = all_mfma && mfma_eligible(j.in1 + j.in2, j.out, j.act);
+        if (all_mfma) {
+            for (auto& j : jobs) {
+                int ntm = ceil_div(j.B, MT_M), ntn = ceil_div(j.out, MT_N);
+                hipLaunchKernelGGL(k_mfma_fwd, dim3(ntm * ntn), dim3(256),
+                                   0, stream, j.x1, j.x2, j.wt, j.bias, j.y,
+                                   j.B, j.in1, j.in2, j.out, j.act);
+            }
+            return;
+        }
         for (auto& j : jobs) {
             a[n++] = j;
             wgs = j.wg0 + j.nwg_b * j.nwg_o;
@@ -2042,6 +2072,36 @@ public:
                     const float* slab, float* gslab, const LayerDesc& l,
                     float* dx1, float* dx2, const float* h1, int prev_act,
                     bool want_dw) {
+        int in_total = l.in1 + l.in2;
+        if (cfg.batch >= 512 && l.out >= 64 && in_total >= 32) {
+            if (want_dw) {
+                int ntm = ceil_div(in_total, MT_M), ntn = ceil_div(l.out, MT_N);
+                hipLaunchKernelGGL(k_mfma_dw, dim3(ntm * ntn), dim3(256),
+                                   0, stream, dz, x1, x2, gslab + l.w_off,
+                                   cfg.batch, l.in1, l.in2, l.out);
+                hipLaunchKernelGGL(k_bias_reduce,
+                                   dim3(ceil_div(l.out, 256)), dim3(256),
+                                   0, stream, dz, gslab + l.b_off,
+                                   cfg.batch, l.out);
+            }
+            if (dx1) {
+                int ntm = ceil_div(cfg.batch, MT_M);
+                int ntn = ceil_div(l.in1, MT_N);
+                hipLaunchKernelGGL(k_mfma_dx, dim3(ntm * ntn), dim3(256),
+                                   0, stream, dz, slab + l.w_off, h1, dx1,
+                                   cfg.batch, 0, l.in1, l.out, prev_act);
+            }
+            if (dx2) {
+                int ntm = ceil_div(cfg.batch, MT_M);
+                int ntn = ceil_div(l.in2, MT_N);
+                hipLaunchKernelGGL(k_mfma_dx, dim3(ntm * ntn), dim3(256),
+                                   0, stream, dz, slab + l.w_off,
+                                   (const float*)nullptr, dx2,
+                                   cfg.batch, l.in1, in_total, l.out,
+                                   ACT_NONE);
+            }
+            return;
+        }
         BwdJob j{};
         j.dz = dz; j.x1 = x1; j.x2 = x2;
         j.wt = slab + l.w_off;
@@ -2106,7 +2166,7 @@ public:
         g.pc_h1 = pc_h1; g.pc_h2 = pc_h2; g.pc_h3 = pc_h3; g.pq = pq;
         g.pd3 = pd3; g.pd2 = pd2; g.pdh1 = pdh1; g.adz = adz;
         g.az1 = az1; g.az2 = az2; g.az3 = az3;
-        g.cnt = cnt; g.gbar = gbar;
+        g.cnt = cnt; g.gbar = gbar; g.tstamp = tstamp;
         return g;
     }
 

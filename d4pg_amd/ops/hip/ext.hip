@@ -180,6 +180,7 @@ static torch::Tensor read_buffer(int64_t h, std::string name) {
         {"c_h2", {e.c_h2, {B, H}, false, false}},
         {"c_h3", {e.c_h3, {B, H}, false, false}},
         {"at_h1", {e.at_h1, {B, H}, false, false}},
+        {"tstamp", {e.tstamp, {64}, true, false}},
         {"sum_tree", {e.sum_tree, {2 * e.tree_cap}, false, true}},
         {"min_tree", {e.min_tree, {2 * e.tree_cap}, false, true}},
     };

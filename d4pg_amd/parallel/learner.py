@@ -152,6 +152,12 @@ class DistributedD4PG:
         self.ewma = None
         self.writer = writer
         self.run_dir = run_dir_name(args)
+        # failure visibility: the reference silently loses
+        # throughput when a worker dies; here the learner tracks per-rank
+        # heartbeats (a rank that gathers 0 transitions for many
+        # consecutive rounds is flagged).
+        self.last_seen = {r: 0 for r in range(1, self.world)}
+        self.heartbeat_warn_rounds = 10
 
     # -- round phases -----------------------------------------------------
 
@@ -194,6 +200,7 @@ class DistributedD4PG:
         cnt = torch.tensor([float(len(lb))], device=dev)
         counts = [torch.zeros_like(cnt) for _ in range(self.world)]
         dist.all_gather(counts, cnt)
+        self._last_counts = counts
         buf, n = _encode(lb.items, self.push_cap, self.obs_dim,
                          self.act_dim, dev)
         blocks = [torch.zeros_like(buf) for _ in range(self.world)]
@@ -232,11 +239,25 @@ class DistributedD4PG:
 
     # -- main loop --------------------------------------------------------
 
+    def _heartbeat(self, counts, rnd):
+        for r in range(1, self.world):
+            if r == self.eval_rank:
+                continue
+            if int(counts[r].item()) > 0:
+                self.last_seen[r] = rnd
+            elif rnd - self.last_seen[r] >= self.heartbeat_warn_rounds:
+                print(f"[learner] WARNING: actor rank {r} silent for "
+                      f"{rnd - self.last_seen[r]} rounds (dead actor? the "
+                      f"learner keeps training on remaining actors)",
+                      flush=True)
+
     def run(self, rounds: int, save: bool = False):
         for rnd in range(rounds):
             self._broadcast_params()
             lb = self._collect()
             ingested = self._exchange(lb)
+            if self.is_learner:
+                self._heartbeat(self._last_counts, rnd)
             self._train()
             if self.is_learner:
                 if self.writer is not None:
