@@ -1242,7 +1242,9 @@ __global__ void k_bias_reduce(const float* __restrict__ dz,
 #define PNWG 64
 #define PROWS 16
 #define PXMAX 512
-#define PLDS_FLOATS (PROWS * PXMAX + 256)
+// pool: [x/dz stage 16*512][w chunk 64*65][pad]
+#define PWOFF (PROWS * PXMAX)
+#define PLDS_FLOATS (PROWS * PXMAX + 64 * 65 + 64)
 
 struct PStepArgs {
     int B, O, A, H, K;
@@ -1339,18 +1341,36 @@ __device__ __noinline__ void p_fwd(float* lds, const float* x1, const float* x2,
         int rq = tid >> 6, c = tid & 63;
         int o = c0 + c;
         float acc0 = 0.f, acc1 = 0.f, acc2 = 0.f, acc3 = 0.f;
-        if (o < out) {
-            const float* wcol = wt + o;
-            const float* xr0 = lds + (rq + 0) * in_total;
-            const float* xr1 = lds + (rq + 4) * in_total;
-            const float* xr2 = lds + (rq + 8) * in_total;
-            const float* xr3 = lds + (rq + 12) * in_total;
-#pragma unroll 8
-            for (int k = 0; k < in_total; ++k) {
-                float wv = wcol[(long)k * out];
-                acc0 += xr0[k] * wv; acc1 += xr1[k] * wv;
-                acc2 += xr2[k] * wv; acc3 += xr3[k] * wv;
+        // K-chunked: stage the 64-deep weight slice in LDS with COALESCED
+        // 256-thread loads (the direct per-lane column walk has stride
+        // out*4 B and zero memory-level parallelism — measured 66 us per
+        // 64x256x256 GEMM phase vs ~6 us with this staging), then FMA
+        // against conflict-free LDS reads.
+        float* ws = lds + PWOFF;                 // [64][65]
+        const float* xr0 = lds + (rq + 0) * in_total;
+        const float* xr1 = lds + (rq + 4) * in_total;
+        const float* xr2 = lds + (rq + 8) * in_total;
+        const float* xr3 = lds + (rq + 12) * in_total;
+        for (int kc = 0; kc < in_total; kc += 64) {
+            int klen = min(64, in_total - kc);
+            for (int e = tid; e < (klen << 6); e += 256) {
+                int kk = e >> 6, cc = e & 63;
+                int gc = c0 + cc;
+                ws[kk * 65 + cc] = (gc < out)
+                    ? wt[(long)(kc + kk) * out + gc] : 0.f;
             }
+            __syncthreads();
+            if (o < out) {
+#pragma unroll 8
+                for (int k = 0; k < klen; ++k) {
+                    float wv = ws[k * 65 + c];
+                    acc0 += xr0[kc + k] * wv; acc1 += xr1[kc + k] * wv;
+                    acc2 += xr2[kc + k] * wv; acc3 += xr3[kc + k] * wv;
+                }
+            }
+            __syncthreads();
+        }
+        if (o < out) {
             float bv = bias[o];
             acc0 += bv; acc1 += bv; acc2 += bv; acc3 += bv;
         }
@@ -1406,18 +1426,32 @@ __device__ inline void p_bwd_dx(float* lds, const float* dz, const float* wt,
         int rq = tid >> 6, c = tid & 63;
         int i = i0 + c;
         float acc0 = 0.f, acc1 = 0.f, acc2 = 0.f, acc3 = 0.f;
-        if (i < in_hi) {
-            const float* wrow = wt + (long)i * out;
-            const float* z0 = lds + (rq + 0) * out;
-            const float* z1 = lds + (rq + 4) * out;
-            const float* z2 = lds + (rq + 8) * out;
-            const float* z3 = lds + (rq + 12) * out;
-#pragma unroll 8
-            for (int o = 0; o < out; ++o) {
-                float wv = wrow[o];
-                acc0 += z0[o] * wv; acc1 += z1[o] * wv;
-                acc2 += z2[o] * wv; acc3 += z3[o] * wv;
+        // O-chunked LDS staging of the weight tile (transposed write, so
+        // the compute reads ws[o][i] conflict-free) — same latency fix as
+        // the forward path.
+        float* ws = lds + PWOFF;                 // [64][65] = [o][i]
+        const float* z0 = lds + (rq + 0) * out;
+        const float* z1 = lds + (rq + 4) * out;
+        const float* z2 = lds + (rq + 8) * out;
+        const float* z3 = lds + (rq + 12) * out;
+        for (int oc = 0; oc < out; oc += 64) {
+            int olen = min(64, out - oc);
+            for (int e = tid; e < 4096; e += 256) {
+                int ii = e >> 6, oo = e & 63;     // oo fastest => coalesced
+                int gi = i0 + ii;
+                ws[oo * 65 + ii] = (gi < in_hi && oo < olen)
+                    ? wt[(long)gi * out + (oc + oo)] : 0.f;
             }
+            __syncthreads();
+            if (i < in_hi) {
+#pragma unroll 8
+                for (int o = 0; o < olen; ++o) {
+                    float wv = ws[o * 65 + c];
+                    acc0 += z0[oc + o] * wv; acc1 += z1[oc + o] * wv;
+                    acc2 += z2[oc + o] * wv; acc3 += z3[oc + o] * wv;
+                }
+            }
+            __syncthreads();
         }
         if (i < in_hi) {
             float accs[4] = {acc0, acc1, acc2, acc3};
