@@ -1283,25 +1283,42 @@ struct PStepArgs {
     do { if (blockIdx.x == 0 && threadIdx.x == 0 && (s) == 0) \
         (g).tstamp[i] = __builtin_amdgcn_s_memrealtime(); } while (0)
 
-// software grid barrier: monotonic arrival counter, per-thread local target.
-// Safe because all PNWG workgroups are co-resident (64 wgs of 256 threads /
-// 34 KB LDS on a 256-CU chip).  The base survives across launches in
-// gbar[1] (written by wg 0 after the final barrier; stream order makes the
+// software grid barrier — tree arrival + write-once go-flag.  Safe because
+// all PNWG workgroups are co-resident (64 wgs of 256 threads on 256 CUs).
+//
+// Layout in gbar[]: 8 group counters at [g*16] (separate 128 B lines),
+// root counter at [128], go-flag at [144], cross-launch round base at
+// [160] (written by wg 0 after the final barrier; stream order makes the
 // next launch's read race-free).
-__device__ inline void p_bar(unsigned long long* ctr,
-                             unsigned long long& tgt) {
+//
+// Design notes (measured): a single central atomic that everyone arrives
+// on AND polls makes idle workgroups' poll storm contend with the active
+// workgroups' real memory traffic — single-GEMM phases went from ~6 us to
+// ~40 us under it.  Here arrivals fan in through per-group lines, pollers
+// read only the once-written flag, and the spin backs off exponentially
+// (s_sleep needs an immediate, hence the two-stage backoff).
+__device__ inline void p_bar(unsigned long long* gb,
+                             unsigned long long& round) {
     __syncthreads();
-    tgt += PNWG;
+    round += 1;
     if (threadIdx.x == 0) {
         __threadfence();                       // release: flush our writes
-        atomicAdd(ctr, 1ull);
-        volatile unsigned long long* vctr = ctr;
+        int grp = blockIdx.x & 7;
+        unsigned long long old = atomicAdd(&gb[grp << 4], 1ull);
+        if (old + 1 == round * (PNWG >> 3)) {  // last of the group
+            unsigned long long r = atomicAdd(&gb[128], 1ull);
+            if (r + 1 == round * 8)            // last group overall
+                __hip_atomic_store(&gb[144], round, __ATOMIC_RELEASE,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+        }
+        volatile unsigned long long* f = &gb[144];
         long spins = 0;
-        while (*vctr < tgt) {
-            __builtin_amdgcn_s_sleep(2);
-            // safety valve (~10 s): a barrier logic bug must never
-            // hard-hang the GPU — desynchronize and terminate instead
-            if (++spins > (1L << 28)) break;
+        while (*f < round) {
+            // safety valve: a barrier logic bug must never hard-hang the
+            // GPU — desynchronize and terminate instead (~10 s)
+            if (++spins > (1L << 26)) break;
+            if (spins < 8) __builtin_amdgcn_s_sleep(2);
+            else __builtin_amdgcn_s_sleep(32);
         }
         __threadfence();                       // acquire: invalidate L1
     }
@@ -1512,8 +1529,90 @@ __device__ inline void p_dw(float* lds, const float* dz, const float* x1,
     }
 }
 
-__device__ inline void p_adam_lerp(float* p, const float* gr, float* m,
-                                   float* v, float* tgt_slab, long n,
+// dW tile [64 i x 64 o] per wg: x^T and dz chunks staged in LDS (coalesced),
+// each thread owns a 4x4 output patch (ig = tid>>4 picks 4 i's, og = tid&15
+// picks 4 o's) — 8 LDS reads + 16 FMAs per batch row.  Bias rows fold in on
+// the i0==0 tiles.  Accumulation over b ascending, same order as p_dw/bwd_one.
+__device__ __noinline__ void p_dw2(float* lds, const float* dz,
+                                   const float* x1, const float* x2,
+                                   float* dwt, float* dbias,
+                                   int B, int in1, int in2, int out,
+                                   int wg_rel, int nwg) {
+    int in_total = in1 + in2;
+    int nti = (in_total + 63) >> 6, nto = (out + 63) >> 6;
+    int ntiles = nti * nto;
+    float* xs = lds;                       // [64 b][65] i-slice
+    float* zs = lds + 64 * 65;             // [64 b][65] o-slice
+    int tid = threadIdx.x;
+    int ig = tid >> 4, og = tid & 15;
+    for (int t = wg_rel; t < ntiles; t += nwg) {
+        int i0 = (t / nto) << 6, o0 = (t % nto) << 6;
+        float acc[4][4] = {};
+        float accb[4] = {};
+        for (int bc = 0; bc < B; bc += 64) {
+            for (int e = tid; e < 4096; e += 256) {
+                int bb = e >> 6, ii = e & 63;
+                int gb = bc + bb, gi = i0 + ii;
+                float xv = 0.f;
+                if (gb < B && gi < in_total)
+                    xv = (gi < in1) ? x1[(long)gb * in1 + gi]
+                                    : x2[(long)gb * in2 + (gi - in1)];
+                xs[bb * 65 + ii] = xv;
+                int go = o0 + ii;
+                zs[bb * 65 + ii] = (gb < B && go < out)
+                    ? dz[(long)gb * out + go] : 0.f;
+            }
+            __syncthreads();
+            int blim = min(64, B - bc);
+            for (int b = 0; b < blim; ++b) {
+                float xv0 = xs[b * 65 + ig * 4 + 0];
+                float xv1 = xs[b * 65 + ig * 4 + 1];
+                float xv2 = xs[b * 65 + ig * 4 + 2];
+                float xv3 = xs[b * 65 + ig * 4 + 3];
+                float zv0 = zs[b * 65 + og * 4 + 0];
+                float zv1 = zs[b * 65 + og * 4 + 1];
+                float zv2 = zs[b * 65 + og * 4 + 2];
+                float zv3 = zs[b * 65 + og * 4 + 3];
+                acc[0][0] += xv0 * zv0; acc[0][1] += xv0 * zv1;
+                acc[0][2] += xv0 * zv2; acc[0][3] += xv0 * zv3;
+                acc[1][0] += xv1 * zv0; acc[1][1] += xv1 * zv1;
+                acc[1][2] += xv1 * zv2; acc[1][3] += xv1 * zv3;
+                acc[2][0] += xv2 * zv0; acc[2][1] += xv2 * zv1;
+                acc[2][2] += xv2 * zv2; acc[2][3] += xv2 * zv3;
+                acc[3][0] += xv3 * zv0; acc[3][1] += xv3 * zv1;
+                acc[3][2] += xv3 * zv2; acc[3][3] += xv3 * zv3;
+                if (ig == 0 && i0 == 0) {
+                    accb[0] += zv0; accb[1] += zv1;
+                    accb[2] += zv2; accb[3] += zv3;
+                }
+            }
+            __syncthreads();
+        }
+#pragma unroll
+        for (int a = 0; a < 4; ++a) {
+            int gi = i0 + ig * 4 + a;
+            if (gi >= in_total) continue;
+#pragma unroll
+            for (int bb = 0; bb < 4; ++bb) {
+                int go = o0 + og * 4 + bb;
+                if (go < out) dwt[(long)gi * out + go] = acc[a][bb];
+            }
+        }
+        if (ig == 0 && i0 == 0 && dbias) {
+#pragma unroll
+            for (int bb = 0; bb < 4; ++bb) {
+                int go = o0 + og * 4 + bb;
+                if (go < out) dbias[go] = accb[bb];
+            }
+        }
+    }
+}
+
+__device__ inline void p_adam_lerp(float* __restrict__ p,
+                                   const float* __restrict__ gr,
+                                   float* __restrict__ m,
+                                   float* __restrict__ v,
+                                   float* __restrict__ tgt_slab, long n,
                                    float lr, float tau, long long t) {
     const float b1 = 0.9f, b2 = 0.999f, eps = 1e-8f;
     float bc1 = 1.f - __powf(b1, (float)t);
@@ -1688,7 +1787,7 @@ __device__ inline void p_per_update(const PStepArgs& g) {
 __global__ void __launch_bounds__(256, 1)
 k_step_persistent(PStepArgs g, int nsteps) {
     __shared__ float lds[PLDS_FLOATS];
-    unsigned long long tgt = g.gbar[1];
+    unsigned long long tgt = g.gbar[160];   // cross-launch round base
     unsigned long long* ctr = g.gbar;
     int wg = blockIdx.x;
     NetPtrs a = net_ptrs(g.p_actor, g.al);
@@ -1777,16 +1876,16 @@ k_step_persistent(PStepArgs g, int nsteps) {
         p_bar(ctr, tgt); PTIME(g, s, 13);
         // PH13: critic dW, 4 jobs split by tile count (l1 is the biggest)
         if (wg < 4)
-            p_dw(lds, g.d1, g.bs, nullptr, g.g_critic + g.cl[0].w_off,
+            p_dw2(lds, g.d1, g.bs, nullptr, g.g_critic + g.cl[0].w_off,
                  g.g_critic + g.cl[0].b_off, B, O, 0, H, wg, 4);
         else if (wg < 32)
-            p_dw(lds, g.d2, g.c_h1, g.ba, g.g_critic + g.cl[1].w_off,
+            p_dw2(lds, g.d2, g.c_h1, g.ba, g.g_critic + g.cl[1].w_off,
                  g.g_critic + g.cl[1].b_off, B, H, A, H, wg - 4, 28);
         else if (wg < 56)
-            p_dw(lds, g.d3, g.c_h2, nullptr, g.g_critic + g.cl[2].w_off,
+            p_dw2(lds, g.d3, g.c_h2, nullptr, g.g_critic + g.cl[2].w_off,
                  g.g_critic + g.cl[2].b_off, B, H, 0, H, wg - 32, 24);
         else
-            p_dw(lds, g.dlog, g.c_h3, nullptr, g.g_critic + g.cl[3].w_off,
+            p_dw2(lds, g.dlog, g.c_h3, nullptr, g.g_critic + g.cl[3].w_off,
                  g.g_critic + g.cl[3].b_off, B, H, 0, K, wg - 56, 8);
         p_bar(ctr, tgt); PTIME(g, s, 14);
         // PH14: Adam + target soft-update, critic
@@ -1833,16 +1932,16 @@ k_step_persistent(PStepArgs g, int nsteps) {
         p_bar(ctr, tgt); PTIME(g, s, 26);
         // PH26: actor dW
         if (wg < 4)
-            p_dw(lds, g.az1, g.bs, nullptr, g.g_actor + g.al[0].w_off,
+            p_dw2(lds, g.az1, g.bs, nullptr, g.g_actor + g.al[0].w_off,
                  g.g_actor + g.al[0].b_off, B, O, 0, H, wg, 4);
         else if (wg < 30)
-            p_dw(lds, g.az2, g.pa_h1, nullptr, g.g_actor + g.al[1].w_off,
+            p_dw2(lds, g.az2, g.pa_h1, nullptr, g.g_actor + g.al[1].w_off,
                  g.g_actor + g.al[1].b_off, B, H, 0, H, wg - 4, 26);
         else if (wg < 56)
-            p_dw(lds, g.az3, g.pa_h2, nullptr, g.g_actor + g.al[2].w_off,
+            p_dw2(lds, g.az3, g.pa_h2, nullptr, g.g_actor + g.al[2].w_off,
                  g.g_actor + g.al[2].b_off, B, H, 0, H, wg - 30, 26);
         else
-            p_dw(lds, g.adz, g.pa_h3, nullptr, g.g_actor + g.al[3].w_off,
+            p_dw2(lds, g.adz, g.pa_h3, nullptr, g.g_actor + g.al[3].w_off,
                  g.g_actor + g.al[3].b_off, B, H, 0, A, wg - 56, 8);
         p_bar(ctr, tgt); PTIME(g, s, 27);
         // PH27: Adam + soft-update, actor
@@ -1854,7 +1953,7 @@ k_step_persistent(PStepArgs g, int nsteps) {
         p_per_update(g);
         p_bar(ctr, tgt); PTIME(g, s, 29);
     }
-    if (wg == 0 && threadIdx.x == 0) g.gbar[1] = tgt;
+    if (wg == 0 && threadIdx.x == 0) g.gbar[160] = tgt;
 }
 
 // ===========================================================================
@@ -2024,7 +2123,7 @@ public:
         az1 = carve<float>((long)B * H, off);
         az2 = carve<float>((long)B * H, off);
         az3 = carve<float>((long)B * H, off);
-        gbar = carve<unsigned long long>(2, off);
+        gbar = carve<unsigned long long>(192, off);
         tstamp = carve<unsigned long long>(64, off);
         ing_s = carve<float>((long)ing_cap * O, off);
         ing_a = carve<float>((long)ing_cap * A, off);
