@@ -1345,31 +1345,82 @@ __device__ __noinline__ void p_fwd(float* lds, const float* x1, const float* x2,
     int tid = threadIdx.x;
     for (int t = wg_rel; t < ntiles; t += nwg) {
         int r0 = (t / nct) * PROWS, c0 = (t % nct) * 64;
-        for (int e = tid; e < PROWS * in_total; e += 256) {
-            int rr_ = e / in_total, kk = e % in_total;
-            int gb = r0 + rr_;
-            float v = 0.f;
-            if (gb < B)
-                v = (kk < in1) ? x1[(long)gb * in1 + kk]
-                               : x2[(long)gb * in2 + (kk - in1)];
-            lds[rr_ * in_total + kk] = v;
+        // x-stage, register-batched: fire ALL loads (independent), then
+        // write LDS.  A naive load->ds_write loop serializes one memory
+        // round-trip per 256-thread sweep (measured ~9 us per 16 KB).
+        {
+            int tot = PROWS * in_total;
+            for (int base = 0; base < tot; base += 256 * 8) {
+                float tmp[8];
+#pragma unroll
+                for (int u = 0; u < 8; ++u) {
+                    int e = base + u * 256 + tid;
+                    float v = 0.f;
+                    if (e < tot) {
+                        int rr_ = e / in_total, kk = e % in_total;
+                        int gb = r0 + rr_;
+                        if (gb < B)
+                            v = (kk < in1)
+                                ? x1[(long)gb * in1 + kk]
+                                : x2[(long)gb * in2 + (kk - in1)];
+                    }
+                    tmp[u] = v;
+                }
+#pragma unroll
+                for (int u = 0; u < 8; ++u) {
+                    int e = base + u * 256 + tid;
+                    if (e < tot) lds[e] = tmp[u];
+                }
+            }
         }
         __syncthreads();
         int rq = tid >> 6, c = tid & 63;
         int o = c0 + c;
         float acc0 = 0.f, acc1 = 0.f, acc2 = 0.f, acc3 = 0.f;
-        // K-chunked: stage the 64-deep weight slice in LDS with COALESCED
-        // 256-thread loads (the direct per-lane column walk has stride
-        // out*4 B and zero memory-level parallelism — measured 66 us per
-        // 64x256x256 GEMM phase vs ~6 us with this staging), then FMA
-        // against conflict-free LDS reads.
+        // Weight slice K-chunked through LDS: 64x64 chunks staged with
+        // register-batched coalesced loads, DOUBLE-BUFFERED so chunk k+1's
+        // global loads are in flight while chunk k's FMAs run.
         float* ws = lds + PWOFF;                 // [64][65]
         const float* xr0 = lds + (rq + 0) * in_total;
         const float* xr1 = lds + (rq + 4) * in_total;
         const float* xr2 = lds + (rq + 8) * in_total;
         const float* xr3 = lds + (rq + 12) * in_total;
-        for (int kc = 0; kc < in_total; kc += 64) {
-            int klen = min(64, in_total - kc);
+        int kk16 = tid >> 6;                     // chunk-load geometry:
+        int cc16 = tid & 63;                     // thread u covers row
+        float wreg[16];                          // kk16 + 4u, col cc16
+        int nfull = in_total >> 6;               // # of full 64-chunks
+        auto preload = [&](int kc) {
+#pragma unroll
+            for (int u = 0; u < 16; ++u) {
+                int gk = kc + kk16 + 4 * u, gc = c0 + cc16;
+                wreg[u] = (gc < out)
+                    ? wt[(long)gk * out + gc] : 0.f;
+            }
+        };
+        if (nfull > 0) preload(0);
+        for (int ch = 0; ch < nfull; ++ch) {
+            int kc = ch << 6;
+            float wb[16];
+#pragma unroll
+            for (int u = 0; u < 16; ++u) wb[u] = wreg[u];
+            if (kc + 64 < (nfull << 6)) preload(kc + 64);
+#pragma unroll
+            for (int u = 0; u < 16; ++u)
+                ws[(kk16 + 4 * u) * 65 + cc16] = wb[u];
+            __syncthreads();
+            if (o < out) {
+#pragma unroll 8
+                for (int k = 0; k < 64; ++k) {
+                    float wv = ws[k * 65 + c];
+                    acc0 += xr0[kc + k] * wv; acc1 += xr1[kc + k] * wv;
+                    acc2 += xr2[kc + k] * wv; acc3 += xr3[kc + k] * wv;
+                }
+            }
+            __syncthreads();
+        }
+        // K tail (in_total % 64): simple staging
+        int kc = nfull << 6, klen = in_total - kc;
+        if (klen > 0) {
             for (int e = tid; e < (klen << 6); e += 256) {
                 int kk = e >> 6, cc = e & 63;
                 int gc = c0 + cc;
@@ -1378,7 +1429,6 @@ __device__ __noinline__ void p_fwd(float* lds, const float* x1, const float* x2,
             }
             __syncthreads();
             if (o < out) {
-#pragma unroll 8
                 for (int k = 0; k < klen; ++k) {
                     float wv = ws[k * 65 + c];
                     acc0 += xr0[kc + k] * wv; acc1 += xr1[kc + k] * wv;
@@ -1434,34 +1484,87 @@ __device__ inline void p_bwd_dx(float* lds, const float* dz, const float* wt,
     int tid = threadIdx.x;
     for (int t = wg_rel; t < ntiles; t += nwg) {
         int r0 = (t / nit) * PROWS, i0 = in_lo + (t % nit) * 64;
-        for (int e = tid; e < PROWS * out; e += 256) {
-            int rr_ = e / out, oo = e % out;
-            int gb = r0 + rr_;
-            lds[rr_ * out + oo] = (gb < B) ? dz[(long)gb * out + oo] : 0.f;
+        // dz-stage, register-batched (see p_fwd)
+        {
+            int tot = PROWS * out;
+            for (int base = 0; base < tot; base += 256 * 8) {
+                float tmp[8];
+#pragma unroll
+                for (int u = 0; u < 8; ++u) {
+                    int e = base + u * 256 + tid;
+                    float v = 0.f;
+                    if (e < tot) {
+                        int rr_ = e / out, oo = e % out;
+                        int gb = r0 + rr_;
+                        if (gb < B) v = dz[(long)gb * out + oo];
+                    }
+                    tmp[u] = v;
+                }
+#pragma unroll
+                for (int u = 0; u < 8; ++u) {
+                    int e = base + u * 256 + tid;
+                    if (e < tot) lds[e] = tmp[u];
+                }
+            }
         }
         __syncthreads();
         int rq = tid >> 6, c = tid & 63;
         int i = i0 + c;
         float acc0 = 0.f, acc1 = 0.f, acc2 = 0.f, acc3 = 0.f;
-        // O-chunked LDS staging of the weight tile (transposed write, so
-        // the compute reads ws[o][i] conflict-free) — same latency fix as
-        // the forward path.
+        // Weight tile O-chunked through LDS, transposed write (compute
+        // reads ws[o][i] conflict-free), register-batched + double-
+        // buffered like the forward path.  Load geometry: thread u covers
+        // (i row = cc16, o = oc + kk16 + 4u) — oo fastest => coalesced.
         float* ws = lds + PWOFF;                 // [64][65] = [o][i]
         const float* z0 = lds + (rq + 0) * out;
         const float* z1 = lds + (rq + 4) * out;
         const float* z2 = lds + (rq + 8) * out;
         const float* z3 = lds + (rq + 12) * out;
-        for (int oc = 0; oc < out; oc += 64) {
-            int olen = min(64, out - oc);
+        int oo16 = tid & 3;                      // o sub-offset (0..3)
+        int ii16 = tid >> 2;                     // i row (0..63)
+        float wreg[16];                          // o = oc + oo16 + 4u
+        int gi_ld = i0 + ii16;
+        int nfull = out >> 6;
+        auto preload = [&](int oc) {
+#pragma unroll
+            for (int u = 0; u < 16; ++u) {
+                int go = oc + oo16 + 4 * u;
+                wreg[u] = (gi_ld < in_hi)
+                    ? wt[(long)gi_ld * out + go] : 0.f;
+            }
+        };
+        if (nfull > 0) preload(0);
+        for (int ch = 0; ch < nfull; ++ch) {
+            int oc = ch << 6;
+            float wb[16];
+#pragma unroll
+            for (int u = 0; u < 16; ++u) wb[u] = wreg[u];
+            if (oc + 64 < (nfull << 6)) preload(oc + 64);
+#pragma unroll
+            for (int u = 0; u < 16; ++u)
+                ws[(oo16 + 4 * u) * 65 + ii16] = wb[u];
+            __syncthreads();
+            if (i < in_hi) {
+#pragma unroll 8
+                for (int o = 0; o < 64; ++o) {
+                    float wv = ws[o * 65 + c];
+                    acc0 += z0[oc + o] * wv; acc1 += z1[oc + o] * wv;
+                    acc2 += z2[oc + o] * wv; acc3 += z3[oc + o] * wv;
+                }
+            }
+            __syncthreads();
+        }
+        // O tail
+        int oc = nfull << 6, olen = out - oc;
+        if (olen > 0) {
             for (int e = tid; e < 4096; e += 256) {
-                int ii = e >> 6, oo = e & 63;     // oo fastest => coalesced
+                int ii = e >> 6, oo = e & 63;
                 int gi = i0 + ii;
                 ws[oo * 65 + ii] = (gi < in_hi && oo < olen)
                     ? wt[(long)gi * out + (oc + oo)] : 0.f;
             }
             __syncthreads();
             if (i < in_hi) {
-#pragma unroll 8
                 for (int o = 0; o < olen; ++o) {
                     float wv = ws[o * 65 + c];
                     acc0 += z0[oc + o] * wv; acc1 += z1[oc + o] * wv;
