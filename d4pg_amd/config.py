@@ -74,6 +74,11 @@ def make_parser() -> argparse.ArgumentParser:
     p.add_argument("--eval_trials", default=10, type=int,
                    help="greedy eval rollouts per cycle (reference main.py:309)")
     p.add_argument("--seed", default=0, type=int, help="RNG seed")
+    p.add_argument("--vector_envs", default=0, type=int,
+                   help="MI355X extension: batch M vectorized envs per "
+                        "actor rank (one [M,obs] policy forward per tick) "
+                        "instead of one env per process; 0 = scalar parity "
+                        "mode (distributed learner only)")
     return p
 
 

@@ -144,6 +144,22 @@ class DistributedD4PG:
             1, args.episodes_per_cycle // max(1, self.world - 1))
         mult = 2 if args.her else 1
         self.push_cap = self.episodes_per_round * args.max_steps * mult
+        # vectorized actor mode: M batched envs per actor rank, one [M,obs]
+        # policy forward per tick (Pendulum only; HER needs dict obs)
+        self.vector = None
+        if (int(getattr(args, "vector_envs", 0)) > 0 and not args.her
+                and args.env.startswith("Pendulum")
+                and not self.is_learner and not self.is_evaluator):
+            from ..envs.vector import VecNStep, VectorPendulum
+            m = int(args.vector_envs)
+            self.vector = VectorPendulum(m, seed=seed + 17,
+                                         horizon=args.max_steps)
+            self.vfold = VecNStep(m, self.obs_dim, self.act_dim,
+                                  args.n_steps, args.gamma)
+        if int(getattr(args, "vector_envs", 0)) > 0 and not args.her \
+                and args.env.startswith("Pendulum"):
+            # every rank must agree on the wire size
+            self.push_cap = int(args.vector_envs) * args.max_steps
         self.blob_len = pack_net(self.agent.actor).numel() + 1
 
         self.grad_meter = Meter()
@@ -186,6 +202,25 @@ class DistributedD4PG:
                 0.95 * self.ewma + 0.05 * R
             print(f"[eval] step {self.global_step} return {R:.2f} "
                   f"ewma {self.ewma:.2f}", flush=True)
+            return lb
+        if self.vector is not None:
+            import torch as _t
+            obs = self.vector.reset()
+            self.vfold.reset()
+            actor = self.agent.actor
+            with _t.no_grad():
+                for t in range(self.vector.horizon):
+                    a = actor(_t.from_numpy(obs)).numpy()
+                    a = np.clip(a + 0.3 * self.rng.standard_normal(a.shape),
+                                -1.0, 1.0).astype(np.float32)
+                    obs2, r, done = self.vector.step(a)
+                    out = self.vfold.push(obs, a, r, obs2, done)
+                    if out is not None:
+                        S, A, R, S2, D = out
+                        for e in range(len(R)):
+                            lb.add(S[e], A[e], R[e], S2[e], D[e])
+                    obs = obs2
+            self.env_meter.add(self.vector.horizon * self.vector.n)
             return lb
         for _ in range(self.episodes_per_round):
             ep, _, _ = rollout_episode(self.agent, self.env, noise=True)

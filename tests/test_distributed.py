@@ -91,3 +91,12 @@ def test_encode_decode_roundtrip():
     np.testing.assert_allclose(a[5], lb.items[5][1], rtol=1e-6)
     np.testing.assert_allclose(r, [it[2] for it in lb.items], rtol=1e-6)
     np.testing.assert_allclose(d, [it[4] for it in lb.items])
+
+
+def test_two_rank_vector_actors():
+    out = _run_world(2, extra=("--vector_envs", "8", "--max_steps", "30"),
+                     port=29617)
+    step, replay_len = out["learner"]
+    # 8 envs x ~(30 - n_steps + 1) matured transitions per round x 4 rounds
+    assert replay_len > 500
+    assert step > 0
