@@ -903,6 +903,50 @@ __global__ void k_per_update(double* __restrict__ sum_tree,
     }
 }
 
+// ---- PER write-back, wide-batch variant -------------------------------------
+// At B >= 512 the single-workgroup level-synced k_per_update serializes
+// (measured 305 us at B=4096); instead: one leaves kernel + one kernel per
+// tree level (grid-wide), + a tiny counter-tick kernel.  Duplicate parents
+// recompute identical values (benign, same as the in-wg version).
+__global__ void k_per_leaves(double* __restrict__ sum_tree,
+                             double* __restrict__ min_tree, long tree_cap,
+                             const long* __restrict__ idx,
+                             const float* __restrict__ pri,
+                             int B, float alpha, Counters* cnt) {
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < B;
+         i += (long)gridDim.x * blockDim.x) {
+        float p = pri[i];
+        double pa = pow((double)p, (double)alpha);
+        long leaf = tree_cap + idx[i];
+        sum_tree[leaf] = pa;
+        min_tree[leaf] = pa;
+        // positive floats compare correctly as int bits
+        atomicMax((int*)&cnt->max_priority, __float_as_int(p));
+    }
+}
+
+__global__ void k_per_level(double* __restrict__ sum_tree,
+                            double* __restrict__ min_tree, long tree_cap,
+                            const long* __restrict__ idx, int B, long lv) {
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < B;
+         i += (long)gridDim.x * blockDim.x) {
+        long node = (tree_cap + idx[i]) >> (lv + 1);
+        if (node >= 1) {
+            sum_tree[node] = sum_tree[2 * node] + sum_tree[2 * node + 1];
+            min_tree[node] = fmin(min_tree[2 * node], min_tree[2 * node + 1]);
+        }
+    }
+}
+
+__global__ void k_tick_end(Counters* cnt) {
+    if (threadIdx.x == 0) {
+        cnt->beta_t += 1;
+        cnt->adam_t_actor += 1;
+        cnt->adam_t_critic += 1;
+        cnt->rng_epoch += 1;
+    }
+}
+
 // ---- replay ingestion (batched add) -----------------------------------------
 // T transitions appended at the ring position with priority max_priority^alpha,
 // then one level-synced repair pass.  Single workgroup (T can exceed threads).
@@ -1019,51 +1063,106 @@ __global__ void k_tree_build_level(double* sum_tree, double* min_tree,
 
 typedef float f32x16 __attribute__((ext_vector_type(16)));
 
+// Software-pipelined MFMA tile core shared by fwd/dX/dW: double-buffered
+// LDS (global loads for chunk ch+1 are issued before chunk ch's MFMAs and
+// written to the other buffer after them — one barrier per chunk), with
+// register-batched 16-load staging so every global load is in flight at
+// once.  `ldA`/`ldB` load chunk k0's A/B fragments into 16 registers.
+// If `bias_acc` is non-null, B-fragment values for waves with wm0==0 are
+// summed into bias_acc[0..1] during the MFMA loop (used by dW for db).
+template <typename FA, typename FB>
+__device__ inline void mfma_pipeline(FA ldA, FB ldB, int nch,
+                                     float* As0, float* As1,
+                                     float* Bs0, float* Bs1,
+                                     f32x16& a00, f32x16& a01,
+                                     f32x16& a10, f32x16& a11,
+                                     float* bias_acc) {
+    int tid = threadIdx.x, wid = tid >> 6, lane = tid & 63;
+    int wm0 = (wid >> 1) * 64, wn0 = (wid & 1) * 64;
+    int r = lane & 31, kk2 = lane >> 5;
+    float ta[16], tb[16];
+    ldA(0, ta);
+    ldB(0, tb);
+    auto wr = [&](float* As, float* Bs) {
+#pragma unroll
+        for (int u = 0; u < 16; ++u) {
+            int e = u * 256 + tid;
+            As[(e & 31) * (MT_M + 4) + (e >> 5)] = ta[u];
+            Bs[(e >> 7) * (MT_N + 4) + (e & 127)] = tb[u];
+        }
+    };
+    wr(As0, Bs0);
+    __syncthreads();
+    for (int ch = 0; ch < nch; ++ch) {
+        float* As = (ch & 1) ? As1 : As0;
+        float* Bs = (ch & 1) ? Bs1 : Bs0;
+        bool more = ch + 1 < nch;
+        if (more) {
+            ldA((ch + 1) * MT_K, ta);
+            ldB((ch + 1) * MT_K, tb);
+        }
+#pragma unroll
+        for (int ks = 0; ks < MT_K; ks += 2) {
+            float a0 = As[(ks + kk2) * (MT_M + 4) + wm0 + r];
+            float a1 = As[(ks + kk2) * (MT_M + 4) + wm0 + 32 + r];
+            float b0 = Bs[(ks + kk2) * (MT_N + 4) + wn0 + r];
+            float b1 = Bs[(ks + kk2) * (MT_N + 4) + wn0 + 32 + r];
+            a00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, a00, 0, 0, 0);
+            a01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, a01, 0, 0, 0);
+            a10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, a10, 0, 0, 0);
+            a11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, a11, 0, 0, 0);
+            if (bias_acc && wm0 == 0) {
+                bias_acc[0] += b0;
+                bias_acc[1] += b1;
+            }
+        }
+        if (more) wr((ch & 1) ? As0 : As1, (ch & 1) ? Bs0 : Bs1);
+        __syncthreads();
+    }
+}
+
+#define MFMA_LDS_DECL \
+    __shared__ float As2[2][MT_K * (MT_M + 4)]; \
+    __shared__ float Bs2[2][MT_K * (MT_N + 4)]
+
 // C[B,out] = act(X[B,in1 (++ concat in2)] @ Wt[in,out] + bias)
 __global__ void __launch_bounds__(256, 1)
 k_mfma_fwd(const float* __restrict__ x1, const float* __restrict__ x2,
            const float* __restrict__ wt, const float* __restrict__ bias,
            float* __restrict__ y, int B, int in1, int in2, int out,
            int act_kind) {
-    __shared__ float As[MT_K][MT_M + 4];
-    __shared__ float Bs[MT_K][MT_N + 4];
+    MFMA_LDS_DECL;
     int in_total = in1 + in2;
     int ntn = (out + MT_N - 1) / MT_N;
     int m0 = (blockIdx.x / ntn) * MT_M, n0 = (blockIdx.x % ntn) * MT_N;
-    int tid = threadIdx.x, wid = tid >> 6, lane = tid & 63;
-    int wm0 = (wid >> 1) * 64, wn0 = (wid & 1) * 64;
-    int r = lane & 31, kk2 = lane >> 5;
+    int tid = threadIdx.x, lane = tid & 63;
     f32x16 acc00 = {}, acc01 = {}, acc10 = {}, acc11 = {};
-    for (int k0 = 0; k0 < in_total; k0 += MT_K) {
-        for (int e = tid; e < MT_M * MT_K; e += 256) {
-            int mm = e / MT_K, kk = e % MT_K;
-            int gm = m0 + mm, gk = k0 + kk;
+    auto ldA = [&](int k0, float* t) {
+#pragma unroll
+        for (int u = 0; u < 16; ++u) {
+            int e = u * 256 + tid;
+            int gm = m0 + (e >> 5), gk = k0 + (e & 31);
             float v = 0.f;
             if (gm < B && gk < in_total)
                 v = (gk < in1) ? x1[(long)gm * in1 + gk]
                                : x2[(long)gm * in2 + (gk - in1)];
-            As[kk][mm] = v;
+            t[u] = v;
         }
-        for (int e = tid; e < MT_K * MT_N; e += 256) {
-            int kk = e / MT_N, nn = e % MT_N;
-            int gk = k0 + kk, gn = n0 + nn;
-            Bs[kk][nn] = (gk < in_total && gn < out)
+    };
+    auto ldB = [&](int k0, float* t) {
+#pragma unroll
+        for (int u = 0; u < 16; ++u) {
+            int e = u * 256 + tid;
+            int gk = k0 + (e >> 7), gn = n0 + (e & 127);
+            t[u] = (gk < in_total && gn < out)
                 ? wt[(long)gk * out + gn] : 0.f;
         }
-        __syncthreads();
-#pragma unroll
-        for (int ks = 0; ks < MT_K; ks += 2) {
-            float a0 = As[ks + kk2][wm0 + r];
-            float a1 = As[ks + kk2][wm0 + 32 + r];
-            float b0 = Bs[ks + kk2][wn0 + r];
-            float b1 = Bs[ks + kk2][wn0 + 32 + r];
-            acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
-            acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
-            acc10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
-            acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);
-        }
-        __syncthreads();
-    }
+    };
+    int nch = (in_total + MT_K - 1) / MT_K;
+    mfma_pipeline(ldA, ldB, nch, As2[0], As2[1], Bs2[0], Bs2[1],
+                  acc00, acc01, acc10, acc11, nullptr);
+    int wid = tid >> 6;
+    int wm0 = (wid >> 1) * 64, wn0 = (wid & 1) * 64;
     const f32x16* accs[4] = {&acc00, &acc01, &acc10, &acc11};
 #pragma unroll
     for (int t = 0; t < 4; ++t) {
@@ -1090,45 +1189,88 @@ __global__ void __launch_bounds__(256, 1)
 k_mfma_dx(const float* __restrict__ dz, const float* __restrict__ wt,
           const float* __restrict__ hprev, float* __restrict__ dx,
           int B, int in_lo, int in_hi, int out, int prev_act) {
-    __shared__ float As[MT_K][MT_M + 4];
-    __shared__ float Bs[MT_K][MT_N + 4];
+    MFMA_LDS_DECL;
     int span = in_hi - in_lo;
     int ntn = (span + MT_N - 1) / MT_N;
     int m0 = (blockIdx.x / ntn) * MT_M;
     int n0 = (blockIdx.x % ntn) * MT_N;             // relative i tile
-    int tid = threadIdx.x, wid = tid >> 6, lane = tid & 63;
-    int wm0 = (wid >> 1) * 64, wn0 = (wid & 1) * 64;
-    int r = lane & 31, kk2 = lane >> 5;
+    int tid = threadIdx.x, lane = tid & 63;
     f32x16 acc00 = {}, acc01 = {}, acc10 = {}, acc11 = {};
-    for (int k0 = 0; k0 < out; k0 += MT_K) {
-        for (int e = tid; e < MT_M * MT_K; e += 256) {
-            int mm = e / MT_K, kk = e % MT_K;
-            int gm = m0 + mm, gk = k0 + kk;
-            As[kk][mm] = (gm < B && gk < out)
-                ? dz[(long)gm * out + gk] : 0.f;
+    auto ldA = [&](int k0, float* t) {
+#pragma unroll
+        for (int u = 0; u < 16; ++u) {
+            int e = u * 256 + tid;
+            int gm = m0 + (e >> 5), gk = k0 + (e & 31);
+            t[u] = (gm < B && gk < out) ? dz[(long)gm * out + gk] : 0.f;
         }
-        // B tile = Wt^T: Bs[k(=o)][n(=i rel)] = wt[i][o]; o fastest on the
-        // read side for coalescing
-        for (int e = tid; e < MT_N * MT_K; e += 256) {
-            int nn = e / MT_K, kk = e % MT_K;
+    };
+    // B tile = Wt^T: t holds Bs[k(=o)][n(=i rel)] = wt[i][o]; the load
+    // geometry keeps 32 consecutive o per thread-sweep (128 B segments)
+    auto ldB = [&](int k0, float* t) {
+#pragma unroll
+        for (int u = 0; u < 16; ++u) {
+            int e = u * 256 + tid;
+            int nn = e >> 5, kk = e & 31;         // [n][k] load order
             int gi = in_lo + n0 + nn, gk = k0 + kk;
-            Bs[kk][nn] = (gi < in_hi && gk < out)
+            t[u] = (gi < in_hi && gk < out)
                 ? wt[(long)gi * out + gk] : 0.f;
         }
-        __syncthreads();
+    };
+    // ldB writes must land at Bs[kk][nn] but wr() assumes [k][n] from the
+    // (e>>7, e&127) mapping; so remap via a custom pipeline here:
+    {
+        float ta[16], tb[16];
+        ldA(0, ta);
+        ldB(0, tb);
+        auto wrA = [&](float* As) {
 #pragma unroll
-        for (int ks = 0; ks < MT_K; ks += 2) {
-            float a0 = As[ks + kk2][wm0 + r];
-            float a1 = As[ks + kk2][wm0 + 32 + r];
-            float b0 = Bs[ks + kk2][wn0 + r];
-            float b1 = Bs[ks + kk2][wn0 + 32 + r];
-            acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
-            acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
-            acc10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
-            acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);
-        }
+            for (int u = 0; u < 16; ++u) {
+                int e = u * 256 + tid;
+                As[(e & 31) * (MT_M + 4) + (e >> 5)] = ta[u];
+            }
+        };
+        auto wrB = [&](float* Bs) {
+#pragma unroll
+            for (int u = 0; u < 16; ++u) {
+                int e = u * 256 + tid;
+                Bs[(e & 31) * (MT_N + 4) + (e >> 5)] = tb[u];
+            }
+        };
+        wrA(As2[0]);
+        wrB(Bs2[0]);
         __syncthreads();
+        int wid = tid >> 6;
+        int wm0 = (wid >> 1) * 64, wn0 = (wid & 1) * 64;
+        int r = lane & 31, kk2 = lane >> 5;
+        int nch = (out + MT_K - 1) / MT_K;
+        for (int ch = 0; ch < nch; ++ch) {
+            float* As = As2[ch & 1];
+            float* Bs = Bs2[ch & 1];
+            bool more = ch + 1 < nch;
+            if (more) {
+                ldA((ch + 1) * MT_K, ta);
+                ldB((ch + 1) * MT_K, tb);
+            }
+#pragma unroll
+            for (int ks = 0; ks < MT_K; ks += 2) {
+                float a0 = As[(ks + kk2) * (MT_M + 4) + wm0 + r];
+                float a1 = As[(ks + kk2) * (MT_M + 4) + wm0 + 32 + r];
+                float b0 = Bs[(ks + kk2) * (MT_N + 4) + wn0 + r];
+                float b1 = Bs[(ks + kk2) * (MT_N + 4) + wn0 + 32 + r];
+                acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
+                acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
+                acc10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
+                acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);
+            }
+            if (more) {
+                wrA(As2[(ch + 1) & 1]);
+                wrB(Bs2[(ch + 1) & 1]);
+            }
+            __syncthreads();
+        }
     }
+    int wid = tid >> 6;
+    int wm0 = (wid >> 1) * 64, wn0 = (wid & 1) * 64;
     const f32x16* accs[4] = {&acc00, &acc01, &acc10, &acc11};
 #pragma unroll
     for (int t = 0; t < 4; ++t) {
@@ -1148,52 +1290,110 @@ k_mfma_dx(const float* __restrict__ dz, const float* __restrict__ wt,
     }
 }
 
-// dWt[in,out] = X^T[in,B] @ dz[B,out]  (K dimension = batch)
+// dWt[in,out] = X^T[in,B] @ dz[B,out]  (K dimension = batch).  Split-K
+// over `ksplit` segments (grid = nti*nto*ksplit) with fp32 atomicAdd
+// epilogue when ksplit > 1 (dwt/db must be pre-zeroed by the caller in
+// that case); bias db[o] = sum_b dz[b][o] accumulates for free out of the
+// B-fragment during the MFMA loop (wm0==0 waves of the m0==0 tiles).
 __global__ void __launch_bounds__(256, 1)
 k_mfma_dw(const float* __restrict__ dz, const float* __restrict__ x1,
           const float* __restrict__ x2, float* __restrict__ dwt,
-          int B, int in1, int in2, int out) {
-    __shared__ float As[MT_K][MT_M + 4];
-    __shared__ float Bs[MT_K][MT_N + 4];
+          float* __restrict__ db, int B, int in1, int in2, int out,
+          int ksplit) {
+    MFMA_LDS_DECL;
     int in_total = in1 + in2;
     int ntn = (out + MT_N - 1) / MT_N;
-    int m0 = (blockIdx.x / ntn) * MT_M;                // i tile
-    int n0 = (blockIdx.x % ntn) * MT_N;                // o tile
-    int tid = threadIdx.x, wid = tid >> 6, lane = tid & 63;
-    int wm0 = (wid >> 1) * 64, wn0 = (wid & 1) * 64;
-    int r = lane & 31, kk2 = lane >> 5;
+    int nti = (in_total + MT_M - 1) / MT_M;
+    int seg = blockIdx.x / (nti * ntn);
+    int tile = blockIdx.x % (nti * ntn);
+    int m0 = (tile / ntn) * MT_M;                      // i tile
+    int n0 = (tile % ntn) * MT_N;                      // o tile
+    int nch_total = (B + MT_K - 1) / MT_K;
+    int nch_seg = (nch_total + ksplit - 1) / ksplit;
+    int ch0 = seg * nch_seg;
+    int nch = min(nch_seg, nch_total - ch0);
+    if (nch <= 0) return;
+    long b_base = (long)ch0 * MT_K;
+    int tid = threadIdx.x, lane = tid & 63;
     f32x16 acc00 = {}, acc01 = {}, acc10 = {}, acc11 = {};
-    for (int k0 = 0; k0 < B; k0 += MT_K) {
-        // As[k(=b)][m(=i)] = X[b][i], i fastest on the read side
-        for (int e = tid; e < MT_K * MT_M; e += 256) {
-            int kk = e / MT_M, mm = e % MT_M;
-            int gb = k0 + kk, gi = m0 + mm;
+    // As[k(=b)][m(=i)] = X[b][i], i fastest on the read side
+    auto ldA = [&](int k0, float* t) {
+#pragma unroll
+        for (int u = 0; u < 16; ++u) {
+            int e = u * 256 + tid;
+            long gb = b_base + k0 + (e >> 7);
+            int gi = m0 + (e & 127);
             float v = 0.f;
             if (gb < B && gi < in_total)
-                v = (gi < in1) ? x1[(long)gb * in1 + gi]
-                               : x2[(long)gb * in2 + (gi - in1)];
-            As[kk][mm] = v;
+                v = (gi < in1) ? x1[gb * in1 + gi]
+                               : x2[gb * in2 + (gi - in1)];
+            t[u] = v;
         }
-        for (int e = tid; e < MT_K * MT_N; e += 256) {
-            int kk = e / MT_N, nn = e % MT_N;
-            int gb = k0 + kk, go = n0 + nn;
-            Bs[kk][nn] = (gb < B && go < out)
-                ? dz[(long)gb * out + go] : 0.f;
-        }
-        __syncthreads();
+    };
+    auto ldB = [&](int k0, float* t) {
 #pragma unroll
-        for (int ks = 0; ks < MT_K; ks += 2) {
-            float a0 = As[ks + kk2][wm0 + r];
-            float a1 = As[ks + kk2][wm0 + 32 + r];
-            float b0 = Bs[ks + kk2][wn0 + r];
-            float b1 = Bs[ks + kk2][wn0 + 32 + r];
-            acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
-            acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
-            acc10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
-            acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);
+        for (int u = 0; u < 16; ++u) {
+            int e = u * 256 + tid;
+            long gb = b_base + k0 + (e >> 7);
+            int go = n0 + (e & 127);
+            t[u] = (gb < B && go < out) ? dz[gb * out + go] : 0.f;
         }
+    };
+    // custom pipeline (A write geometry differs from mfma_pipeline's)
+    float bias0 = 0.f, bias1 = 0.f;
+    {
+        int wid = tid >> 6;
+        int wm0 = (wid >> 1) * 64, wn0 = (wid & 1) * 64;
+        int r = lane & 31, kk2 = lane >> 5;
+        float ta[16], tb[16];
+        ldA(0, ta);
+        ldB(0, tb);
+        auto wrA = [&](float* As) {
+#pragma unroll
+            for (int u = 0; u < 16; ++u) {
+                int e = u * 256 + tid;
+                As[(e >> 7) * (MT_M + 4) + (e & 127)] = ta[u];
+            }
+        };
+        auto wrB = [&](float* Bs) {
+#pragma unroll
+            for (int u = 0; u < 16; ++u) {
+                int e = u * 256 + tid;
+                Bs[(e >> 7) * (MT_N + 4) + (e & 127)] = tb[u];
+            }
+        };
+        wrA(As2[0]);
+        wrB(Bs2[0]);
         __syncthreads();
+        for (int ch = 0; ch < nch; ++ch) {
+            float* As = As2[ch & 1];
+            float* Bs = Bs2[ch & 1];
+            bool more = ch + 1 < nch;
+            if (more) {
+                ldA((ch + 1) * MT_K, ta);
+                ldB((ch + 1) * MT_K, tb);
+            }
+#pragma unroll
+            for (int ks = 0; ks < MT_K; ks += 2) {
+                float a0 = As[(ks + kk2) * (MT_M + 4) + wm0 + r];
+                float a1 = As[(ks + kk2) * (MT_M + 4) + wm0 + 32 + r];
+                float b0 = Bs[(ks + kk2) * (MT_N + 4) + wn0 + r];
+                float b1 = Bs[(ks + kk2) * (MT_N + 4) + wn0 + 32 + r];
+                acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
+                acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
+                acc10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
+                acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);
+                if (wm0 == 0) { bias0 += b0; bias1 += b1; }
+            }
+            if (more) {
+                wrA(As2[(ch + 1) & 1]);
+                wrB(Bs2[(ch + 1) & 1]);
+            }
+            __syncthreads();
+        }
     }
+    int wid = tid >> 6;
+    int wm0 = (wid >> 1) * 64, wn0 = (wid & 1) * 64;
     const f32x16* accs[4] = {&acc00, &acc01, &acc10, &acc11};
 #pragma unroll
     for (int t = 0; t < 4; ++t) {
@@ -1204,21 +1404,30 @@ k_mfma_dw(const float* __restrict__ dz, const float* __restrict__ x1,
             int col = lane & 31;
             int gi = m0 + wm0 + i * 32 + row;
             int go = n0 + wn0 + j * 32 + col;
-            if (gi < in_total && go < out)
-                dwt[(long)gi * out + go] = (*accs[t])[reg];
+            if (gi < in_total && go < out) {
+                if (ksplit > 1)
+                    atomicAdd(&dwt[(long)gi * out + go], (*accs[t])[reg]);
+                else
+                    dwt[(long)gi * out + go] = (*accs[t])[reg];
+            }
         }
     }
-}
-
-// db[o] = sum_b dz[b][o] — column reduce, one thread per column with a
-// grid-stride batch loop (coalesced across o).
-__global__ void k_bias_reduce(const float* __restrict__ dz,
-                              float* __restrict__ db, int B, int out) {
-    int o = blockIdx.x * blockDim.x + threadIdx.x;
-    if (o >= out) return;
-    float acc = 0.f;
-    for (int b = 0; b < B; ++b) acc += dz[(long)b * out + o];
-    db[o] = acc;
+    if (db && wm0 == 0 && m0 == 0) {
+        // combine the two k-parity halves (lane l and l^32 share a column)
+        bias0 += __shfl_xor(bias0, 32, 64);
+        bias1 += __shfl_xor(bias1, 32, 64);
+        int r = lane & 31;
+        if ((lane >> 5) == 0) {
+            int go0 = n0 + wn0 + r, go1 = n0 + wn0 + 32 + r;
+            if (ksplit > 1) {
+                if (go0 < out) atomicAdd(&db[go0], bias0);
+                if (go1 < out) atomicAdd(&db[go1], bias1);
+            } else {
+                if (go0 < out) db[go0] = bias0;
+                if (go1 < out) db[go1] = bias1;
+            }
+        }
+    }
 }
 
 // ===========================================================================
@@ -2312,13 +2521,19 @@ public:
         if (cfg.batch >= 512 && l.out >= 64 && in_total >= 32) {
             if (want_dw) {
                 int ntm = ceil_div(in_total, MT_M), ntn = ceil_div(l.out, MT_N);
-                hipLaunchKernelGGL(k_mfma_dw, dim3(ntm * ntn), dim3(256),
-                                   0, stream, dz, x1, x2, gslab + l.w_off,
-                                   cfg.batch, l.in1, l.in2, l.out);
-                hipLaunchKernelGGL(k_bias_reduce,
-                                   dim3(ceil_div(l.out, 256)), dim3(256),
-                                   0, stream, dz, gslab + l.b_off,
-                                   cfg.batch, l.out);
+                // split K (= batch) until the grid covers the chip
+                int ksplit = 1;
+                while (ntm * ntn * ksplit < 256 &&
+                       ksplit * 2 * MT_K <= cfg.batch)
+                    ksplit *= 2;
+                if (ksplit > 1)
+                    HIP_CHECK(hipMemsetAsync(
+                        gslab + l.w_off, 0,
+                        ((long)in_total * l.out + l.out) * 4, stream));
+                hipLaunchKernelGGL(k_mfma_dw, dim3(ntm * ntn * ksplit),
+                                   dim3(256), 0, stream, dz, x1, x2,
+                                   gslab + l.w_off, gslab + l.b_off,
+                                   cfg.batch, l.in1, l.in2, l.out, ksplit);
             }
             if (dx1) {
                 int ntm = ceil_div(cfg.batch, MT_M);
@@ -2645,10 +2860,25 @@ public:
         hipLaunchKernelGGL(k_soft_update, dim3(256), dim3(256), 0, stream,
                            p_actor_t, p_actor, anet.n_params,
                            p_critic_t, p_critic, cnet.n_params, cfg.tau);
-        // P34: PER priority write-back
-        hipLaunchKernelGGL(k_per_update, dim3(1), dim3(256), 0, stream,
-                           sum_tree, min_tree, tree_cap, bidx, pri, B,
-                           cfg.per_alpha, cnt);
+        // P34: PER priority write-back.  Large batches use the per-level
+        // grid-wide repair (the one-wg level-synced kernel serializes).
+        if (B >= 512) {
+            hipLaunchKernelGGL(k_per_leaves, dim3(ceil_div(B, 256)),
+                               dim3(256), 0, stream, sum_tree, min_tree,
+                               tree_cap, bidx, pri, B, cfg.per_alpha, cnt);
+            long levels = 0;
+            for (long c = tree_cap; c > 1; c >>= 1) ++levels;
+            for (long lv = 0; lv < levels; ++lv)
+                hipLaunchKernelGGL(k_per_level, dim3(ceil_div(B, 256)),
+                                   dim3(256), 0, stream, sum_tree, min_tree,
+                                   tree_cap, bidx, B, lv);
+            hipLaunchKernelGGL(k_tick_end, dim3(1), dim3(64), 0, stream,
+                               cnt);
+        } else {
+            hipLaunchKernelGGL(k_per_update, dim3(1), dim3(256), 0, stream,
+                               sum_tree, min_tree, tree_cap, bidx, pri, B,
+                               cfg.per_alpha, cnt);
+        }
     }
 
     void step(int n) {
