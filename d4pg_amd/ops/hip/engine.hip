@@ -1962,7 +1962,11 @@ __device__ inline void p_sample(const PStepArgs& g) {
         double mass = (double)u01(r.v[0]) * total;
         long node = 1;
         while (node < g.tree_cap) {
-            double ls = g.sum_tree[2 * node];
+            // deep tree levels are random-access, once-per-step: bypass L2
+            // for them, keep the hot top of the tree cached
+            double ls = (node >= 16384)
+                ? __builtin_nontemporal_load(&g.sum_tree[2 * node])
+                : g.sum_tree[2 * node];
             if (mass > ls) { mass -= ls; node = 2 * node + 1; }
             else           { node = 2 * node; }
         }
@@ -1977,12 +1981,17 @@ __device__ inline void p_sample(const PStepArgs& g) {
         g.bd[probe] = g.rd[idx];
     }
     idx = __shfl(idx, 0, 64);
+    // replay rows are random and never re-read — keep them out of L2
+    // (the weights need it; SURVEY hot-loop note)
     for (int k = lane; k < g.O; k += 64) {
-        g.bs[(long)probe * g.O + k] = g.rs[idx * g.O + k];
-        g.bs2[(long)probe * g.O + k] = g.rs2[idx * g.O + k];
+        g.bs[(long)probe * g.O + k] =
+            __builtin_nontemporal_load(&g.rs[idx * g.O + k]);
+        g.bs2[(long)probe * g.O + k] =
+            __builtin_nontemporal_load(&g.rs2[idx * g.O + k]);
     }
     for (int k = lane; k < g.A; k += 64)
-        g.ba[(long)probe * g.A + k] = g.ra[idx * g.A + k];
+        g.ba[(long)probe * g.A + k] =
+            __builtin_nontemporal_load(&g.ra[idx * g.A + k]);
 }
 
 // C51 projection phase (k_project semantics; one wave per row, LDS row)
@@ -2009,6 +2018,76 @@ __device__ inline void p_project(const PStepArgs& g, float* lds) {
     if (row < g.B)
         for (int k = lane; k < K; k += 64)
             g.m_proj[(long)row * K + k] = mrow[k];
+}
+
+// merged projection + CE-grad phase: the wave that projects row b's target
+// distribution keeps it in LDS and computes the CE gradient/priority
+// immediately (saves a barrier and the m_proj global round-trip; q comes
+// from PH4).  m_proj is still written out for introspection/parity tests.
+__device__ inline void p_project_ce(const PStepArgs& g, float* lds) {
+    int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    int row = blockIdx.x * 4 + wid;
+    int K = g.K;
+    float* mrow = lds + wid * 64;
+    for (int k = lane; k < K; k += 64) mrow[k] = 0.f;
+    __builtin_amdgcn_wave_barrier();
+    if (row < g.B && lane < K) {
+        float delta = (g.v_max - g.v_min) / (K - 1);
+        float z = g.v_min + lane * delta;
+        float tz = g.br[row] + g.gamma_n * (1.f - g.bd[row]) * z;
+        tz = fminf(g.v_max, fmaxf(g.v_min, tz));
+        float b = (tz - g.v_min) / delta;
+        int l = (int)floorf(b), u = (int)ceilf(b);
+        if (l == u) { if (u > 0) l -= 1; else u += 1; }
+        float p = g.p_t[(long)row * K + lane];
+        atomicAdd(&mrow[l], p * ((float)u - b));
+        atomicAdd(&mrow[u], p * (b - (float)l));
+    }
+    __builtin_amdgcn_wave_barrier();
+    if (row >= g.B) return;
+    float qv = 0.f, mv = 0.f;
+    if (lane < K) {
+        g.m_proj[(long)row * K + lane] = mrow[lane];
+        qv = g.q[(long)row * K + lane];
+        mv = mrow[lane];
+    }
+    float dot = mv * qv;
+    float ce = -mv * __logf(qv + 1e-10f);
+    for (int s = 32; s > 0; s >>= 1) {
+        dot += __shfl_xor(dot, s, 64);
+        ce += __shfl_xor(ce, s, 64);
+    }
+    float scale = (g.is_weighting && g.bw) ? g.bw[row] : 1.f;
+    if (lane < K)
+        g.dlog[(long)row * K + lane] = scale * (qv - mv) / (float)g.B;
+    if (lane == 0) {
+        g.pri[row] = dot + g.per_eps;
+        atomicAdd(&g.cnt->loss_critic, scale * ce / (float)g.B);
+    }
+}
+
+// narrow backward-dX (span <= 8, e.g. the concat action slice): one wave
+// per batch row, lane-parallel dot over `out` with a shfl reduce, tanh
+// mask applied from ymask.
+__device__ inline void p_bwd_dx_narrow(const PStepArgs& g, const float* dz,
+                                       const float* wt, int in_lo,
+                                       int in_hi, int out,
+                                       const float* ymask, float* dx) {
+    int row = blockIdx.x * 4 + (threadIdx.x >> 6);
+    int lane = threadIdx.x & 63;
+    if (row >= g.B) return;
+    int span = in_hi - in_lo;
+    for (int j = 0; j < span; ++j) {
+        const float* wrow = wt + (long)(in_lo + j) * out;
+        float acc = 0.f;
+        for (int o = lane; o < out; o += 64)
+            acc += dz[(long)row * out + o] * wrow[o];
+        for (int s = 32; s > 0; s >>= 1) acc += __shfl_xor(acc, s, 64);
+        if (lane == 0) {
+            float y = ymask[(long)row * span + j];
+            dx[(long)row * span + j] = acc * (1.f - y * y);
+        }
+    }
 }
 
 __device__ inline void p_ce_grad(const PStepArgs& g) {
@@ -2109,7 +2188,7 @@ k_step_persistent(PStepArgs g, int nsteps) {
     const int O = g.O, A = g.A, H = g.H, K = g.K, B = g.B;
 
     for (int s = 0; s < nsteps; ++s) {
-        PTIME(g, s, 0);
+               PTIME(g, s, 0);
         // PH0: PER sample + batch gather
         p_sample(g);
         p_bar(ctr, tgt); PTIME(g, s, 1);
@@ -2171,21 +2250,19 @@ k_step_persistent(PStepArgs g, int nsteps) {
         p_fwd(lds, g.ct_h3, nullptr, ct.w4, ct.b4, g.p_t, B, H, 0, K,
               ACT_SOFTMAX, wg, PNWG);
         p_bar(ctr, tgt); PTIME(g, s, 8);
-        // PH8: C51 projection; PH9: CE grad + priorities
-        p_project(g, lds);
+        // PH8: C51 projection + CE grad + priorities (fused)
+        p_project_ce(g, lds);
         p_bar(ctr, tgt); PTIME(g, s, 9);
-        p_ce_grad(g);
-        p_bar(ctr, tgt); PTIME(g, s, 10);
         // PH10-12: critic dX chain (pre-update weights)
         p_bwd_dx(lds, g.dlog, c.w4, 0, H, K, B, g.c_h3, ACT_RELU, g.d3,
                  wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 11);
+        p_bar(ctr, tgt); PTIME(g, s, 10);
         p_bwd_dx(lds, g.d3, c.w3, 0, H, H, B, g.c_h2, ACT_RELU, g.d2,
                  wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 12);
+        p_bar(ctr, tgt); PTIME(g, s, 11);
         p_bwd_dx(lds, g.d2, c.w2, 0, H, H, B, g.c_h1, ACT_RELU, g.d1,
                  wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 13);
+        p_bar(ctr, tgt); PTIME(g, s, 12);
         // PH13: critic dW, 4 jobs split by tile count (l1 is the biggest)
         if (wg < 4)
             p_dw2(lds, g.d1, g.bs, nullptr, g.g_critic + g.cl[0].w_off,
@@ -2199,49 +2276,48 @@ k_step_persistent(PStepArgs g, int nsteps) {
         else
             p_dw2(lds, g.dlog, g.c_h3, nullptr, g.g_critic + g.cl[3].w_off,
                  g.g_critic + g.cl[3].b_off, B, H, 0, K, wg - 56, 8);
-        p_bar(ctr, tgt); PTIME(g, s, 14);
+        p_bar(ctr, tgt); PTIME(g, s, 13);
         // PH14: Adam + target soft-update, critic
         p_adam_lerp(g.p_critic, g.g_critic, g.m_critic, g.v_critic,
                     g.p_critic_t, g.n_critic, g.lr_critic, g.tau,
                     g.cnt->adam_t_critic);
-        p_bar(ctr, tgt); PTIME(g, s, 15);
+        p_bar(ctr, tgt); PTIME(g, s, 14);
         // PH15-18: critic(s, a_out) with UPDATED critic params
         p_fwd(lds, g.bs, nullptr, c.w1, c.b1, g.pc_h1, B, O, 0, H,
               ACT_RELU, wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 16);
+        p_bar(ctr, tgt); PTIME(g, s, 15);
         p_fwd(lds, g.pc_h1, g.a_out, c.w2, c.b2, g.pc_h2, B, H, A, H,
               ACT_RELU, wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 17);
+        p_bar(ctr, tgt); PTIME(g, s, 16);
         p_fwd(lds, g.pc_h2, nullptr, c.w3, c.b3, g.pc_h3, B, H, 0, H,
               ACT_RELU, wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 18);
+        p_bar(ctr, tgt); PTIME(g, s, 17);
         p_fwd(lds, g.pc_h3, nullptr, c.w4, c.b4, g.pq, B, H, 0, K,
               ACT_SOFTMAX, wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 19);
+        p_bar(ctr, tgt); PTIME(g, s, 18);
         // PH19: policy head gradient
         p_policy_grad(g);
-        p_bar(ctr, tgt); PTIME(g, s, 20);
+        p_bar(ctr, tgt); PTIME(g, s, 19);
         // PH20-22: dX back through the critic, ending at the tanh-masked
         // action slice (adz)
         p_bwd_dx(lds, g.pd3, c.w4, 0, H, K, B, g.pc_h3, ACT_RELU, g.pd2,
                  wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 21);
+        p_bar(ctr, tgt); PTIME(g, s, 20);
         p_bwd_dx(lds, g.pd2, c.w3, 0, H, H, B, g.pc_h2, ACT_RELU, g.pdh1,
                  wg, PNWG);
+        p_bar(ctr, tgt); PTIME(g, s, 21);
+        p_bwd_dx_narrow(g, g.pdh1, c.w2, H, H + A, H, g.a_out, g.adz);
         p_bar(ctr, tgt); PTIME(g, s, 22);
-        p_bwd_dx(lds, g.pdh1, c.w2, H, H + A, H, B, g.a_out, ACT_TANH,
-                 g.adz, wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 23);
         // PH23-25: dX through the actor (az3, az2, az1)
         p_bwd_dx(lds, g.adz, a.w4, 0, H, A, B, g.pa_h3, ACT_RELU, g.az3,
                  wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 24);
+        p_bar(ctr, tgt); PTIME(g, s, 23);
         p_bwd_dx(lds, g.az3, a.w3, 0, H, H, B, nullptr, ACT_NONE, g.az2,
                  wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 25);
+        p_bar(ctr, tgt); PTIME(g, s, 24);
         p_bwd_dx(lds, g.az2, a.w2, 0, H, H, B, g.pa_h1, ACT_RELU, g.az1,
                  wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 26);
+        p_bar(ctr, tgt); PTIME(g, s, 25);
         // PH26: actor dW
         if (wg < 4)
             p_dw2(lds, g.az1, g.bs, nullptr, g.g_actor + g.al[0].w_off,
@@ -2255,15 +2331,15 @@ k_step_persistent(PStepArgs g, int nsteps) {
         else
             p_dw2(lds, g.adz, g.pa_h3, nullptr, g.g_actor + g.al[3].w_off,
                  g.g_actor + g.al[3].b_off, B, H, 0, A, wg - 56, 8);
-        p_bar(ctr, tgt); PTIME(g, s, 27);
+        p_bar(ctr, tgt); PTIME(g, s, 26);
         // PH27: Adam + soft-update, actor
         p_adam_lerp(g.p_actor, g.g_actor, g.m_actor, g.v_actor,
                     g.p_actor_t, g.n_actor, g.lr_actor, g.tau,
                     g.cnt->adam_t_actor);
-        p_bar(ctr, tgt); PTIME(g, s, 28);
+        p_bar(ctr, tgt); PTIME(g, s, 27);
         // PH28: PER priority write-back + counter tick
         p_per_update(g);
-        p_bar(ctr, tgt); PTIME(g, s, 29);
+        p_bar(ctr, tgt); PTIME(g, s, 28);
     }
     if (wg == 0 && threadIdx.x == 0) g.gbar[160] = tgt;
 }

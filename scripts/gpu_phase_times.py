@@ -10,11 +10,11 @@ from scripts.gpu_microbench import make_engine  # noqa: E402
 
 PHASES = [
     "PH0 sample", "PH1 L1x4", "PH2 L2x3", "PH3 L3x3", "PH4 headsx3",
-    "PH5 ct.L2", "PH6 ct.L3", "PH7 ct.L4sm", "PH8 project", "PH9 ce_grad",
-    "PH10 c.dX4", "PH11 c.dX3", "PH12 c.dX2", "PH13 c.dW", "PH14 c.adam",
-    "PH15 pc.L1", "PH16 pc.L2", "PH17 pc.L3", "PH18 pc.L4sm", "PH19 pgrad",
-    "PH20 p.dX4", "PH21 p.dX3", "PH22 p.dX2a", "PH23 a.dX4", "PH24 a.dX3",
-    "PH25 a.dX2", "PH26 a.dW", "PH27 a.adam", "PH28 per_upd",
+    "PH5 ct.L2", "PH6 ct.L3", "PH7 ct.L4sm", "PH8 proj+ce",
+    "PH9 c.dX4", "PH10 c.dX3", "PH11 c.dX2", "PH12 c.dW", "PH13 c.adam",
+    "PH14 pc.L1", "PH15 pc.L2", "PH16 pc.L3", "PH17 pc.L4sm", "PH18 pgrad",
+    "PH19 p.dX4", "PH20 p.dX3", "PH21 p.dX2a", "PH22 a.dX4", "PH23 a.dX3",
+    "PH24 a.dX2", "PH25 a.dW", "PH26 a.adam", "PH27 per_upd",
 ]
 
 eng = make_engine()
