@@ -2131,8 +2131,9 @@ __device__ inline void p_policy_grad(const PStepArgs& g) {
 }
 
 // PER write-back + schedule-counter tick (k_per_update semantics; wg 0 only)
-__device__ inline void p_per_update(const PStepArgs& g) {
-    if (blockIdx.x != 0) return;
+__device__ inline void p_per_update(const PStepArgs& g, bool do_tick = true,
+                                    int owner_wg = 0) {
+    if ((int)blockIdx.x != owner_wg) return;
     int tid = threadIdx.x;
     float local_max = 0.f;
     for (int i = tid; i < g.B; i += 256) {
@@ -2167,12 +2168,428 @@ __device__ inline void p_per_update(const PStepArgs& g) {
         }
         __syncthreads();
     }
-    if (tid == 0) {
+    if (do_tick && tid == 0) {
         g.cnt->beta_t += 1;
         g.cnt->adam_t_actor += 1;
         g.cnt->adam_t_critic += 1;
         g.cnt->rng_epoch += 1;
     }
+}
+
+// ===========================================================================
+// Chain-fused persistent step (flagship path v2)
+// ===========================================================================
+// Forward layers and backward-dX layers are ROW-LOCAL: y[b] depends only on
+// x[b].  So a workgroup that owns PR complete rows chains through an entire
+// network (and back) with only __syncthreads — no grid barriers between
+// layers.  The step collapses to ~10 grid-barrier phases:
+//   sample | {actor_t, critic, actor} fwd chains | critic_t chain+proj+CE |
+//   critic dX chain | critic dW | critic Adam | policy megachain (fwd +
+//   policy grad + dX back through critic AND actor, with the PER tree
+//   write-back overlapped on a spare workgroup) | actor dW | actor Adam |
+//   tick.
+// Activations stay in LDS between layers ([PR][PXMAX] ping-pong rows);
+// weights stream through double-buffered 64x64 LDS chunks with register-
+// batched preloads (same staging discipline as p_fwd).
+#define PR 4
+// LDS pool reuse during chain phases: XA | XB | WS0 | WS1
+#define PX_A 0
+#define PX_B (PR * PXMAX)
+#define PW_0 (2 * PR * PXMAX)
+#define PW_1 (2 * PR * PXMAX + 64 * 65)
+
+// stage PR rows of src[B][width] into x_l rows at column offset `off`,
+// zero-padding columns [off+width, pad_to)
+__device__ inline void p_rows_load(float* x_l, const float* src, long r0,
+                                   int B, int width, int off, int pad_to) {
+    int tid = threadIdx.x;
+    int tot = PR * width;
+    for (int base = 0; base < tot; base += 256 * 8) {
+        float tmp[8];
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+            int e = base + u * 256 + tid;
+            float v = 0.f;
+            if (e < tot) {
+                int rr = e / width, kk = e % width;
+                if (r0 + rr < B) v = src[(r0 + rr) * width + kk];
+            }
+            tmp[u] = v;
+        }
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+            int e = base + u * 256 + tid;
+            if (e < tot)
+                x_l[(e / width) * PXMAX + off + (e % width)] = tmp[u];
+        }
+    }
+    for (int e = tid; e < PR * (pad_to - off - width); e += 256) {
+        int span = pad_to - off - width;
+        x_l[(e / span) * PXMAX + off + width + (e % span)] = 0.f;
+    }
+}
+
+// one dense layer for a PR-row group: y_l = act(x_l @ W + b), LDS-resident;
+// optionally mirrors y to global ysave[B][out].  Columns beyond `out` are
+// zero-padded to the next 64 multiple so the next layer can run full-64
+// K chunks.
+__device__ __noinline__ void p_layer_rows(float* lds, const float* x_l,
+                                          float* y_l, const float* wt,
+                                          const float* bias, int in_total,
+                                          int out, int act_kind,
+                                          float* ysave, long r0, int B) {
+    int tid = threadIdx.x, rq = tid >> 6, c = tid & 63;
+    int kk16 = tid >> 6, cc16 = tid & 63;
+    int nk = (in_total + 63) >> 6;
+    int nc = (out + 63) >> 6;
+    int total_ch = nc * nk;
+    float* ws0 = lds + PW_0;
+    float* ws1 = lds + PW_1;
+    float wreg[16];
+    auto preload = [&](int t) {
+        int c0 = (t / nk) << 6, kc = (t % nk) << 6;
+#pragma unroll
+        for (int u = 0; u < 16; ++u) {
+            int gk = kc + kk16 + 4 * u, gc = c0 + cc16;
+            wreg[u] = (gk < in_total && gc < out)
+                ? wt[(long)gk * out + gc] : 0.f;
+        }
+    };
+    preload(0);
+    float acc = 0.f;
+    const float* xr = x_l + rq * PXMAX;
+    for (int t = 0; t < total_ch; ++t) {
+        int ci = t / nk, ki = t % nk;
+        float* ws = (t & 1) ? ws1 : ws0;
+        float wb[16];
+#pragma unroll
+        for (int u = 0; u < 16; ++u) wb[u] = wreg[u];
+        if (t + 1 < total_ch) preload(t + 1);
+#pragma unroll
+        for (int u = 0; u < 16; ++u)
+            ws[(kk16 + 4 * u) * 65 + cc16] = wb[u];
+        __syncthreads();
+        if (ki == 0) acc = 0.f;
+        int kc = ki << 6;
+#pragma unroll 8
+        for (int k = 0; k < 64; ++k)
+            acc += xr[kc + k] * ws[k * 65 + c];
+        if (ki == nk - 1) {
+            int o = (ci << 6) + c;
+            float v = acc;
+            bool live = (o < out) && (r0 + rq < B);
+            if (o < out) {
+                v += bias[o];
+                if (act_kind == ACT_RELU) v = fmaxf(v, 0.f);
+                else if (act_kind == ACT_TANH) v = tanhf(v);
+            }
+            if (act_kind == ACT_SOFTMAX) {
+                // out <= 64 => single col chunk; wave == one row
+                float lv = (o < out) ? v : -INFINITY;
+                float mx = lv;
+                for (int s = 32; s > 0; s >>= 1)
+                    mx = fmaxf(mx, __shfl_xor(mx, s, 64));
+                float e = (o < out) ? __expf(lv - mx) : 0.f;
+                float sum = e;
+                for (int s = 32; s > 0; s >>= 1)
+                    sum += __shfl_xor(sum, s, 64);
+                v = (sum > 0.f) ? e / sum : 0.f;
+            }
+            int pad_to = nc << 6;
+            if (o < pad_to)
+                y_l[rq * PXMAX + o] = (o < out) ? v : 0.f;
+            if (live && ysave) ysave[(r0 + rq) * out + o] = v;
+        }
+        __syncthreads();
+    }
+}
+
+// backward-dX for a PR-row group: dx_l = (dz_l @ W^T) * act'(h), LDS ->
+// LDS, h/dx mirrors read/written against global [B][span] arrays.
+__device__ __noinline__ void p_dx_rows(float* lds, const float* dz_l,
+                                       float* dx_l, const float* wt,
+                                       int in_lo, int in_hi, int out,
+                                       const float* hsave, int prev_act,
+                                       float* dxsave, long r0, int B) {
+    int tid = threadIdx.x, rq = tid >> 6, c = tid & 63;
+    int oo16 = tid & 3, ii16 = tid >> 2;
+    int span = in_hi - in_lo;
+    int no = (out + 63) >> 6;
+    int ni = (span + 63) >> 6;
+    int total_ch = ni * no;
+    float* ws0 = lds + PW_0;
+    float* ws1 = lds + PW_1;
+    float wreg[16];
+    auto preload = [&](int t) {
+        int i0 = (t / no) << 6, oc = (t % no) << 6;
+        int gi = in_lo + i0 + ii16;
+#pragma unroll
+        for (int u = 0; u < 16; ++u) {
+            int go = oc + oo16 + 4 * u;
+            wreg[u] = (gi < in_hi && go < out)
+                ? wt[(long)gi * out + go] : 0.f;
+        }
+    };
+    preload(0);
+    float acc = 0.f;
+    const float* zr = dz_l + rq * PXMAX;
+    for (int t = 0; t < total_ch; ++t) {
+        int ii = t / no, oi = t % no;
+        float* ws = (t & 1) ? ws1 : ws0;
+        float wb[16];
+#pragma unroll
+        for (int u = 0; u < 16; ++u) wb[u] = wreg[u];
+        if (t + 1 < total_ch) preload(t + 1);
+#pragma unroll
+        for (int u = 0; u < 16; ++u)
+            ws[(oo16 + 4 * u) * 65 + ii16] = wb[u];
+        __syncthreads();
+        if (oi == 0) acc = 0.f;
+        int oc = oi << 6;
+#pragma unroll 8
+        for (int o = 0; o < 64; ++o)
+            acc += zr[oc + o] * ws[o * 65 + c];
+        if (oi == no - 1) {
+            int i = (ii << 6) + c;
+            int pad_to = ni << 6;
+            bool live = (i < span) && (r0 + rq < B);
+            float v = acc;
+            if (live && hsave)
+                v *= act_mask(prev_act, hsave[(r0 + rq) * span + i]);
+            if (i < pad_to)
+                dx_l[rq * PXMAX + i] = (i < span) ? v : 0.f;
+            if (live && dxsave) dxsave[(r0 + rq) * span + i] = v;
+        }
+        __syncthreads();
+    }
+}
+
+// projection + CE for PR LDS-resident p_t rows (wave rq owns row r0+rq;
+// scratch m rows in m_l)
+__device__ inline void p_proj_ce_rows(const PStepArgs& g, const float* pt_l,
+                                      float* m_l, long r0) {
+    int rq = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    long row = r0 + rq;
+    int K = g.K;
+    float* mrow = m_l + rq * PXMAX;
+    mrow[lane] = 0.f;
+    __builtin_amdgcn_wave_barrier();
+    bool live = row < g.B;
+    if (live && lane < K) {
+        float delta = (g.v_max - g.v_min) / (K - 1);
+        float z = g.v_min + lane * delta;
+        float tz = g.br[row] + g.gamma_n * (1.f - g.bd[row]) * z;
+        tz = fminf(g.v_max, fmaxf(g.v_min, tz));
+        float b = (tz - g.v_min) / delta;
+        int l = (int)floorf(b), u = (int)ceilf(b);
+        if (l == u) { if (u > 0) l -= 1; else u += 1; }
+        float p = pt_l[rq * PXMAX + lane];
+        atomicAdd(&mrow[l], p * ((float)u - b));
+        atomicAdd(&mrow[u], p * (b - (float)l));
+    }
+    __builtin_amdgcn_wave_barrier();
+    if (!live) return;
+    float mv = (lane < K) ? mrow[lane] : 0.f;
+    float qv = (lane < K) ? g.q[row * K + lane] : 0.f;
+    if (lane < K) g.m_proj[row * K + lane] = mv;
+    float dot = mv * qv;
+    float ce = -mv * __logf(qv + 1e-10f);
+    for (int s = 32; s > 0; s >>= 1) {
+        dot += __shfl_xor(dot, s, 64);
+        ce += __shfl_xor(ce, s, 64);
+    }
+    float scale = (g.is_weighting && g.bw) ? g.bw[row] : 1.f;
+    if (lane < K)
+        g.dlog[row * K + lane] = scale * (qv - mv) / (float)g.B;
+    if (lane == 0) {
+        g.pri[row] = dot + g.per_eps;
+        atomicAdd(&g.cnt->loss_critic, scale * ce / (float)g.B);
+    }
+}
+
+// policy-head gradient for PR LDS-resident pq rows -> pd rows (zeroed pad)
+__device__ inline void p_pgrad_rows(const PStepArgs& g, const float* pq_l,
+                                    float* pd_l, long r0) {
+    int rq = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    long row = r0 + rq;
+    int K = g.K;
+    bool live = row < g.B;
+    float delta = (g.v_max - g.v_min) / (K - 1);
+    float z = g.v_min + lane * delta;
+    float qv = (live && lane < K) ? pq_l[rq * PXMAX + lane] : 0.f;
+    float e = qv * z;
+    for (int s = 32; s > 0; s >>= 1) e += __shfl_xor(e, s, 64);
+    pd_l[rq * PXMAX + lane] = (live && lane < K)
+        ? -qv * (z - e) / (float)g.B : 0.f;
+    if (live && lane == 0)
+        atomicAdd(&g.cnt->loss_actor, -e / (float)g.B);
+}
+
+__global__ void __launch_bounds__(256, 1)
+k_step_chain(PStepArgs g, int nsteps) {
+    __shared__ float lds[PLDS_FLOATS];
+    unsigned long long tgt = g.gbar[160];
+    unsigned long long* ctr = g.gbar;
+    int wg = blockIdx.x;
+    NetPtrs a = net_ptrs(g.p_actor, g.al);
+    NetPtrs at = net_ptrs(g.p_actor_t, g.al);
+    NetPtrs c = net_ptrs(g.p_critic, g.cl);
+    NetPtrs ct = net_ptrs(g.p_critic_t, g.cl);
+    const int O = g.O, A = g.A, H = g.H, K = g.K, B = g.B;
+    const int ngr = (B + PR - 1) / PR;
+    float* XA = lds + PX_A;
+    float* XB = lds + PX_B;
+    const int padO = (O + 63) & ~63;
+    const int padHA = (H + A + 63) & ~63;
+
+    for (int s = 0; s < nsteps; ++s) {
+        PTIME(g, s, 32);
+        p_sample(g);
+        p_bar(ctr, tgt); PTIME(g, s, 33);
+        // C1: three independent forward chains
+        if (wg < ngr) {                        // actor_target(s2) -> a2
+            long r0 = (long)wg * PR;
+            p_rows_load(XA, g.bs2, r0, B, O, 0, padO);
+            p_layer_rows(lds, XA, XB, at.w1, at.b1, O, H, ACT_RELU,
+                         nullptr, r0, B);
+            p_layer_rows(lds, XB, XA, at.w2, at.b2, H, H, ACT_NONE,
+                         nullptr, r0, B);
+            p_layer_rows(lds, XA, XB, at.w3, at.b3, H, H, ACT_RELU,
+                         nullptr, r0, B);
+            p_layer_rows(lds, XB, XA, at.w4, at.b4, H, A, ACT_TANH,
+                         g.a2, r0, B);
+        } else if (wg < 2 * ngr) {             // critic(s, a) -> q
+            long r0 = (long)(wg - ngr) * PR;
+            p_rows_load(XA, g.bs, r0, B, O, 0, padO);
+            p_layer_rows(lds, XA, XB, c.w1, c.b1, O, H, ACT_RELU,
+                         g.c_h1, r0, B);
+            p_rows_load(XB, g.ba, r0, B, A, H, padHA);
+            p_layer_rows(lds, XB, XA, c.w2, c.b2, H + A, H, ACT_RELU,
+                         g.c_h2, r0, B);
+            p_layer_rows(lds, XA, XB, c.w3, c.b3, H, H, ACT_RELU,
+                         g.c_h3, r0, B);
+            p_layer_rows(lds, XB, XA, c.w4, c.b4, H, K, ACT_SOFTMAX,
+                         g.q, r0, B);
+        } else if (wg < 3 * ngr) {             // actor(s) -> a_out
+            long r0 = (long)(wg - 2 * ngr) * PR;
+            p_rows_load(XA, g.bs, r0, B, O, 0, padO);
+            p_layer_rows(lds, XA, XB, a.w1, a.b1, O, H, ACT_RELU,
+                         g.pa_h1, r0, B);
+            p_layer_rows(lds, XB, XA, a.w2, a.b2, H, H, ACT_NONE,
+                         g.pa_h2, r0, B);
+            p_layer_rows(lds, XA, XB, a.w3, a.b3, H, H, ACT_RELU,
+                         g.pa_h3, r0, B);
+            p_layer_rows(lds, XB, XA, a.w4, a.b4, H, A, ACT_TANH,
+                         g.a_out, r0, B);
+        }
+        p_bar(ctr, tgt); PTIME(g, s, 34);
+        // C2: critic_target(s2, a2) chain + projection + CE (row-local)
+        if (wg < ngr) {
+            long r0 = (long)wg * PR;
+            p_rows_load(XA, g.bs2, r0, B, O, 0, padO);
+            p_layer_rows(lds, XA, XB, ct.w1, ct.b1, O, H, ACT_RELU,
+                         nullptr, r0, B);
+            p_rows_load(XB, g.a2, r0, B, A, H, padHA);
+            p_layer_rows(lds, XB, XA, ct.w2, ct.b2, H + A, H, ACT_RELU,
+                         nullptr, r0, B);
+            p_layer_rows(lds, XA, XB, ct.w3, ct.b3, H, H, ACT_RELU,
+                         nullptr, r0, B);
+            p_layer_rows(lds, XB, XA, ct.w4, ct.b4, H, K, ACT_SOFTMAX,
+                         g.p_t, r0, B);
+            p_proj_ce_rows(g, XA, XB, r0);
+        }
+        p_bar(ctr, tgt); PTIME(g, s, 35);
+        // C3: critic backward-dX chain (pre-update weights)
+        if (wg < ngr) {
+            long r0 = (long)wg * PR;
+            p_rows_load(XA, g.dlog, r0, B, K, 0, 64);
+            p_dx_rows(lds, XA, XB, c.w4, 0, H, K, g.c_h3, ACT_RELU,
+                      g.d3, r0, B);
+            p_dx_rows(lds, XB, XA, c.w3, 0, H, H, g.c_h2, ACT_RELU,
+                      g.d2, r0, B);
+            p_dx_rows(lds, XA, XB, c.w2, 0, H, H, g.c_h1, ACT_RELU,
+                      g.d1, r0, B);
+        }
+        p_bar(ctr, tgt); PTIME(g, s, 36);
+        // C4: critic dW (whole grid)
+        if (wg < 4)
+            p_dw2(lds, g.d1, g.bs, nullptr, g.g_critic + g.cl[0].w_off,
+                  g.g_critic + g.cl[0].b_off, B, O, 0, H, wg, 4);
+        else if (wg < 32)
+            p_dw2(lds, g.d2, g.c_h1, g.ba, g.g_critic + g.cl[1].w_off,
+                  g.g_critic + g.cl[1].b_off, B, H, A, H, wg - 4, 28);
+        else if (wg < 56)
+            p_dw2(lds, g.d3, g.c_h2, nullptr, g.g_critic + g.cl[2].w_off,
+                  g.g_critic + g.cl[2].b_off, B, H, 0, H, wg - 32, 24);
+        else
+            p_dw2(lds, g.dlog, g.c_h3, nullptr, g.g_critic + g.cl[3].w_off,
+                  g.g_critic + g.cl[3].b_off, B, H, 0, K, wg - 56, 8);
+        p_bar(ctr, tgt); PTIME(g, s, 37);
+        // C5: critic Adam + target soft-update
+        p_adam_lerp(g.p_critic, g.g_critic, g.m_critic, g.v_critic,
+                    g.p_critic_t, g.n_critic, g.lr_critic, g.tau,
+                    g.cnt->adam_t_critic);
+        p_bar(ctr, tgt); PTIME(g, s, 38);
+        // C6: policy megachain (UPDATED critic), with the PER tree
+        // write-back overlapped on the last workgroup
+        if (wg < ngr) {
+            long r0 = (long)wg * PR;
+            p_rows_load(XA, g.bs, r0, B, O, 0, padO);
+            p_layer_rows(lds, XA, XB, c.w1, c.b1, O, H, ACT_RELU,
+                         g.pc_h1, r0, B);
+            p_rows_load(XB, g.a_out, r0, B, A, H, padHA);
+            p_layer_rows(lds, XB, XA, c.w2, c.b2, H + A, H, ACT_RELU,
+                         g.pc_h2, r0, B);
+            p_layer_rows(lds, XA, XB, c.w3, c.b3, H, H, ACT_RELU,
+                         g.pc_h3, r0, B);
+            p_layer_rows(lds, XB, XA, c.w4, c.b4, H, K, ACT_SOFTMAX,
+                         g.pq, r0, B);
+            p_pgrad_rows(g, XA, XB, r0);
+            p_dx_rows(lds, XB, XA, c.w4, 0, H, K, g.pc_h3, ACT_RELU,
+                      nullptr, r0, B);
+            p_dx_rows(lds, XA, XB, c.w3, 0, H, H, g.pc_h2, ACT_RELU,
+                      nullptr, r0, B);
+            p_dx_rows(lds, XB, XA, c.w2, H, H + A, H, g.a_out, ACT_TANH,
+                      g.adz, r0, B);
+            p_dx_rows(lds, XA, XB, a.w4, 0, H, A, g.pa_h3, ACT_RELU,
+                      g.az3, r0, B);
+            p_dx_rows(lds, XB, XA, a.w3, 0, H, H, nullptr, ACT_NONE,
+                      g.az2, r0, B);
+            p_dx_rows(lds, XA, XB, a.w2, 0, H, H, g.pa_h1, ACT_RELU,
+                      g.az1, r0, B);
+        } else if (wg == PNWG - 1) {
+            p_per_update(g, false, PNWG - 1);
+        }
+        p_bar(ctr, tgt); PTIME(g, s, 39);
+        // C7: actor dW
+        if (wg < 4)
+            p_dw2(lds, g.az1, g.bs, nullptr, g.g_actor + g.al[0].w_off,
+                  g.g_actor + g.al[0].b_off, B, O, 0, H, wg, 4);
+        else if (wg < 30)
+            p_dw2(lds, g.az2, g.pa_h1, nullptr, g.g_actor + g.al[1].w_off,
+                  g.g_actor + g.al[1].b_off, B, H, 0, H, wg - 4, 26);
+        else if (wg < 56)
+            p_dw2(lds, g.az3, g.pa_h2, nullptr, g.g_actor + g.al[2].w_off,
+                  g.g_actor + g.al[2].b_off, B, H, 0, H, wg - 30, 26);
+        else
+            p_dw2(lds, g.adz, g.pa_h3, nullptr, g.g_actor + g.al[3].w_off,
+                  g.g_actor + g.al[3].b_off, B, H, 0, A, wg - 56, 8);
+        p_bar(ctr, tgt); PTIME(g, s, 40);
+        // C8: actor Adam + soft-update
+        p_adam_lerp(g.p_actor, g.g_actor, g.m_actor, g.v_actor,
+                    g.p_actor_t, g.n_actor, g.lr_actor, g.tau,
+                    g.cnt->adam_t_actor);
+        p_bar(ctr, tgt); PTIME(g, s, 41);
+        // C9: schedule-counter tick (after both Adams read their t)
+        if (wg == 0 && threadIdx.x == 0) {
+            g.cnt->beta_t += 1;
+            g.cnt->adam_t_actor += 1;
+            g.cnt->adam_t_critic += 1;
+            g.cnt->rng_epoch += 1;
+        }
+        p_bar(ctr, tgt); PTIME(g, s, 42);
+    }
+    if (wg == 0 && threadIdx.x == 0) g.gbar[160] = tgt;
 }
 
 __global__ void __launch_bounds__(256, 1)
@@ -2697,10 +3114,21 @@ public:
         return g;
     }
 
+    bool use_chain() const {
+        // chain-fused variant: three parallel PR-row chains must fit the
+        // grid, H must be a 64-multiple (LDS row padding), heads <= a wave
+        return use_persistent() && cfg.hidden % 64 == 0 &&
+               3 * ceil_div(cfg.batch, PR) <= PNWG && cfg.act <= 64;
+    }
+
     void enqueue_persistent(int nsteps) {
         PStepArgs g = ps_args();
-        hipLaunchKernelGGL(k_step_persistent, dim3(PNWG), dim3(256), 0,
-                           stream, g, nsteps);
+        if (use_chain())
+            hipLaunchKernelGGL(k_step_chain, dim3(PNWG), dim3(256), 0,
+                               stream, g, nsteps);
+        else
+            hipLaunchKernelGGL(k_step_persistent, dim3(PNWG), dim3(256), 0,
+                               stream, g, nsteps);
     }
 
     bool use_row_block() const {
