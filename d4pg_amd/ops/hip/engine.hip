@@ -1862,17 +1862,28 @@ __device__ __noinline__ void p_dw2(float* lds, const float* dz,
         float acc[4][4] = {};
         float accb[4] = {};
         for (int bc = 0; bc < B; bc += 64) {
-            for (int e = tid; e < 4096; e += 256) {
+            // register-batched staging (all 32 loads in flight)
+            float tx[16], tz[16];
+#pragma unroll
+            for (int u = 0; u < 16; ++u) {
+                int e = u * 256 + tid;
                 int bb = e >> 6, ii = e & 63;
                 int gb = bc + bb, gi = i0 + ii;
                 float xv = 0.f;
                 if (gb < B && gi < in_total)
                     xv = (gi < in1) ? x1[(long)gb * in1 + gi]
                                     : x2[(long)gb * in2 + (gi - in1)];
-                xs[bb * 65 + ii] = xv;
+                tx[u] = xv;
                 int go = o0 + ii;
-                zs[bb * 65 + ii] = (gb < B && go < out)
+                tz[u] = (gb < B && go < out)
                     ? dz[(long)gb * out + go] : 0.f;
+            }
+#pragma unroll
+            for (int u = 0; u < 16; ++u) {
+                int e = u * 256 + tid;
+                int bb = e >> 6, ii = e & 63;
+                xs[bb * 65 + ii] = tx[u];
+                zs[bb * 65 + ii] = tz[u];
             }
             __syncthreads();
             int blim = min(64, B - bc);
