@@ -2256,45 +2256,29 @@ __device__ __noinline__ void p_layer_rows(float* lds, const float* x_l,
     int total_ch = nc * nk;
     float* ws0 = lds + PW_0;
     float* ws1 = lds + PW_1;
-    // prefetch depth 2: chunk t's registers were loaded at t-2, so the
-    // vmcnt wait at the LDS write has had two chunks of cover
-    float wregA[16], wregB[16];
-    auto preloadA = [&](int t) {
+    float wreg[16];
+    auto preload = [&](int t) {
         int c0 = (t / nk) << 6, kc = (t % nk) << 6;
 #pragma unroll
         for (int u = 0; u < 16; ++u) {
             int gk = kc + kk16 + 4 * u, gc = c0 + cc16;
-            wregA[u] = (gk < in_total && gc < out)
+            wreg[u] = (gk < in_total && gc < out)
                 ? wt[(long)gk * out + gc] : 0.f;
         }
     };
-    auto preloadB = [&](int t) {
-        int c0 = (t / nk) << 6, kc = (t % nk) << 6;
-#pragma unroll
-        for (int u = 0; u < 16; ++u) {
-            int gk = kc + kk16 + 4 * u, gc = c0 + cc16;
-            wregB[u] = (gk < in_total && gc < out)
-                ? wt[(long)gk * out + gc] : 0.f;
-        }
-    };
-    preloadA(0);
-    if (total_ch > 1) preloadB(1);
+    preload(0);
     float acc = 0.f;
     const float* xr = x_l + rq * PXMAX;
     for (int t = 0; t < total_ch; ++t) {
         int ci = t / nk, ki = t % nk;
         float* ws = (t & 1) ? ws1 : ws0;
-        if ((t & 1) == 0) {
+        float wb[16];
 #pragma unroll
-            for (int u = 0; u < 16; ++u)
-                ws[(kk16 + 4 * u) * 65 + cc16] = wregA[u];
-            if (t + 2 < total_ch) preloadA(t + 2);
-        } else {
+        for (int u = 0; u < 16; ++u) wb[u] = wreg[u];
+        if (t + 1 < total_ch) preload(t + 1);
 #pragma unroll
-            for (int u = 0; u < 16; ++u)
-                ws[(kk16 + 4 * u) * 65 + cc16] = wregB[u];
-            if (t + 2 < total_ch) preloadB(t + 2);
-        }
+        for (int u = 0; u < 16; ++u)
+            ws[(kk16 + 4 * u) * 65 + cc16] = wb[u];
         __syncthreads();
         if (ki == 0) acc = 0.f;
         int kc = ki << 6;
@@ -2346,45 +2330,30 @@ __device__ __noinline__ void p_dx_rows(float* lds, const float* dz_l,
     int total_ch = ni * no;
     float* ws0 = lds + PW_0;
     float* ws1 = lds + PW_1;
-    float wregA[16], wregB[16];
-    auto preloadA = [&](int t) {
+    float wreg[16];
+    auto preload = [&](int t) {
         int i0 = (t / no) << 6, oc = (t % no) << 6;
         int gi = in_lo + i0 + ii16;
 #pragma unroll
         for (int u = 0; u < 16; ++u) {
             int go = oc + oo16 + 4 * u;
-            wregA[u] = (gi < in_hi && go < out)
+            wreg[u] = (gi < in_hi && go < out)
                 ? wt[(long)gi * out + go] : 0.f;
         }
     };
-    auto preloadB = [&](int t) {
-        int i0 = (t / no) << 6, oc = (t % no) << 6;
-        int gi = in_lo + i0 + ii16;
-#pragma unroll
-        for (int u = 0; u < 16; ++u) {
-            int go = oc + oo16 + 4 * u;
-            wregB[u] = (gi < in_hi && go < out)
-                ? wt[(long)gi * out + go] : 0.f;
-        }
-    };
-    preloadA(0);
-    if (total_ch > 1) preloadB(1);
+    preload(0);
     float acc = 0.f;
     const float* zr = dz_l + rq * PXMAX;
     for (int t = 0; t < total_ch; ++t) {
         int ii = t / no, oi = t % no;
         float* ws = (t & 1) ? ws1 : ws0;
-        if ((t & 1) == 0) {
+        float wb[16];
 #pragma unroll
-            for (int u = 0; u < 16; ++u)
-                ws[(oo16 + 4 * u) * 65 + ii16] = wregA[u];
-            if (t + 2 < total_ch) preloadA(t + 2);
-        } else {
+        for (int u = 0; u < 16; ++u) wb[u] = wreg[u];
+        if (t + 1 < total_ch) preload(t + 1);
 #pragma unroll
-            for (int u = 0; u < 16; ++u)
-                ws[(oo16 + 4 * u) * 65 + ii16] = wregB[u];
-            if (t + 2 < total_ch) preloadB(t + 2);
-        }
+        for (int u = 0; u < 16; ++u)
+            ws[(oo16 + 4 * u) * 65 + ii16] = wb[u];
         __syncthreads();
         if (oi == 0) acc = 0.f;
         int oc = oi << 6;
