@@ -23,6 +23,7 @@
 #include <cstdint>
 #include <cstdio>
 #include <cstring>
+#include <cstdlib>
 #include <cmath>
 #include <vector>
 #include <stdexcept>
@@ -3127,7 +3128,12 @@ public:
 
     bool use_chain() const {
         // chain-fused variant: three parallel PR-row chains must fit the
-        // grid, H must be a 64-multiple (LDS row padding), heads <= a wave
+        // grid, H must be a 64-multiple (LDS row padding), heads <= a wave.
+        // Measured SLOWER than the layer-parallel phase kernel (1596 vs
+        // 2274 steps/s at B=64 — the 10-layer serial policy megachain
+        // dominates), so opt-in via D4PG_CHAIN=1 for further tuning.
+        const char* e = getenv("D4PG_CHAIN");
+        if (!e || e[0] != '1') return false;
         return use_persistent() && cfg.hidden % 64 == 0 &&
                3 * ceil_div(cfg.batch, PR) <= PNWG && cfg.act <= 64;
     }
