@@ -138,6 +138,31 @@ def measure_env_steps_per_sec(seed: int, n: int = 3000) -> float:
     return n / dt
 
 
+def measure_env_steps_vector(seed: int, m: int = 64, ticks: int = 1500):
+    """Aux metric: vectorized-actor env-steps/sec — M batched native
+    Pendulums + one [M,obs] policy forward per tick (the MI355X-native
+    actor mode, parallel/learner.py --vector_envs)."""
+    import numpy as np
+    from d4pg_amd.envs.vector import VectorPendulum
+    from d4pg_amd.models import actor
+    env = VectorPendulum(m, seed=seed)
+    net = actor(3, 1)
+    net.eval()
+    rng = np.random.default_rng(seed)
+    obs = env.reset()
+    with torch.no_grad():
+        t0 = time.perf_counter()
+        for t in range(ticks):
+            a = net(torch.from_numpy(obs)).numpy()
+            a = np.clip(a + 0.3 * rng.standard_normal(a.shape),
+                        -1, 1).astype(np.float32)
+            obs, r, done = env.step(a)
+            if done:
+                obs = env.reset()
+        dt = time.perf_counter() - t0
+    return ticks * m / dt
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -178,6 +203,7 @@ def main():
 
     elapsed = _max_over_ranks(dist, elapsed)
     env_sps = measure_env_steps_per_sec(seed=1234 + rank) if rank == 0 else 0.0
+    vec_sps = measure_env_steps_vector(seed=4321 + rank) if rank == 0 else 0.0
 
     if rank == 0:
         out = {
@@ -204,6 +230,7 @@ def main():
                        "replay_capacity": FLAGSHIP["capacity"],
                        "device": "cuda" if use_gpu else "cpu-fallback"},
             "env_steps_per_sec_1actor": env_sps,
+            "env_steps_per_sec_vector64": vec_sps,
         }
         print(json.dumps(out), flush=True)
     if dist is not None:
