@@ -117,8 +117,14 @@ class DDPG:
             self.replayBuffer = Replay(memory_size, env, n_steps=n_steps,
                                        gamma=gamma, rng=rng)
 
-        self._fused = None   # lazily-built HIP fused-step engine
+        self._fused = None   # lazily-built HIP fused-step engine bridge
         self.train_steps_done = 0
+
+    @property
+    def engine(self):
+        """The fused HIP engine bridge (backend='hip' after the first
+        train step), else None."""
+        return self._fused
 
     # ------------------------------------------------------------------
     # parameter plumbing (reference ddpg.py:92-120 parity)

@@ -179,7 +179,7 @@ class DistributedD4PG:
 
     def _broadcast_params(self):
         if self.is_learner:
-            if self.agent.backend == "hip":
+            if self.agent.backend == "hip" and self.agent.engine is not None:
                 self.agent.engine.sync_params_if_dirty()
             blob = torch.cat([pack_net(self.agent.actor),
                               torch.tensor([float(self.global_step)])])
@@ -263,7 +263,10 @@ class DistributedD4PG:
             return
         n = self.args.train_steps_per_cycle
         if self.agent.backend == "hip":
-            self.agent.engine.ddpg.replayBuffer.flush()
+            if self.agent.engine is None:
+                from ..ops import build_fused_engine
+                self.agent._fused = build_fused_engine(self.agent)
+            self.agent.replayBuffer.flush()
             self.agent.engine.engine.train_steps(n)
             self.agent.train_steps_done += n
         else:
