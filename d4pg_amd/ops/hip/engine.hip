@@ -1127,7 +1127,7 @@ __device__ inline void mfma_pipeline(FA ldA, FB ldB, int nch,
     __shared__ float Bs2[2][MT_K * (MT_N + 4)]
 
 // C[B,out] = act(X[B,in1 (++ concat in2)] @ Wt[in,out] + bias)
-__global__ void __launch_bounds__(256, 1)
+__global__ void __launch_bounds__(256, 2)
 k_mfma_fwd(const float* __restrict__ x1, const float* __restrict__ x2,
            const float* __restrict__ wt, const float* __restrict__ bias,
            float* __restrict__ y, int B, int in1, int in2, int out,
@@ -1186,7 +1186,7 @@ k_mfma_fwd(const float* __restrict__ x1, const float* __restrict__ x2,
 
 // dX[B, in_lo:in_hi] = (dz[B,out] @ Wt^T) * act'(hprev); hprev/dx are
 // [B][span] (span = in_hi - in_lo), hprev null => no mask.
-__global__ void __launch_bounds__(256, 1)
+__global__ void __launch_bounds__(256, 2)
 k_mfma_dx(const float* __restrict__ dz, const float* __restrict__ wt,
           const float* __restrict__ hprev, float* __restrict__ dx,
           int B, int in_lo, int in_hi, int out, int prev_act) {
@@ -1296,7 +1296,7 @@ k_mfma_dx(const float* __restrict__ dz, const float* __restrict__ wt,
 // epilogue when ksplit > 1 (dwt/db must be pre-zeroed by the caller in
 // that case); bias db[o] = sum_b dz[b][o] accumulates for free out of the
 // B-fragment during the MFMA loop (wm0==0 waves of the m0==0 tiles).
-__global__ void __launch_bounds__(256, 1)
+__global__ void __launch_bounds__(256, 2)
 k_mfma_dw(const float* __restrict__ dz, const float* __restrict__ x1,
           const float* __restrict__ x2, float* __restrict__ dwt,
           float* __restrict__ db, int B, int in1, int in2, int out,
@@ -1450,11 +1450,15 @@ k_mfma_dw(const float* __restrict__ dz, const float* __restrict__ x1,
 // between steps (the hipGraph path is only needed for the wide-batch
 // configs that use the per-layer kernels above).
 #define PNWG 64
-#define PROWS 16
+// 4-row x 64-col tiles: a 64x256x256 GEMM phase becomes 64 tiles — the
+// whole grid — with ~1 us of FMA per tile (16-row tiles measured ~6 us of
+// per-wave compute and left 48 workgroups idle; scripts/tile_bench.hip)
+#define PROWS 4
 #define PXMAX 512
-// pool: [x/dz stage 16*512][w chunk 64*65][pad]
 #define PWOFF (PROWS * PXMAX)
-#define PLDS_FLOATS (PROWS * PXMAX + 64 * 65 + 64)
+// pool sized for the chain variant's layout (2 activation row-buffers +
+// 2 weight chunks), which is the largest user
+#define PLDS_FLOATS (2 * 4 * PXMAX + 2 * 64 * 65 + 64)
 
 struct PStepArgs {
     int B, O, A, H, K;
@@ -1555,9 +1559,7 @@ __device__ __noinline__ void p_fwd(float* lds, const float* x1, const float* x2,
     int tid = threadIdx.x;
     for (int t = wg_rel; t < ntiles; t += nwg) {
         int r0 = (t / nct) * PROWS, c0 = (t % nct) * 64;
-        // x-stage, register-batched: fire ALL loads (independent), then
-        // write LDS.  A naive load->ds_write loop serializes one memory
-        // round-trip per 256-thread sweep (measured ~9 us per 16 KB).
+        // x-stage, register-batched (all loads in flight, then ds_writes)
         {
             int tot = PROWS * in_total;
             for (int base = 0; base < tot; base += 256 * 8) {
@@ -1586,19 +1588,15 @@ __device__ __noinline__ void p_fwd(float* lds, const float* x1, const float* x2,
         __syncthreads();
         int rq = tid >> 6, c = tid & 63;
         int o = c0 + c;
-        float acc0 = 0.f, acc1 = 0.f, acc2 = 0.f, acc3 = 0.f;
-        // Weight slice K-chunked through LDS: 64x64 chunks staged with
-        // register-batched coalesced loads, DOUBLE-BUFFERED so chunk k+1's
-        // global loads are in flight while chunk k's FMAs run.
+        int r = r0 + rq;
+        float acc = 0.f;
+        // weight slice K-chunked through LDS, register-batched + double-
+        // buffered (chunk k+1's loads fly during chunk k's FMAs)
         float* ws = lds + PWOFF;                 // [64][65]
-        const float* xr0 = lds + (rq + 0) * in_total;
-        const float* xr1 = lds + (rq + 4) * in_total;
-        const float* xr2 = lds + (rq + 8) * in_total;
-        const float* xr3 = lds + (rq + 12) * in_total;
-        int kk16 = tid >> 6;                     // chunk-load geometry:
-        int cc16 = tid & 63;                     // thread u covers row
-        float wreg[16];                          // kk16 + 4u, col cc16
-        int nfull = in_total >> 6;               // # of full 64-chunks
+        const float* xr = lds + rq * in_total;
+        int kk16 = tid >> 6, cc16 = tid & 63;
+        float wreg[16];
+        int nfull = in_total >> 6;
         auto preload = [&](int kc) {
 #pragma unroll
             for (int u = 0; u < 16; ++u) {
@@ -1620,15 +1618,11 @@ __device__ __noinline__ void p_fwd(float* lds, const float* x1, const float* x2,
             __syncthreads();
             if (o < out) {
 #pragma unroll 8
-                for (int k = 0; k < 64; ++k) {
-                    float wv = ws[k * 65 + c];
-                    acc0 += xr0[kc + k] * wv; acc1 += xr1[kc + k] * wv;
-                    acc2 += xr2[kc + k] * wv; acc3 += xr3[kc + k] * wv;
-                }
+                for (int k = 0; k < 64; ++k)
+                    acc += xr[kc + k] * ws[k * 65 + c];
             }
             __syncthreads();
         }
-        // K tail (in_total % 64): simple staging
         int kc = nfull << 6, klen = in_total - kc;
         if (klen > 0) {
             for (int e = tid; e < (klen << 6); e += 256) {
@@ -1639,42 +1633,26 @@ __device__ __noinline__ void p_fwd(float* lds, const float* x1, const float* x2,
             }
             __syncthreads();
             if (o < out) {
-                for (int k = 0; k < klen; ++k) {
-                    float wv = ws[k * 65 + c];
-                    acc0 += xr0[kc + k] * wv; acc1 += xr1[kc + k] * wv;
-                    acc2 += xr2[kc + k] * wv; acc3 += xr3[kc + k] * wv;
-                }
+                for (int k = 0; k < klen; ++k)
+                    acc += xr[kc + k] * ws[k * 65 + c];
             }
             __syncthreads();
         }
-        if (o < out) {
-            float bv = bias[o];
-            acc0 += bv; acc1 += bv; acc2 += bv; acc3 += bv;
-        }
-        float accs[4] = {acc0, acc1, acc2, acc3};
         if (act_kind == ACT_SOFTMAX) {
-#pragma unroll
-            for (int j = 0; j < 4; ++j) {
-                int r = r0 + rq + 4 * j;
-                float v = (o < out) ? accs[j] : -INFINITY;
-                float mx = v;
-                for (int s = 32; s > 0; s >>= 1)
-                    mx = fmaxf(mx, __shfl_xor(mx, s, 64));
-                float e = (o < out) ? __expf(v - mx) : 0.f;
-                float sum = e;
-                for (int s = 32; s > 0; s >>= 1)
-                    sum += __shfl_xor(sum, s, 64);
-                if (r < B && o < out) y[(long)r * out + o] = e / sum;
-            }
-        } else if (o < out) {
-#pragma unroll
-            for (int j = 0; j < 4; ++j) {
-                int r = r0 + rq + 4 * j;
-                float v = accs[j];
-                if (act_kind == ACT_RELU) v = fmaxf(v, 0.f);
-                else if (act_kind == ACT_TANH) v = tanhf(v);
-                if (r < B) y[(long)r * out + o] = v;
-            }
+            // out <= 64, single col-chunk; wave == one row
+            float v = (o < out) ? acc + bias[o] : -INFINITY;
+            float mx = v;
+            for (int s = 32; s > 0; s >>= 1)
+                mx = fmaxf(mx, __shfl_xor(mx, s, 64));
+            float e = (o < out) ? __expf(v - mx) : 0.f;
+            float sum = e;
+            for (int s = 32; s > 0; s >>= 1) sum += __shfl_xor(sum, s, 64);
+            if (r < B && o < out) y[(long)r * out + o] = e / sum;
+        } else if (r < B && o < out) {
+            float v = acc + bias[o];
+            if (act_kind == ACT_RELU) v = fmaxf(v, 0.f);
+            else if (act_kind == ACT_TANH) v = tanhf(v);
+            y[(long)r * out + o] = v;
         }
         __syncthreads();
     }
@@ -1694,7 +1672,7 @@ __device__ inline void p_bwd_dx(float* lds, const float* dz, const float* wt,
     int tid = threadIdx.x;
     for (int t = wg_rel; t < ntiles; t += nwg) {
         int r0 = (t / nit) * PROWS, i0 = in_lo + (t % nit) * 64;
-        // dz-stage, register-batched (see p_fwd)
+        // dz-stage, register-batched
         {
             int tot = PROWS * out;
             for (int base = 0; base < tot; base += 256 * 8) {
@@ -1720,19 +1698,15 @@ __device__ inline void p_bwd_dx(float* lds, const float* dz, const float* wt,
         __syncthreads();
         int rq = tid >> 6, c = tid & 63;
         int i = i0 + c;
-        float acc0 = 0.f, acc1 = 0.f, acc2 = 0.f, acc3 = 0.f;
-        // Weight tile O-chunked through LDS, transposed write (compute
-        // reads ws[o][i] conflict-free), register-batched + double-
-        // buffered like the forward path.  Load geometry: thread u covers
-        // (i row = cc16, o = oc + kk16 + 4u) — oo fastest => coalesced.
+        int r = r0 + rq;
+        float acc = 0.f;
+        // weight tile O-chunked through LDS, transposed write, double-
+        // buffered register-batched staging
         float* ws = lds + PWOFF;                 // [64][65] = [o][i]
-        const float* z0 = lds + (rq + 0) * out;
-        const float* z1 = lds + (rq + 4) * out;
-        const float* z2 = lds + (rq + 8) * out;
-        const float* z3 = lds + (rq + 12) * out;
-        int oo16 = tid & 3;                      // o sub-offset (0..3)
-        int ii16 = tid >> 2;                     // i row (0..63)
-        float wreg[16];                          // o = oc + oo16 + 4u
+        const float* zr = lds + rq * out;
+        int oo16 = tid & 3;
+        int ii16 = tid >> 2;
+        float wreg[16];
         int gi_ld = i0 + ii16;
         int nfull = out >> 6;
         auto preload = [&](int oc) {
@@ -1756,15 +1730,11 @@ __device__ inline void p_bwd_dx(float* lds, const float* dz, const float* wt,
             __syncthreads();
             if (i < in_hi) {
 #pragma unroll 8
-                for (int o = 0; o < 64; ++o) {
-                    float wv = ws[o * 65 + c];
-                    acc0 += z0[oc + o] * wv; acc1 += z1[oc + o] * wv;
-                    acc2 += z2[oc + o] * wv; acc3 += z3[oc + o] * wv;
-                }
+                for (int o = 0; o < 64; ++o)
+                    acc += zr[oc + o] * ws[o * 65 + c];
             }
             __syncthreads();
         }
-        // O tail
         int oc = nfull << 6, olen = out - oc;
         if (olen > 0) {
             for (int e = tid; e < 4096; e += 256) {
@@ -1775,25 +1745,15 @@ __device__ inline void p_bwd_dx(float* lds, const float* dz, const float* wt,
             }
             __syncthreads();
             if (i < in_hi) {
-                for (int o = 0; o < olen; ++o) {
-                    float wv = ws[o * 65 + c];
-                    acc0 += z0[oc + o] * wv; acc1 += z1[oc + o] * wv;
-                    acc2 += z2[oc + o] * wv; acc3 += z3[oc + o] * wv;
-                }
+                for (int o = 0; o < olen; ++o)
+                    acc += zr[oc + o] * ws[o * 65 + c];
             }
             __syncthreads();
         }
-        if (i < in_hi) {
-            float accs[4] = {acc0, acc1, acc2, acc3};
-#pragma unroll
-            for (int j = 0; j < 4; ++j) {
-                int r = r0 + rq + 4 * j;
-                if (r < B) {
-                    long rel = (long)r * span + (i - in_lo);
-                    float m_ = hprev ? act_mask(prev_act, hprev[rel]) : 1.f;
-                    dx[rel] = accs[j] * m_;
-                }
-            }
+        if (r < B && i < in_hi) {
+            long rel = (long)r * span + (i - in_lo);
+            float m_ = hprev ? act_mask(prev_act, hprev[rel]) : 1.f;
+            dx[rel] = acc * m_;
         }
         __syncthreads();
     }
@@ -2989,10 +2949,11 @@ public:
     }
 
     bool mfma_eligible(int in_total, int out, int act_kind) const {
-        // wide-batch GEMMs go to the matrix cores; the tiny heads and the
-        // wave-wide softmax stay on the per-layer VALU kernels
-        return cfg.batch >= 512 && out >= 64 && in_total >= 32 &&
-               act_kind != ACT_SOFTMAX;
+        // wide-batch GEMMs go to the matrix cores (small dims pad into
+        // the 128x128 tile — still far faster than the VALU fallback);
+        // only the wave-wide softmax head stays on k_fwd3
+        (void)out; (void)in_total;
+        return cfg.batch >= 512 && act_kind != ACT_SOFTMAX;
     }
 
     void launch_fwd(std::initializer_list<FwdJob> jobs) {
@@ -3023,7 +2984,7 @@ public:
                     float* dx1, float* dx2, const float* h1, int prev_act,
                     bool want_dw) {
         int in_total = l.in1 + l.in2;
-        if (cfg.batch >= 512 && l.out >= 64 && in_total >= 32) {
+        if (cfg.batch >= 512) {
             if (want_dw) {
                 int ntm = ceil_div(in_total, MT_M), ntn = ceil_div(l.out, MT_N);
                 // split K (= batch) until the grid covers the chip
