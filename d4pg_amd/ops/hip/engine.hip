@@ -1901,15 +1901,35 @@ __device__ inline void p_adam_lerp(float* __restrict__ p,
     const float b1 = 0.9f, b2 = 0.999f, eps = 1e-8f;
     float bc1 = 1.f - __powf(b1, (float)t);
     float bc2 = 1.f - __powf(b2, (float)t);
-    for (long i = (long)blockIdx.x * 256 + threadIdx.x; i < n;
-         i += (long)PNWG * 256) {
-        float gi = gr[i];
-        float mi = b1 * m[i] + (1.f - b1) * gi;
-        float vi = b2 * v[i] + (1.f - b2) * gi * gi;
-        m[i] = mi; v[i] = vi;
-        float pn = p[i] - lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
-        p[i] = pn;
-        tgt_slab[i] += tau * (pn - tgt_slab[i]);
+    // register-batched: fire all 5 streams' loads for 4 grid-strides at
+    // once (a serial load->compute->store loop pays one memory round-trip
+    // per element batch)
+    long stride = (long)PNWG * 256;
+    long i0 = (long)blockIdx.x * 256 + threadIdx.x;
+    for (long base = i0; base < n; base += stride * 4) {
+        float gi_[4], mi_[4], vi_[4], pi_[4], ti_[4];
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+            long i = base + u * stride;
+            bool ok = i < n;
+            gi_[u] = ok ? gr[i] : 0.f;
+            mi_[u] = ok ? m[i] : 0.f;
+            vi_[u] = ok ? v[i] : 0.f;
+            pi_[u] = ok ? p[i] : 0.f;
+            ti_[u] = ok ? tgt_slab[i] : 0.f;
+        }
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+            long i = base + u * stride;
+            if (i >= n) continue;
+            float gi = gi_[u];
+            float mi = b1 * mi_[u] + (1.f - b1) * gi;
+            float vi = b2 * vi_[u] + (1.f - b2) * gi * gi;
+            m[i] = mi; v[i] = vi;
+            float pn = pi_[u] - lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+            p[i] = pn;
+            tgt_slab[i] = ti_[u] + tau * (pn - ti_[u]);
+        }
     }
 }
 
@@ -2707,7 +2727,10 @@ k_step_persistent(PStepArgs g, int nsteps) {
         p_bwd_dx(lds, g.az2, a.w2, 0, H, H, B, g.pa_h1, ACT_RELU, g.az1,
                  wg, PNWG);
         p_bar(ctr, tgt); PTIME(g, s, 25);
-        // PH26: actor dW
+        // PH26: actor dW, with the PER tree write-back overlapped on the
+        // last workgroup (priorities have been final since the proj+CE
+        // phase; the counter tick moves to the final phase so both Adams
+        // still read this step's t)
         if (wg < 4)
             p_dw2(lds, g.az1, g.bs, nullptr, g.g_actor + g.al[0].w_off,
                  g.g_actor + g.al[0].b_off, B, O, 0, H, wg, 4);
@@ -2717,17 +2740,24 @@ k_step_persistent(PStepArgs g, int nsteps) {
         else if (wg < 56)
             p_dw2(lds, g.az3, g.pa_h2, nullptr, g.g_actor + g.al[2].w_off,
                  g.g_actor + g.al[2].b_off, B, H, 0, H, wg - 30, 26);
-        else
+        else if (wg < 63)
             p_dw2(lds, g.adz, g.pa_h3, nullptr, g.g_actor + g.al[3].w_off,
-                 g.g_actor + g.al[3].b_off, B, H, 0, A, wg - 56, 8);
+                 g.g_actor + g.al[3].b_off, B, H, 0, A, wg - 56, 7);
+        else
+            p_per_update(g, false, PNWG - 1);
         p_bar(ctr, tgt); PTIME(g, s, 26);
         // PH27: Adam + soft-update, actor
         p_adam_lerp(g.p_actor, g.g_actor, g.m_actor, g.v_actor,
                     g.p_actor_t, g.n_actor, g.lr_actor, g.tau,
                     g.cnt->adam_t_actor);
         p_bar(ctr, tgt); PTIME(g, s, 27);
-        // PH28: PER priority write-back + counter tick
-        p_per_update(g);
+        // PH28: schedule-counter tick (tree write-back ran in PH26)
+        if (wg == 0 && threadIdx.x == 0) {
+            g.cnt->beta_t += 1;
+            g.cnt->adam_t_actor += 1;
+            g.cnt->adam_t_critic += 1;
+            g.cnt->rng_epoch += 1;
+        }
         p_bar(ctr, tgt); PTIME(g, s, 28);
     }
     if (wg == 0 && threadIdx.x == 0) g.gbar[160] = tgt;
