@@ -1897,14 +1897,15 @@ __device__ inline void p_adam_lerp(float* __restrict__ p,
                                    float* __restrict__ m,
                                    float* __restrict__ v,
                                    float* __restrict__ tgt_slab, long n,
-                                   float lr, float tau, long long t) {
+                                   float lr, float tau, long long t,
+                                   int nwg = PNWG) {
     const float b1 = 0.9f, b2 = 0.999f, eps = 1e-8f;
     float bc1 = 1.f - __powf(b1, (float)t);
     float bc2 = 1.f - __powf(b2, (float)t);
     // register-batched: fire all 5 streams' loads for 4 grid-strides at
     // once (a serial load->compute->store loop pays one memory round-trip
     // per element batch)
-    long stride = (long)PNWG * 256;
+    long stride = (long)nwg * 256;
     long i0 = (long)blockIdx.x * 256 + threadIdx.x;
     for (long base = i0; base < n; base += stride * 4) {
         float gi_[4], mi_[4], vi_[4], pi_[4], ti_[4];
@@ -1934,22 +1935,27 @@ __device__ inline void p_adam_lerp(float* __restrict__ p,
 }
 
 // PER sample + gather phase (k_per_sample semantics; one wave per probe)
-__device__ inline void p_sample(const PStepArgs& g) {
-    int probe = blockIdx.x * 4 + (threadIdx.x >> 6);
+__device__ inline void p_sample(const PStepArgs& g, int wg0 = 0,
+                                int sched_off = 0) {
+    // sched_off = +1 when pre-sampling the NEXT step's batch before the
+    // counter tick (overlapped under the actor-Adam phase)
+    int probe = ((int)blockIdx.x - wg0) * 4 + (threadIdx.x >> 6);
     int lane = threadIdx.x & 63;
-    if (probe == 0 && lane == 0) {
+    if (sched_off == 0 && probe == 0 && lane == 0) {
         g.cnt->loss_critic = 0.f;
         g.cnt->loss_actor = 0.f;
     }
     if (probe >= g.B) return;
     long long n = g.cnt->size;
     double total = g.sum_tree[1];
-    float frac = fminf((float)((double)g.cnt->beta_t / g.per_beta_iters),
-                       1.0f);
+    float frac = fminf(
+        (float)((double)(g.cnt->beta_t + sched_off) / g.per_beta_iters),
+        1.0f);
     float beta = g.per_beta0 + frac * (1.0f - g.per_beta0);
     long idx;
     if (lane == 0) {
-        Philox4 r = philox4(g.seed, (uint64_t)g.cnt->rng_epoch,
+        Philox4 r = philox4(g.seed,
+                            (uint64_t)(g.cnt->rng_epoch + sched_off),
                             (uint64_t)probe);
         double mass = (double)u01(r.v[0]) * total;
         long node = 1;
@@ -2598,9 +2604,13 @@ k_step_persistent(PStepArgs g, int nsteps) {
 
     for (int s = 0; s < nsteps; ++s) {
                PTIME(g, s, 0);
-        // PH0: PER sample + batch gather
-        p_sample(g);
-        p_bar(ctr, tgt); PTIME(g, s, 1);
+        // PH0: PER sample + batch gather (only on a launch's first step —
+        // later steps were pre-sampled under the previous actor-Adam)
+        if (s == 0 || B > 64) {      // 16 overlap wgs cover 64 probes
+            p_sample(g);
+            p_bar(ctr, tgt);
+        }
+        PTIME(g, s, 1);
         // PH1: four independent L1s (16 wgs each)
         if (wg < 16)
             p_fwd(lds, g.bs2, nullptr, at.w1, at.b1, g.at_h1, B, O, 0, H,
@@ -2746,10 +2756,15 @@ k_step_persistent(PStepArgs g, int nsteps) {
         else
             p_per_update(g, false, PNWG - 1);
         p_bar(ctr, tgt); PTIME(g, s, 26);
-        // PH27: Adam + soft-update, actor
-        p_adam_lerp(g.p_actor, g.g_actor, g.m_actor, g.v_actor,
-                    g.p_actor_t, g.n_actor, g.lr_actor, g.tau,
-                    g.cnt->adam_t_actor);
+        // PH27: Adam + soft-update, actor (wgs 0-47) overlapped with the
+        // NEXT step's PER sample + gather (wgs 48-63; tree was repaired in
+        // PH26, schedule counters offset by the pending tick)
+        if (wg < 48)
+            p_adam_lerp(g.p_actor, g.g_actor, g.m_actor, g.v_actor,
+                        g.p_actor_t, g.n_actor, g.lr_actor, g.tau,
+                        g.cnt->adam_t_actor, 48);
+        else if (s + 1 < nsteps && B <= 64)
+            p_sample(g, 48, 1);
         p_bar(ctr, tgt); PTIME(g, s, 27);
         // PH28: schedule-counter tick (tree write-back ran in PH26)
         if (wg == 0 && threadIdx.x == 0) {
@@ -2757,6 +2772,10 @@ k_step_persistent(PStepArgs g, int nsteps) {
             g.cnt->adam_t_actor += 1;
             g.cnt->adam_t_critic += 1;
             g.cnt->rng_epoch += 1;
+            if (s + 1 < nsteps) {
+                g.cnt->loss_critic = 0.f;
+                g.cnt->loss_actor = 0.f;
+            }
         }
         p_bar(ctr, tgt); PTIME(g, s, 28);
     }
