@@ -8,7 +8,7 @@ second beta) and a per-process plain-int step count.
 
 This optimizer exists for the reference-parity shared-memory mode
 (parallel/hogwild.py) and for CPU tests.  The MI355X learner path instead
-uses the fused multi-tensor Adam HIP kernel (ops/hip/fused_step.hip, K8 in
+uses the fused multi-tensor Adam HIP kernel (ops/hip/engine.hip, K8 in
 SURVEY.md §2c) over a flat parameter slab.
 """
 

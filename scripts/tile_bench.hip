@@ -16,9 +16,10 @@
 #define IN 256
 #define OUT 256
 
-// variant 0: depth-1 prefetch with copy (current p_fwd)
+// variant 0: depth-1 prefetch with copy (16-row tile, 4 accs)
 // variant 1: depth-2 prefetch, two reg banks
 // variant 2: no LDS staging — per-lane strided global reads
+// variant 3: 4-row tile, single acc (the shipped p_fwd shape)
 template <int V>
 __global__ void __launch_bounds__(256, 1)
 k_tile(const float* __restrict__ wt, float* __restrict__ y,
@@ -88,6 +89,36 @@ k_tile(const float* __restrict__ wt, float* __restrict__ y,
                 }
                 __syncthreads();
             }
+        } else if (V == 3) {
+            // 4-row tile: wave rq owns row rq, single acc
+            float acc = 0.f;
+            const float* xr = xs + rq * IN;
+            float wreg[16];
+#pragma unroll
+            for (int u = 0; u < 16; ++u)
+                wreg[u] = wt[(long)(kk16 + 4 * u) * OUT + cc16];
+            for (int ch = 0; ch < 4; ++ch) {
+                int kc = ch << 6;
+                float wb[16];
+#pragma unroll
+                for (int u = 0; u < 16; ++u) wb[u] = wreg[u];
+                if (ch + 1 < 4) {
+#pragma unroll
+                    for (int u = 0; u < 16; ++u)
+                        wreg[u] = wt[(long)(kc + 64 + kk16 + 4 * u) * OUT
+                                     + cc16];
+                }
+                float* ws = (ch & 1) ? ws1 : ws0;
+#pragma unroll
+                for (int u = 0; u < 16; ++u)
+                    ws[(kk16 + 4 * u) * 65 + cc16] = wb[u];
+                __syncthreads();
+#pragma unroll 8
+                for (int k = 0; k < 64; ++k)
+                    acc += xr[kc + k] * ws[k * 65 + c];
+                __syncthreads();
+            }
+            acc0 = acc;
         } else {   // V == 1
             float wa[16], wb2[16];
 #pragma unroll
@@ -176,5 +207,6 @@ int main(int argc, char** argv) {
     run<1>("v1 depth2", wt, y, iters, 16, flag);
     run<2>("v2 direct strided", wt, y, iters, 16, flag);
     run<2>("v2 direct strided all-busy", wt, y, iters, 64, flag);
+    run<3>("v3 4-row single-acc", wt, y, iters, 64, flag);
     return 0;
 }
