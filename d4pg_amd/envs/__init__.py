@@ -56,12 +56,18 @@ def make(env_id: str, seed: int | None = None, normalize: bool = True,
 
 def obs_act_dims(env, her: bool = False):
     """Observation/action dims the way the reference derives them
-    (main.py:70-80): flat obs dim, or observation+goal concat under HER."""
+    (main.py:70-80): flat obs dim (observation+goal concat under HER), and
+    discrete-vs-continuous action detection (``Discrete.n`` vs
+    ``Box.shape``)."""
     if her:
         o = env.reset()
         obs_dim = int(np.asarray(o["observation"]).size
                       + np.asarray(o["desired_goal"]).size)
     else:
         obs_dim = int(np.prod(env.observation_space.shape))
-    act_dim = int(np.prod(env.action_space.shape))
+    space = env.action_space
+    if hasattr(space, "n") and not getattr(space, "shape", None):
+        act_dim = int(space.n)          # gym.spaces.Discrete
+    else:
+        act_dim = int(np.prod(space.shape))
     return obs_dim, act_dim
