@@ -1058,7 +1058,11 @@ __global__ void k_tree_build_level(double* sum_tree, double* min_tree,
 // summation order as the per-thread loops of the small-batch kernels.
 // Tile: 128x128 per workgroup (4 waves, each 64x64 = 2x2 of 32x32 MFMA
 // tiles), K staged through LDS in 32-deep slices.
-#define MT_M 128
+// M-tile 64 (not 128): the H=1024 layers' grids then cover the chip with
+// 2 workgroups per CU, so one workgroup's LDS staging hides behind the
+// other's MFMAs (at 128 the 256-wg grid left 1 wg/CU and the per-chunk
+// ds_write+barrier was exposed — ~50% MfmaUtil).
+#define MT_M 64
 #define MT_N 128
 #define MT_K 32
 
@@ -1076,19 +1080,22 @@ __device__ inline void mfma_pipeline(FA ldA, FB ldB, int nch,
                                      float* As0, float* As1,
                                      float* Bs0, float* Bs1,
                                      f32x16& a00, f32x16& a01,
-                                     f32x16& a10, f32x16& a11,
                                      float* bias_acc) {
     int tid = threadIdx.x, wid = tid >> 6, lane = tid & 63;
-    int wm0 = (wid >> 1) * 64, wn0 = (wid & 1) * 64;
+    int wm0 = (wid >> 1) * 32, wn0 = (wid & 1) * 64;
     int r = lane & 31, kk2 = lane >> 5;
-    float ta[16], tb[16];
+    float ta[8], tb[16];
     ldA(0, ta);
     ldB(0, tb);
     auto wr = [&](float* As, float* Bs) {
 #pragma unroll
-        for (int u = 0; u < 16; ++u) {
+        for (int u = 0; u < 8; ++u) {
             int e = u * 256 + tid;
             As[(e & 31) * (MT_M + 4) + (e >> 5)] = ta[u];
+        }
+#pragma unroll
+        for (int u = 0; u < 16; ++u) {
+            int e = u * 256 + tid;
             Bs[(e >> 7) * (MT_N + 4) + (e & 127)] = tb[u];
         }
     };
@@ -1105,13 +1112,10 @@ __device__ inline void mfma_pipeline(FA ldA, FB ldB, int nch,
 #pragma unroll
         for (int ks = 0; ks < MT_K; ks += 2) {
             float a0 = As[(ks + kk2) * (MT_M + 4) + wm0 + r];
-            float a1 = As[(ks + kk2) * (MT_M + 4) + wm0 + 32 + r];
             float b0 = Bs[(ks + kk2) * (MT_N + 4) + wn0 + r];
             float b1 = Bs[(ks + kk2) * (MT_N + 4) + wn0 + 32 + r];
             a00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, a00, 0, 0, 0);
             a01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, a01, 0, 0, 0);
-            a10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, a10, 0, 0, 0);
-            a11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, a11, 0, 0, 0);
             if (bias_acc && wm0 == 0) {
                 bias_acc[0] += b0;
                 bias_acc[1] += b1;
@@ -1137,10 +1141,10 @@ k_mfma_fwd(const float* __restrict__ x1, const float* __restrict__ x2,
     int ntn = (out + MT_N - 1) / MT_N;
     int m0 = (blockIdx.x / ntn) * MT_M, n0 = (blockIdx.x % ntn) * MT_N;
     int tid = threadIdx.x, lane = tid & 63;
-    f32x16 acc00 = {}, acc01 = {}, acc10 = {}, acc11 = {};
+    f32x16 acc00 = {}, acc01 = {};
     auto ldA = [&](int k0, float* t) {
 #pragma unroll
-        for (int u = 0; u < 16; ++u) {
+        for (int u = 0; u < 8; ++u) {
             int e = u * 256 + tid;
             int gm = m0 + (e >> 5), gk = k0 + (e & 31);
             float v = 0.f;
@@ -1161,13 +1165,13 @@ k_mfma_fwd(const float* __restrict__ x1, const float* __restrict__ x2,
     };
     int nch = (in_total + MT_K - 1) / MT_K;
     mfma_pipeline(ldA, ldB, nch, As2[0], As2[1], Bs2[0], Bs2[1],
-                  acc00, acc01, acc10, acc11, nullptr);
+                  acc00, acc01, nullptr);
     int wid = tid >> 6;
-    int wm0 = (wid >> 1) * 64, wn0 = (wid & 1) * 64;
-    const f32x16* accs[4] = {&acc00, &acc01, &acc10, &acc11};
+    int wm0 = (wid >> 1) * 32, wn0 = (wid & 1) * 64;
+    const f32x16* accs[2] = {&acc00, &acc01};
 #pragma unroll
-    for (int t = 0; t < 4; ++t) {
-        int i = t >> 1, j = t & 1;
+    for (int t = 0; t < 2; ++t) {
+        int i = 0, j = t;
 #pragma unroll
         for (int reg = 0; reg < 16; ++reg) {
             int row = (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
@@ -1196,10 +1200,10 @@ k_mfma_dx(const float* __restrict__ dz, const float* __restrict__ wt,
     int m0 = (blockIdx.x / ntn) * MT_M;
     int n0 = (blockIdx.x % ntn) * MT_N;             // relative i tile
     int tid = threadIdx.x, lane = tid & 63;
-    f32x16 acc00 = {}, acc01 = {}, acc10 = {}, acc11 = {};
+    f32x16 acc00 = {}, acc01 = {};
     auto ldA = [&](int k0, float* t) {
 #pragma unroll
-        for (int u = 0; u < 16; ++u) {
+        for (int u = 0; u < 8; ++u) {
             int e = u * 256 + tid;
             int gm = m0 + (e >> 5), gk = k0 + (e & 31);
             t[u] = (gm < B && gk < out) ? dz[(long)gm * out + gk] : 0.f;
@@ -1220,12 +1224,12 @@ k_mfma_dx(const float* __restrict__ dz, const float* __restrict__ wt,
     // ldB writes must land at Bs[kk][nn] but wr() assumes [k][n] from the
     // (e>>7, e&127) mapping; so remap via a custom pipeline here:
     {
-        float ta[16], tb[16];
+        float ta[8], tb[16];
         ldA(0, ta);
         ldB(0, tb);
         auto wrA = [&](float* As) {
 #pragma unroll
-            for (int u = 0; u < 16; ++u) {
+            for (int u = 0; u < 8; ++u) {
                 int e = u * 256 + tid;
                 As[(e & 31) * (MT_M + 4) + (e >> 5)] = ta[u];
             }
@@ -1241,7 +1245,7 @@ k_mfma_dx(const float* __restrict__ dz, const float* __restrict__ wt,
         wrB(Bs2[0]);
         __syncthreads();
         int wid = tid >> 6;
-        int wm0 = (wid >> 1) * 64, wn0 = (wid & 1) * 64;
+        int wm0 = (wid >> 1) * 32, wn0 = (wid & 1) * 64;
         int r = lane & 31, kk2 = lane >> 5;
         int nch = (out + MT_K - 1) / MT_K;
         for (int ch = 0; ch < nch; ++ch) {
@@ -1255,13 +1259,10 @@ k_mfma_dx(const float* __restrict__ dz, const float* __restrict__ wt,
 #pragma unroll
             for (int ks = 0; ks < MT_K; ks += 2) {
                 float a0 = As[(ks + kk2) * (MT_M + 4) + wm0 + r];
-                float a1 = As[(ks + kk2) * (MT_M + 4) + wm0 + 32 + r];
                 float b0 = Bs[(ks + kk2) * (MT_N + 4) + wn0 + r];
                 float b1 = Bs[(ks + kk2) * (MT_N + 4) + wn0 + 32 + r];
                 acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
                 acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
-                acc10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
-                acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);
             }
             if (more) {
                 wrA(As2[(ch + 1) & 1]);
@@ -1271,11 +1272,11 @@ k_mfma_dx(const float* __restrict__ dz, const float* __restrict__ wt,
         }
     }
     int wid = tid >> 6;
-    int wm0 = (wid >> 1) * 64, wn0 = (wid & 1) * 64;
-    const f32x16* accs[4] = {&acc00, &acc01, &acc10, &acc11};
+    int wm0 = (wid >> 1) * 32, wn0 = (wid & 1) * 64;
+    const f32x16* accs[2] = {&acc00, &acc01};
 #pragma unroll
-    for (int t = 0; t < 4; ++t) {
-        int i = t >> 1, j = t & 1;
+    for (int t = 0; t < 2; ++t) {
+        int i = 0, j = t;
 #pragma unroll
         for (int reg = 0; reg < 16; ++reg) {
             int row = (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
@@ -1316,14 +1317,14 @@ k_mfma_dw(const float* __restrict__ dz, const float* __restrict__ x1,
     if (nch <= 0) return;
     long b_base = (long)ch0 * MT_K;
     int tid = threadIdx.x, lane = tid & 63;
-    f32x16 acc00 = {}, acc01 = {}, acc10 = {}, acc11 = {};
+    f32x16 acc00 = {}, acc01 = {};
     // As[k(=b)][m(=i)] = X[b][i], i fastest on the read side
     auto ldA = [&](int k0, float* t) {
 #pragma unroll
-        for (int u = 0; u < 16; ++u) {
+        for (int u = 0; u < 8; ++u) {
             int e = u * 256 + tid;
-            long gb = b_base + k0 + (e >> 7);
-            int gi = m0 + (e & 127);
+            long gb = b_base + k0 + (e >> 6);
+            int gi = m0 + (e & 63);
             float v = 0.f;
             if (gb < B && gi < in_total)
                 v = (gi < in1) ? x1[gb * in1 + gi]
@@ -1344,16 +1345,16 @@ k_mfma_dw(const float* __restrict__ dz, const float* __restrict__ x1,
     float bias0 = 0.f, bias1 = 0.f;
     {
         int wid = tid >> 6;
-        int wm0 = (wid >> 1) * 64, wn0 = (wid & 1) * 64;
+        int wm0 = (wid >> 1) * 32, wn0 = (wid & 1) * 64;
         int r = lane & 31, kk2 = lane >> 5;
-        float ta[16], tb[16];
+        float ta[8], tb[16];
         ldA(0, ta);
         ldB(0, tb);
         auto wrA = [&](float* As) {
 #pragma unroll
-            for (int u = 0; u < 16; ++u) {
+            for (int u = 0; u < 8; ++u) {
                 int e = u * 256 + tid;
-                As[(e >> 7) * (MT_M + 4) + (e & 127)] = ta[u];
+                As[(e >> 6) * (MT_M + 4) + (e & 63)] = ta[u];
             }
         };
         auto wrB = [&](float* Bs) {
@@ -1377,13 +1378,10 @@ k_mfma_dw(const float* __restrict__ dz, const float* __restrict__ x1,
 #pragma unroll
             for (int ks = 0; ks < MT_K; ks += 2) {
                 float a0 = As[(ks + kk2) * (MT_M + 4) + wm0 + r];
-                float a1 = As[(ks + kk2) * (MT_M + 4) + wm0 + 32 + r];
                 float b0 = Bs[(ks + kk2) * (MT_N + 4) + wn0 + r];
                 float b1 = Bs[(ks + kk2) * (MT_N + 4) + wn0 + 32 + r];
                 acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
                 acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
-                acc10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
-                acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);
                 if (wm0 == 0) { bias0 += b0; bias1 += b1; }
             }
             if (more) {
@@ -1394,11 +1392,11 @@ k_mfma_dw(const float* __restrict__ dz, const float* __restrict__ x1,
         }
     }
     int wid = tid >> 6;
-    int wm0 = (wid >> 1) * 64, wn0 = (wid & 1) * 64;
-    const f32x16* accs[4] = {&acc00, &acc01, &acc10, &acc11};
+    int wm0 = (wid >> 1) * 32, wn0 = (wid & 1) * 64;
+    const f32x16* accs[2] = {&acc00, &acc01};
 #pragma unroll
-    for (int t = 0; t < 4; ++t) {
-        int i = t >> 1, j = t & 1;
+    for (int t = 0; t < 2; ++t) {
+        int i = 0, j = t;
 #pragma unroll
         for (int reg = 0; reg < 16; ++reg) {
             int row = (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
