@@ -1084,10 +1084,10 @@ __device__ inline void mfma_pipeline(FA ldA, FB ldB, int nch,
     int tid = threadIdx.x, wid = tid >> 6, lane = tid & 63;
     int wm0 = (wid >> 1) * 32, wn0 = (wid & 1) * 64;
     int r = lane & 31, kk2 = lane >> 5;
-    // prefetch depth 2: two register banks, so the ds_write for chunk
-    // ch+1 never waits on in-flight loads (they were issued at ch-1)
-    float ta0[8], tb0[16], ta1[8], tb1[16];
-    auto wr = [&](float* As, float* Bs, float* ta, float* tb) {
+    float ta[8], tb[16];
+    ldA(0, ta);
+    ldB(0, tb);
+    auto wr = [&](float* As, float* Bs) {
 #pragma unroll
         for (int u = 0; u < 8; ++u) {
             int e = u * 256 + tid;
@@ -1099,14 +1099,16 @@ __device__ inline void mfma_pipeline(FA ldA, FB ldB, int nch,
             Bs[(e >> 7) * (MT_N + 4) + (e & 127)] = tb[u];
         }
     };
-    ldA(0, ta0); ldB(0, tb0);
-    if (nch > 1) { ldA(MT_K, ta1); ldB(MT_K, tb1); }
-    wr(As0, Bs0, ta0, tb0);
-    if (nch > 2) { ldA(2 * MT_K, ta0); ldB(2 * MT_K, tb0); }
+    wr(As0, Bs0);
     __syncthreads();
     for (int ch = 0; ch < nch; ++ch) {
         float* As = (ch & 1) ? As1 : As0;
         float* Bs = (ch & 1) ? Bs1 : Bs0;
+        bool more = ch + 1 < nch;
+        if (more) {
+            ldA((ch + 1) * MT_K, ta);
+            ldB((ch + 1) * MT_K, tb);
+        }
 #pragma unroll
         for (int ks = 0; ks < MT_K; ks += 2) {
             float a0 = As[(ks + kk2) * (MT_M + 4) + wm0 + r];
@@ -1119,17 +1121,7 @@ __device__ inline void mfma_pipeline(FA ldA, FB ldB, int nch,
                 bias_acc[1] += b1;
             }
         }
-        if (ch + 1 < nch) {
-            if (((ch + 1) & 1) == 1) {
-                wr(As1, Bs1, ta1, tb1);
-                if (ch + 3 < nch) { ldA((ch + 3) * MT_K, ta1);
-                                    ldB((ch + 3) * MT_K, tb1); }
-            } else {
-                wr(As0, Bs0, ta0, tb0);
-                if (ch + 3 < nch) { ldA((ch + 3) * MT_K, ta0);
-                                    ldB((ch + 3) * MT_K, tb0); }
-            }
-        }
+        if (more) wr((ch & 1) ? As0 : As1, (ch & 1) ? Bs0 : Bs1);
         __syncthreads();
     }
 }
@@ -1232,31 +1224,38 @@ k_mfma_dx(const float* __restrict__ dz, const float* __restrict__ wt,
     // ldB writes must land at Bs[kk][nn] but wr() assumes [k][n] from the
     // (e>>7, e&127) mapping; so remap via a custom pipeline here:
     {
-        float ta0[8], tb0[16], ta1[8], tb1[16];
-        auto wrAB = [&](float* As, float* Bs, float* ta, float* tb) {
+        float ta[8], tb[16];
+        ldA(0, ta);
+        ldB(0, tb);
+        auto wrA = [&](float* As) {
 #pragma unroll
             for (int u = 0; u < 8; ++u) {
                 int e = u * 256 + tid;
                 As[(e & 31) * (MT_M + 4) + (e >> 5)] = ta[u];
             }
+        };
+        auto wrB = [&](float* Bs) {
 #pragma unroll
             for (int u = 0; u < 16; ++u) {
                 int e = u * 256 + tid;
                 Bs[(e & 31) * (MT_N + 4) + (e >> 5)] = tb[u];
             }
         };
-        int nch = (out + MT_K - 1) / MT_K;
-        ldA(0, ta0); ldB(0, tb0);
-        if (nch > 1) { ldA(MT_K, ta1); ldB(MT_K, tb1); }
-        wrAB(As2[0], Bs2[0], ta0, tb0);
-        if (nch > 2) { ldA(2 * MT_K, ta0); ldB(2 * MT_K, tb0); }
+        wrA(As2[0]);
+        wrB(Bs2[0]);
         __syncthreads();
         int wid = tid >> 6;
         int wm0 = (wid >> 1) * 32, wn0 = (wid & 1) * 64;
         int r = lane & 31, kk2 = lane >> 5;
+        int nch = (out + MT_K - 1) / MT_K;
         for (int ch = 0; ch < nch; ++ch) {
             float* As = As2[ch & 1];
             float* Bs = Bs2[ch & 1];
+            bool more = ch + 1 < nch;
+            if (more) {
+                ldA((ch + 1) * MT_K, ta);
+                ldB((ch + 1) * MT_K, tb);
+            }
 #pragma unroll
             for (int ks = 0; ks < MT_K; ks += 2) {
                 float a0 = As[(ks + kk2) * (MT_M + 4) + wm0 + r];
@@ -1265,16 +1264,9 @@ k_mfma_dx(const float* __restrict__ dz, const float* __restrict__ wt,
                 acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
                 acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
             }
-            if (ch + 1 < nch) {
-                if (((ch + 1) & 1) == 1) {
-                    wrAB(As2[1], Bs2[1], ta1, tb1);
-                    if (ch + 3 < nch) { ldA((ch + 3) * MT_K, ta1);
-                                        ldB((ch + 3) * MT_K, tb1); }
-                } else {
-                    wrAB(As2[0], Bs2[0], ta0, tb0);
-                    if (ch + 3 < nch) { ldA((ch + 3) * MT_K, ta0);
-                                        ldB((ch + 3) * MT_K, tb0); }
-                }
+            if (more) {
+                wrA(As2[(ch + 1) & 1]);
+                wrB(Bs2[(ch + 1) & 1]);
             }
             __syncthreads();
         }
@@ -1355,27 +1347,34 @@ k_mfma_dw(const float* __restrict__ dz, const float* __restrict__ x1,
         int wid = tid >> 6;
         int wm0 = (wid >> 1) * 32, wn0 = (wid & 1) * 64;
         int r = lane & 31, kk2 = lane >> 5;
-        float ta0[8], tb0[16], ta1[8], tb1[16];
-        auto wrAB = [&](float* As, float* Bs, float* ta, float* tb) {
+        float ta[8], tb[16];
+        ldA(0, ta);
+        ldB(0, tb);
+        auto wrA = [&](float* As) {
 #pragma unroll
             for (int u = 0; u < 8; ++u) {
                 int e = u * 256 + tid;
                 As[(e >> 6) * (MT_M + 4) + (e & 63)] = ta[u];
             }
+        };
+        auto wrB = [&](float* Bs) {
 #pragma unroll
             for (int u = 0; u < 16; ++u) {
                 int e = u * 256 + tid;
                 Bs[(e >> 7) * (MT_N + 4) + (e & 127)] = tb[u];
             }
         };
-        ldA(0, ta0); ldB(0, tb0);
-        if (nch > 1) { ldA(MT_K, ta1); ldB(MT_K, tb1); }
-        wrAB(As2[0], Bs2[0], ta0, tb0);
-        if (nch > 2) { ldA(2 * MT_K, ta0); ldB(2 * MT_K, tb0); }
+        wrA(As2[0]);
+        wrB(Bs2[0]);
         __syncthreads();
         for (int ch = 0; ch < nch; ++ch) {
             float* As = As2[ch & 1];
             float* Bs = Bs2[ch & 1];
+            bool more = ch + 1 < nch;
+            if (more) {
+                ldA((ch + 1) * MT_K, ta);
+                ldB((ch + 1) * MT_K, tb);
+            }
 #pragma unroll
             for (int ks = 0; ks < MT_K; ks += 2) {
                 float a0 = As[(ks + kk2) * (MT_M + 4) + wm0 + r];
@@ -1385,16 +1384,9 @@ k_mfma_dw(const float* __restrict__ dz, const float* __restrict__ x1,
                 acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
                 if (wm0 == 0) { bias0 += b0; bias1 += b1; }
             }
-            if (ch + 1 < nch) {
-                if (((ch + 1) & 1) == 1) {
-                    wrAB(As2[1], Bs2[1], ta1, tb1);
-                    if (ch + 3 < nch) { ldA((ch + 3) * MT_K, ta1);
-                                        ldB((ch + 3) * MT_K, tb1); }
-                } else {
-                    wrAB(As2[0], Bs2[0], ta0, tb0);
-                    if (ch + 3 < nch) { ldA((ch + 3) * MT_K, ta0);
-                                        ldB((ch + 3) * MT_K, tb0); }
-                }
+            if (more) {
+                wrA(As2[(ch + 1) & 1]);
+                wrB(Bs2[(ch + 1) & 1]);
             }
             __syncthreads();
         }
