@@ -316,6 +316,8 @@ class DDPG:
     def state_dict(self) -> dict:
         """Full-resume checkpoint (capability the reference lacks —
         SURVEY.md §5 checkpoint row)."""
+        if self._fused is not None:
+            self._fused.sync_params_if_dirty()
         st = {
             "actor": self.actor.state_dict(),
             "critic": self.critic.state_dict(),
@@ -363,3 +365,6 @@ class DDPG:
                                   self.critic, self.critic_target)
             for k in ("m_actor", "v_actor", "m_critic", "v_critic"):
                 eng.load_slab(k, st["engine"][k])
+            cnt = st["engine"]["counters"]
+            eng.set_schedule(cnt["adam_t_actor"], cnt["max_priority"])
+            self._fused._params_dirty = False

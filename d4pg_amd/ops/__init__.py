@@ -128,6 +128,19 @@ class GPUReplayAdapter:
     def __len__(self):
         return int(self.engine.counters()["size"]) + len(self._pending)
 
+    # -- exact resume (SURVEY §5 checkpoint row): snapshot/restore the
+    # on-HBM SoA store + sum/min trees --
+    def state_dict(self):
+        self.flush()
+        return dict(self.engine.ext.replay_state(self.engine.h))
+
+    def load_state_dict(self, st):
+        self._pending.clear()
+        self.engine.ext.load_replay_state(
+            self.engine.h, st["s"], st["a"], st["r"], st["s2"], st["d"],
+            st["sum_tree"], st["min_tree"], int(st["size"]), int(st["pos"]),
+            float(st["max_priority"]))
+
 
 # ---------------------------------------------------------------------------
 # the fused engine wrapper
@@ -219,9 +232,13 @@ class FusedEngine:
     def sync(self):
         self.ext.sync(self.h)
 
-    # -- introspection --
+    # -- introspection / resume --
     def counters(self):
         return self.ext.counters(self.h)
+
+    def set_schedule(self, steps_done, max_priority=1.0):
+        """Restore schedule counters to 'steps_done completed' (resume)."""
+        self.ext.set_schedule(self.h, int(steps_done), float(max_priority))
 
     def read(self, name):
         return self.ext.read_buffer(self.h, name)

@@ -123,6 +123,88 @@ static void ingest(int64_t h, torch::Tensor s, torch::Tensor a,
              ts2.data_ptr<float>(), td.data_ptr<float>(), T);
 }
 
+// exact-resume support: restore the schedule counters to "steps_done
+// completed" (adam/rng pre-advanced to steps_done+1, beta_t = steps_done —
+// the same convention Engine::alloc establishes at step 0)
+static void set_schedule(int64_t h, int64_t steps_done, double max_priority) {
+    Engine& e = get(h);
+    Counters c{};
+    HIP_CHECK(hipMemcpy(&c, e.cnt, sizeof(c), hipMemcpyDeviceToHost));
+    c.beta_t = steps_done;
+    c.adam_t_actor = c.adam_t_critic = c.rng_epoch = steps_done + 1;
+    c.max_priority = (float)max_priority;
+    HIP_CHECK(hipMemcpy(e.cnt, &c, sizeof(c), hipMemcpyHostToDevice));
+}
+
+// on-HBM replay snapshot: SoA rows [0, size) + both segment trees
+static py::dict replay_state(int64_t h) {
+    Engine& e = get(h);
+    Counters c{};
+    HIP_CHECK(hipMemcpy(&c, e.cnt, sizeof(c), hipMemcpyDeviceToHost));
+    long n = c.size;
+    const int O = e.cfg.obs, A = e.cfg.act;
+    py::dict d;
+    auto grab = [&](const float* src, long rows, long w) {
+        auto t = torch::empty({rows, w}, torch::kFloat32);
+        if (rows)
+            HIP_CHECK(hipMemcpy(t.data_ptr<float>(), src, rows * w * 4,
+                                hipMemcpyDeviceToHost));
+        return t;
+    };
+    d["s"] = grab(e.rs, n, O);
+    d["a"] = grab(e.ra, n, A);
+    d["r"] = grab(e.rr, n, 1);
+    d["s2"] = grab(e.rs2, n, O);
+    d["d"] = grab(e.rd, n, 1);
+    auto st = torch::empty({2 * e.tree_cap}, torch::kFloat64);
+    auto mt = torch::empty({2 * e.tree_cap}, torch::kFloat64);
+    HIP_CHECK(hipMemcpy(st.data_ptr<double>(), e.sum_tree,
+                        2 * e.tree_cap * 8, hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpy(mt.data_ptr<double>(), e.min_tree,
+                        2 * e.tree_cap * 8, hipMemcpyDeviceToHost));
+    d["sum_tree"] = st;
+    d["min_tree"] = mt;
+    d["size"] = c.size;
+    d["pos"] = c.pos;
+    d["max_priority"] = c.max_priority;
+    return d;
+}
+
+static void load_replay_state(int64_t h, torch::Tensor s, torch::Tensor a,
+                              torch::Tensor r, torch::Tensor s2,
+                              torch::Tensor dn, torch::Tensor sum_tree,
+                              torch::Tensor min_tree, int64_t size,
+                              int64_t pos, double max_priority) {
+    Engine& e = get(h);
+    const int O = e.cfg.obs, A = e.cfg.act;
+    long n = size;
+    TORCH_CHECK(sum_tree.numel() == 2 * e.tree_cap, "tree size mismatch");
+    auto put = [&](float* dst, torch::Tensor t, long rows, long w) {
+        auto tt = f32_contig(t);
+        TORCH_CHECK(tt.numel() == rows * w, "replay slab size mismatch");
+        if (rows)
+            HIP_CHECK(hipMemcpy(dst, tt.data_ptr<float>(), rows * w * 4,
+                                hipMemcpyHostToDevice));
+    };
+    put(e.rs, s, n, O);
+    put(e.ra, a, n, A);
+    put(e.rr, r, n, 1);
+    put(e.rs2, s2, n, O);
+    put(e.rd, dn, n, 1);
+    auto st = sum_tree.to(torch::kFloat64).contiguous().cpu();
+    auto mt = min_tree.to(torch::kFloat64).contiguous().cpu();
+    HIP_CHECK(hipMemcpy(e.sum_tree, st.data_ptr<double>(),
+                        2 * e.tree_cap * 8, hipMemcpyHostToDevice));
+    HIP_CHECK(hipMemcpy(e.min_tree, mt.data_ptr<double>(),
+                        2 * e.tree_cap * 8, hipMemcpyHostToDevice));
+    Counters c{};
+    HIP_CHECK(hipMemcpy(&c, e.cnt, sizeof(c), hipMemcpyDeviceToHost));
+    c.size = size;
+    c.pos = pos;
+    c.max_priority = (float)max_priority;
+    HIP_CHECK(hipMemcpy(e.cnt, &c, sizeof(c), hipMemcpyHostToDevice));
+}
+
 static void step(int64_t h, int64_t n) { get(h).step((int)n); }
 static void capture(int64_t h, int64_t n) { get(h).capture((int)n); }
 static void replay(int64_t h, int64_t iters) { get(h).replay((int)iters); }
@@ -216,6 +298,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
     mod.def("store_slab", &d4pg::store_slab);
     mod.def("synth_fill", &d4pg::synth_fill);
     mod.def("ingest", &d4pg::ingest);
+    mod.def("set_schedule", &d4pg::set_schedule);
+    mod.def("replay_state", &d4pg::replay_state);
+    mod.def("load_replay_state", &d4pg::load_replay_state);
     mod.def("step", &d4pg::step);
     mod.def("capture", &d4pg::capture);
     mod.def("replay", &d4pg::replay);
