@@ -45,7 +45,17 @@ def _dist_init(args):
         backend = "nccl" if torch.cuda.is_available() else "gloo"
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29511")
-        dist.init_process_group(backend=backend, rank=rank, world_size=world)
+        try:
+            dist.init_process_group(backend=backend, rank=rank,
+                                    world_size=world)
+        except Exception:
+            # the collectives here only synchronize timing (weak scaling,
+            # independent learners) — gloo is a safe fallback
+            if backend != "gloo":
+                dist.init_process_group(backend="gloo", rank=rank,
+                                        world_size=world)
+            else:
+                raise
         return dist, world, rank, local
     return None, 1, 0, 0
 
