@@ -342,6 +342,7 @@ class DDPG:
                 "m_critic": eng.store_slab("m_critic"),
                 "v_critic": eng.store_slab("v_critic"),
                 "counters": eng.counters(),
+                "seed": eng.info()["seed"],
             }
         return st
 
@@ -366,5 +367,7 @@ class DDPG:
             for k in ("m_actor", "v_actor", "m_critic", "v_critic"):
                 eng.load_slab(k, st["engine"][k])
             cnt = st["engine"]["counters"]
+            if "seed" in st["engine"]:
+                eng.ext.set_seed(eng.h, int(st["engine"]["seed"]))
             eng.set_schedule(cnt["adam_t_actor"], cnt["max_priority"])
             self._fused._params_dirty = False

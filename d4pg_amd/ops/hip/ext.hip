@@ -50,6 +50,7 @@ static py::dict info(int64_t h) {
     d["n_params_critic"] = e.cnet.n_params;
     d["tree_cap"] = e.tree_cap;
     d["persistent"] = e.use_persistent();
+    d["seed"] = (int64_t)e.cfg.seed;
     py::list al, cl;
     for (int i = 0; i < 4; ++i) {
         py::dict ld;
@@ -126,6 +127,10 @@ static void ingest(int64_t h, torch::Tensor s, torch::Tensor a,
 // exact-resume support: restore the schedule counters to "steps_done
 // completed" (adam/rng pre-advanced to steps_done+1, beta_t = steps_done —
 // the same convention Engine::alloc establishes at step 0)
+static void set_seed(int64_t h, int64_t seed) {
+    get(h).cfg.seed = (uint64_t)seed;
+}
+
 static void set_schedule(int64_t h, int64_t steps_done, double max_priority) {
     Engine& e = get(h);
     Counters c{};
@@ -298,6 +303,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
     mod.def("store_slab", &d4pg::store_slab);
     mod.def("synth_fill", &d4pg::synth_fill);
     mod.def("ingest", &d4pg::ingest);
+    mod.def("set_seed", &d4pg::set_seed);
     mod.def("set_schedule", &d4pg::set_schedule);
     mod.def("replay_state", &d4pg::replay_state);
     mod.def("load_replay_state", &d4pg::load_replay_state);
