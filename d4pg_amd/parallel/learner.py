@@ -114,6 +114,13 @@ class DistributedD4PG:
                 device = "cuda"
             else:
                 device = "cpu"
+        # collective-buffer device must match the process-group backend
+        # (gloo = host memory; ranks can share one GPU for compute while
+        # exchanging over gloo — the single-GPU-box topology)
+        if dist.is_initialized() and dist.get_backend() != "nccl":
+            self.comm_device = torch.device("cpu")
+        else:
+            self.comm_device = torch.device(device)
         self.device = torch.device(device)
         self.is_learner = self.rank == 0
         self.eval_rank = (self.world - 1
@@ -183,9 +190,9 @@ class DistributedD4PG:
                 self.agent.engine.sync_params_if_dirty()
             blob = torch.cat([pack_net(self.agent.actor),
                               torch.tensor([float(self.global_step)])])
-            blob = blob.to(self.device)
+            blob = blob.to(self.comm_device)
         else:
-            blob = torch.zeros(self.blob_len, device=self.device)
+            blob = torch.zeros(self.blob_len, device=self.comm_device)
         dist.broadcast(blob, src=0)
         if not self.is_learner:
             blob = blob.cpu()
@@ -231,7 +238,7 @@ class DistributedD4PG:
         return lb
 
     def _exchange(self, lb):
-        dev = self.device
+        dev = self.comm_device
         cnt = torch.tensor([float(len(lb))], device=dev)
         counts = [torch.zeros_like(cnt) for _ in range(self.world)]
         dist.all_gather(counts, cnt)
@@ -322,7 +329,13 @@ def init_process_group(backend: str | None = None):
     if dist.is_initialized():
         return
     if backend is None:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        backend = os.environ.get("D4PG_DIST_BACKEND")
+    if backend is None:
+        # nccl needs one DISTINCT GPU per rank; on a single-GPU box with
+        # several ranks (learner + CPU actor ranks) fall back to gloo
+        n_gpu = torch.cuda.device_count() if torch.cuda.is_available() else 0
+        world = int(os.environ.get("WORLD_SIZE", "1"))
+        backend = "nccl" if n_gpu >= world else "gloo"
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29521")
     os.environ.setdefault("RANK", "0")
