@@ -344,7 +344,11 @@ class DDPG:
                 "counters": eng.counters(),
                 "seed": eng.info()["seed"],
             }
-        return st
+        # nn.Module.state_dict() returns REFERENCES to the live tensors —
+        # continued training would silently mutate the checkpoint.  A
+        # checkpoint must be a snapshot: deep-copy everything.
+        import copy as _copy
+        return _copy.deepcopy(st)
 
     def load_state_dict(self, st: dict, load_replay: bool = True) -> None:
         self.actor.load_state_dict(st["actor"])
