@@ -66,4 +66,6 @@ def test_engine_resume_bitwise():
     c1 = a1.engine.engine.counters()
     c2 = a2.engine.engine.counters()
     assert c1["adam_t_actor"] == c2["adam_t_actor"] == 10
-    assert c1["loss_critic"] == c2["loss_critic"]
+    # the loss scalar accumulates via cross-workgroup fp32 atomics, whose
+    # order is nondeterministic — approx, not bitwise
+    assert c1["loss_critic"] == pytest.approx(c2["loss_critic"], rel=1e-5)
