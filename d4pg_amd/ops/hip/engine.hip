@@ -1557,6 +1557,20 @@ __device__ __noinline__ void p_fwd(float* lds, const float* x1, const float* x2,
     int tid = threadIdx.x;
     for (int t = wg_rel; t < ntiles; t += nwg) {
         int r0 = (t / nct) * PROWS, c0 = (t % nct) * 64;
+        int kk16 = tid >> 6, cc16 = tid & 63;
+        float wreg[16];
+        int nfull = in_total >> 6;
+        auto preload = [&](int kc) {
+#pragma unroll
+            for (int u = 0; u < 16; ++u) {
+                int gk = kc + kk16 + 4 * u, gc = c0 + cc16;
+                wreg[u] = (gc < out)
+                    ? wt[(long)gk * out + gc] : 0.f;
+            }
+        };
+        // first weight chunk's loads fly WITH the x-stage loads (the
+        // x ds_writes only wait on the older x loads)
+        if (nfull > 0) preload(0);
         // x-stage, register-batched (all loads in flight, then ds_writes)
         {
             int tot = PROWS * in_total;
@@ -1592,18 +1606,6 @@ __device__ __noinline__ void p_fwd(float* lds, const float* x1, const float* x2,
         // buffered (chunk k+1's loads fly during chunk k's FMAs)
         float* ws = lds + PWOFF;                 // [64][65]
         const float* xr = lds + rq * in_total;
-        int kk16 = tid >> 6, cc16 = tid & 63;
-        float wreg[16];
-        int nfull = in_total >> 6;
-        auto preload = [&](int kc) {
-#pragma unroll
-            for (int u = 0; u < 16; ++u) {
-                int gk = kc + kk16 + 4 * u, gc = c0 + cc16;
-                wreg[u] = (gc < out)
-                    ? wt[(long)gk * out + gc] : 0.f;
-            }
-        };
-        if (nfull > 0) preload(0);
         for (int ch = 0; ch < nfull; ++ch) {
             int kc = ch << 6;
             float wb[16];
@@ -1670,6 +1672,20 @@ __device__ inline void p_bwd_dx(float* lds, const float* dz, const float* wt,
     int tid = threadIdx.x;
     for (int t = wg_rel; t < ntiles; t += nwg) {
         int r0 = (t / nit) * PROWS, i0 = in_lo + (t % nit) * 64;
+        int oo16 = tid & 3;
+        int ii16 = tid >> 2;
+        float wreg[16];
+        int gi_ld = i0 + ii16;
+        int nfull = out >> 6;
+        auto preload = [&](int oc) {
+#pragma unroll
+            for (int u = 0; u < 16; ++u) {
+                int go = oc + oo16 + 4 * u;
+                wreg[u] = (gi_ld < in_hi)
+                    ? wt[(long)gi_ld * out + go] : 0.f;
+            }
+        };
+        if (nfull > 0) preload(0);
         // dz-stage, register-batched
         {
             int tot = PROWS * out;
@@ -1702,20 +1718,6 @@ __device__ inline void p_bwd_dx(float* lds, const float* dz, const float* wt,
         // buffered register-batched staging
         float* ws = lds + PWOFF;                 // [64][65] = [o][i]
         const float* zr = lds + rq * out;
-        int oo16 = tid & 3;
-        int ii16 = tid >> 2;
-        float wreg[16];
-        int gi_ld = i0 + ii16;
-        int nfull = out >> 6;
-        auto preload = [&](int oc) {
-#pragma unroll
-            for (int u = 0; u < 16; ++u) {
-                int go = oc + oo16 + 4 * u;
-                wreg[u] = (gi_ld < in_hi)
-                    ? wt[(long)gi_ld * out + go] : 0.f;
-            }
-        };
-        if (nfull > 0) preload(0);
         for (int ch = 0; ch < nfull; ++ch) {
             int oc = ch << 6;
             float wb[16];
