@@ -1,0 +1,71 @@
+"""Engine parity at awkward shapes: dims that don't divide the tile sizes
+(64-col chunks, 4-row tiles, wave-wide softmax) and batches that select
+each execution path — persistent (B<=256), row-block (256<B<=512), and
+MFMA (B>=512) — all against the eager fp32 oracle."""
+
+import copy
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+GAMMA_N = 0.99 ** 3
+
+
+def run_one_step(O, A, H, K, B, cap=4096, seed=3):
+    from d4pg_amd.algo.projection import categorical_projection
+    from d4pg_amd.models import actor, critic
+    from d4pg_amd.ops import FusedEngine
+    dist = {"type": "categorical", "v_min": -100.0, "v_max": 50.0,
+            "n_atoms": K}
+    eng = FusedEngine(obs_dim=O, act_dim=A, hidden=H, n_atoms=K, batch=B,
+                      capacity=cap, v_min=-100.0, v_max=50.0,
+                      gamma_n=GAMMA_N, tau=0.01, lr_actor=1e-4,
+                      lr_critic=1e-4, seed=seed)
+    torch.manual_seed(seed)
+    a = actor(O, A, hidden=H)
+    c = critic(O, A, dist, hidden=H)
+    at = copy.deepcopy(a)
+    ct = copy.deepcopy(c)
+    eng.load_from_modules(a, at, c, ct)
+    rng = np.random.default_rng(seed)
+    n = max(2 * B, 512)
+    eng.ingest(torch.from_numpy(rng.standard_normal((n, O)).astype("f")),
+               torch.from_numpy(rng.uniform(-1, 1, (n, A)).astype("f")),
+               torch.from_numpy(rng.uniform(-20, 5, n).astype("f")),
+               torch.from_numpy(rng.standard_normal((n, O)).astype("f")),
+               torch.from_numpy((rng.random(n) < 0.1).astype("f")))
+    eng.step(1)
+
+    s = torch.from_numpy(eng.read("bs").numpy())
+    act = torch.from_numpy(eng.read("ba").numpy())
+    r = torch.from_numpy(eng.read("br").numpy())
+    s2 = torch.from_numpy(eng.read("bs2").numpy())
+    d = torch.from_numpy(eng.read("bd").numpy())
+    with torch.no_grad():
+        a2 = at(s2)
+        p_t = ct(s2, a2)
+        m = categorical_projection(p_t, r, d, -100.0, 50.0, GAMMA_N)
+        q = c(s, act)
+    np.testing.assert_allclose(eng.read("a2").numpy(), a2.numpy(), atol=5e-5)
+    np.testing.assert_allclose(eng.read("p_t").numpy(), p_t.numpy(),
+                               atol=5e-5)
+    np.testing.assert_allclose(eng.read("m_proj").numpy(), m.numpy(),
+                               atol=1e-4)
+    np.testing.assert_allclose(eng.read("q").numpy(), q.numpy(), atol=5e-5)
+    cnt = eng.counters()
+    assert cnt["adam_t_actor"] == 1
+    assert np.isfinite(cnt["loss_critic"]) and np.isfinite(cnt["loss_actor"])
+
+
+@pytest.mark.parametrize("O,A,H,K,B", [
+    (5, 2, 192, 21, 48),       # persistent path, nothing divides nicely
+    (7, 3, 128, 51, 33),       # persistent, odd batch
+    (11, 1, 320, 64, 256),     # persistent upper edge, K == wave
+    (5, 2, 192, 31, 300),      # row-block path (256 < B <= 512)
+    (9, 4, 192, 41, 640),      # MFMA path, dims off the 64/128 tiles
+])
+def test_engine_shape_parity(O, A, H, K, B):
+    run_one_step(O, A, H, K, B)
