@@ -3331,11 +3331,11 @@ public:
                    d1, nullptr, c_h1, ACT_RELU, true);
         launch_bwd(d1, bs, nullptr, p_critic, g_critic, cnet.l[0],
                    nullptr, nullptr, nullptr, ACT_NONE, true);
-        // P15: Adam critic
-        hipLaunchKernelGGL(k_adam, dim3(256), dim3(256), 0, stream,
+        // P15: Adam critic + target soft-update fused
+        hipLaunchKernelGGL(k_adam_lerp, dim3(256), dim3(256), 0, stream,
                            p_critic, g_critic, m_critic, v_critic,
-                           cnet.n_params, cfg.lr_critic, 0.9f, 0.999f, 1e-8f,
-                           cnt, 0);
+                           p_critic_t, cnet.n_params, cfg.lr_critic,
+                           0.9f, 0.999f, 1e-8f, cfg.tau, cnt, 0);
         // P16: actor.L1(s) | critic'.L1(s)
         {
             int wg = 0;
@@ -3382,15 +3382,12 @@ public:
                    pd2, nullptr, pa_h1, ACT_RELU, true);
         launch_bwd(pd2, bs, nullptr, p_actor, g_actor, anet.l[0],
                    nullptr, nullptr, nullptr, ACT_NONE, true);
-        // P32: Adam actor
-        hipLaunchKernelGGL(k_adam, dim3(256), dim3(256), 0, stream,
+        // P32: Adam actor + target soft-update fused (critic target was
+        // lerped in P15; ordering matches the row-block/persistent paths)
+        hipLaunchKernelGGL(k_adam_lerp, dim3(256), dim3(256), 0, stream,
                            p_actor, g_actor, m_actor, v_actor,
-                           anet.n_params, cfg.lr_actor, 0.9f, 0.999f, 1e-8f,
-                           cnt, 1);
-        // P33: soft updates (both nets)
-        hipLaunchKernelGGL(k_soft_update, dim3(256), dim3(256), 0, stream,
-                           p_actor_t, p_actor, anet.n_params,
-                           p_critic_t, p_critic, cnet.n_params, cfg.tau);
+                           p_actor_t, anet.n_params, cfg.lr_actor,
+                           0.9f, 0.999f, 1e-8f, cfg.tau, cnt, 1);
         // P34: PER priority write-back.  Large batches use the per-level
         // grid-wide repair (the one-wg level-synced kernel serializes).
         if (B >= 512) {
