@@ -22,7 +22,7 @@ import time
 
 import numpy as np
 
-from ..her import add_experience, flat_obs, rollout_episode
+from ..her import add_experience, rollout_episode
 from ..utils.logging import Meter, SummaryWriter
 
 

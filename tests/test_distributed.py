@@ -6,7 +6,6 @@ ingest+train) is exercised here without a GPU."""
 import os
 
 import numpy as np
-import pytest
 import torch
 import torch.multiprocessing as mp
 
