@@ -192,3 +192,52 @@ def test_flat_obs_dict_and_array():
     d = {"observation": np.array([1.0]), "desired_goal": np.array([2.0]),
          "achieved_goal": np.array([3.0])}
     assert flat_obs(d).tolist() == [1.0, 2.0]
+
+
+def test_reproj_categorical_dist_matches_projection():
+    """The public vectorized projection method (reference ddpg.py:122-140
+    API parity) must agree with the oracle categorical_projection."""
+    import torch
+    from d4pg_amd.algo.d4pg import DDPG
+    from d4pg_amd.algo.projection import categorical_projection
+    agent = DDPG(3, 1, memory_size=100, batch_size=8,
+                 critic_dist_info={"type": "categorical", "v_min": -10.0,
+                                   "v_max": 10.0, "n_atoms": 11},
+                 n_steps=3, gamma=0.9, seed=0)
+    rng = np.random.default_rng(0)
+    p = rng.random((8, 11)).astype(np.float32)
+    p /= p.sum(1, keepdims=True)
+    r = rng.uniform(-5, 5, 8).astype(np.float32)
+    d = (rng.random(8) < 0.3).astype(np.float32)
+    m1 = agent.reproj_categorical_dist(torch.from_numpy(p),
+                                       torch.from_numpy(r),
+                                       torch.from_numpy(d))
+    m2 = categorical_projection(torch.from_numpy(p), torch.from_numpy(r),
+                                torch.from_numpy(d), -10.0, 10.0, 0.9 ** 3)
+    np.testing.assert_allclose(np.asarray(m1), m2.numpy(), atol=1e-6)
+
+
+def test_evaluator_function_runs():
+    """global_model_eval (reference main.py:103-134 parity): copies global
+    weights, rolls one greedy episode, returns the EWMA return."""
+    import torch
+    from d4pg_amd.algo.d4pg import DDPG
+    from d4pg_amd.config import configure_env_params, make_parser
+    from d4pg_amd.envs import make
+    from d4pg_amd.parallel.worker import global_model_eval
+    args = make_parser().parse_args(["--env", "Pendulum-v1",
+                                     "--max_steps", "20", "--debug", "0"])
+    configure_env_params(args)
+    gm = DDPG(3, 1, memory_size=100, batch_size=8,
+              critic_dist_info={"type": "categorical", "v_min": -300.0,
+                                "v_max": 0.0, "n_atoms": 51}, seed=0)
+    count = torch.zeros(1)
+
+    def factory():
+        env = make("Pendulum-v1", seed=5)
+        env._max_episode_steps = 20
+        return env
+
+    ewma = global_model_eval(gm, count, args, factory, period=0.0,
+                             max_iters=2)
+    assert isinstance(ewma, float) and np.isfinite(ewma)
