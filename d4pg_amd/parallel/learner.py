@@ -356,10 +356,14 @@ def run_distributed(args, rounds: int | None = None, with_evaluator=True):
     t0 = time.perf_counter()
     n_rounds = rounds if rounds is not None else args.n_eps * args.cycles_per_epoch
     step = node.run(n_rounds, save=(rank == 0))
+    dt = time.perf_counter() - t0
     if rank == 0:
-        dt = time.perf_counter() - t0
         print(f"[learner] done: {step} grad steps in {dt:.1f}s "
               f"({step / max(dt, 1e-9):.1f} steps/s)", flush=True)
+    elif not node.is_evaluator:
+        print(f"[actor {rank}] done: {node.env_meter.count} env steps in "
+              f"{dt:.1f}s ({node.env_meter.count / max(dt, 1e-9):.1f} "
+              f"env-steps/s)", flush=True)
     dist.barrier()
     dist.destroy_process_group()
     return step
