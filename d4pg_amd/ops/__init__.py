@@ -103,6 +103,7 @@ class GPUReplayAdapter:
     def __init__(self, engine: "FusedEngine"):
         self.engine = engine
         self._pending = []
+        self._batches = []
 
     def add(self, state, action, reward, next_state, done):
         self._pending.append((
@@ -114,19 +115,36 @@ class GPUReplayAdapter:
         if len(self._pending) >= 32768:
             self.flush()
 
+    def add_batch(self, states, actions, rewards, next_states, dones):
+        """Array-form ingestion (the distributed learner's exchange path
+        decodes whole SoA blocks — a per-transition python loop was the
+        learner's bottleneck at high actor counts)."""
+        self._batches.append((
+            np.asarray(states, np.float32),
+            np.asarray(actions, np.float32),
+            np.asarray(rewards, np.float32).ravel(),
+            np.asarray(next_states, np.float32),
+            np.asarray(dones, np.float32).ravel()))
+
     def flush(self):
-        if not self._pending:
-            return
-        s = torch.from_numpy(np.stack([p[0] for p in self._pending]))
-        a = torch.from_numpy(np.stack([p[1] for p in self._pending]))
-        r = torch.from_numpy(np.asarray([p[2] for p in self._pending]))
-        s2 = torch.from_numpy(np.stack([p[3] for p in self._pending]))
-        d = torch.from_numpy(np.asarray([p[4] for p in self._pending]))
-        self._pending.clear()
-        self.engine.ingest(s, a, r, s2, d)
+        parts = list(self._batches)
+        self._batches.clear()
+        if self._pending:
+            parts.append((
+                np.stack([p[0] for p in self._pending]),
+                np.stack([p[1] for p in self._pending]),
+                np.asarray([p[2] for p in self._pending], np.float32),
+                np.stack([p[3] for p in self._pending]),
+                np.asarray([p[4] for p in self._pending], np.float32)))
+            self._pending.clear()
+        for s, a, r, s2, d in parts:
+            self.engine.ingest(torch.from_numpy(s), torch.from_numpy(a),
+                               torch.from_numpy(r), torch.from_numpy(s2),
+                               torch.from_numpy(d))
 
     def __len__(self):
-        return int(self.engine.counters()["size"]) + len(self._pending)
+        return (int(self.engine.counters()["size"]) + len(self._pending)
+                + sum(len(b[2]) for b in self._batches))
 
     # -- exact resume (SURVEY §5 checkpoint row): snapshot/restore the
     # on-HBM SoA store + sum/min trees --

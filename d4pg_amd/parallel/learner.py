@@ -51,12 +51,14 @@ from ..utils.logging import Meter, SummaryWriter
 
 
 class _ListBuffer:
-    """Replay-interface shim that records transitions into a list, so the
-    actor side can reuse add_experience (n-step fold + HER relabel) and
-    ship the result over the wire instead of into a local buffer."""
+    """Replay-interface shim that records transitions (scalar adds or
+    whole [M,...] batches), so the actor side can reuse add_experience
+    (n-step fold + HER relabel) or the vectorized collector and ship the
+    result over the wire instead of into a local buffer."""
 
     def __init__(self):
         self.items = []
+        self.batches = []
 
     def add(self, state, action, reward, next_state, done):
         self.items.append((np.asarray(state, np.float32).ravel(),
@@ -65,27 +67,51 @@ class _ListBuffer:
                            np.asarray(next_state, np.float32).ravel(),
                            np.float32(done)))
 
+    def add_batch(self, S, A, R, S2, D):
+        self.batches.append((np.asarray(S, np.float32),
+                             np.asarray(A, np.float32),
+                             np.asarray(R, np.float32).ravel(),
+                             np.asarray(S2, np.float32),
+                             np.asarray(D, np.float32).ravel()))
+
+    def to_arrays(self, obs_dim, act_dim):
+        parts = list(self.batches)
+        if self.items:
+            parts.append((
+                np.stack([p[0] for p in self.items]),
+                np.stack([p[1] for p in self.items]),
+                np.asarray([p[2] for p in self.items], np.float32),
+                np.stack([p[3] for p in self.items]),
+                np.asarray([p[4] for p in self.items], np.float32)))
+        if not parts:
+            z = np.zeros((0, obs_dim), np.float32)
+            za = np.zeros((0, act_dim), np.float32)
+            zr = np.zeros(0, np.float32)
+            return z, za, zr, z.copy(), zr.copy()
+        return tuple(np.concatenate([p[i] for p in parts])
+                     for i in range(5))
+
     def __len__(self):
-        return len(self.items)
+        return len(self.items) + sum(len(b[2]) for b in self.batches)
 
 
 def _row_width(obs_dim: int, act_dim: int) -> int:
     return 2 * obs_dim + act_dim + 2
 
 
-def _encode(items, cap, obs_dim, act_dim, device):
+def _encode(lb, cap, obs_dim, act_dim, device):
     buf = torch.zeros(cap, _row_width(obs_dim, act_dim), device=device)
-    n = min(len(items), cap)
+    S, A, R, S2, D = (lb.to_arrays(obs_dim, act_dim)
+                      if hasattr(lb, "to_arrays") else lb)
+    n = min(len(R), cap)
     if n:
-        o = obs_dim
-        a = act_dim
-        arr = np.zeros((n, _row_width(o, a)), np.float32)
-        for i, (s, ac, r, s2, d) in enumerate(items[:n]):
-            arr[i, :o] = s
-            arr[i, o:o + a] = ac
-            arr[i, o + a] = r
-            arr[i, o + a + 1:2 * o + a + 1] = s2
-            arr[i, 2 * o + a + 1] = d
+        o, a = obs_dim, act_dim
+        arr = np.empty((n, _row_width(o, a)), np.float32)
+        arr[:, :o] = S[:n]
+        arr[:, o:o + a] = A[:n]
+        arr[:, o + a] = R[:n]
+        arr[:, o + a + 1:2 * o + a + 1] = S2[:n]
+        arr[:, 2 * o + a + 1] = D[:n]
         buf[:n] = torch.from_numpy(arr).to(device)
     return buf, n
 
@@ -223,9 +249,7 @@ class DistributedD4PG:
                     obs2, r, done = self.vector.step(a)
                     out = self.vfold.push(obs, a, r, obs2, done)
                     if out is not None:
-                        S, A, R, S2, D = out
-                        for e in range(len(R)):
-                            lb.add(S[e], A[e], R[e], S2[e], D[e])
+                        lb.add_batch(*out)
                     obs = obs2
             self.env_meter.add(self.vector.horizon * self.vector.n)
             return lb
@@ -243,7 +267,7 @@ class DistributedD4PG:
         counts = [torch.zeros_like(cnt) for _ in range(self.world)]
         dist.all_gather(counts, cnt)
         self._last_counts = counts
-        buf, n = _encode(lb.items, self.push_cap, self.obs_dim,
+        buf, n = _encode(lb, self.push_cap, self.obs_dim,
                          self.act_dim, dev)
         blocks = [torch.zeros_like(buf) for _ in range(self.world)]
         dist.all_gather(blocks, buf)
@@ -256,8 +280,12 @@ class DistributedD4PG:
                 continue
             s, a, rw, s2, d = _decode(blocks[r], n_r, self.obs_dim,
                                       self.act_dim)
-            for i in range(n_r):
-                self.agent.replayBuffer.add(s[i], a[i], rw[i], s2[i], d[i])
+            buf = self.agent.replayBuffer
+            if hasattr(buf, "add_batch"):
+                buf.add_batch(s, a, rw, s2, d)
+            else:
+                for i in range(n_r):
+                    buf.add(s[i], a[i], rw[i], s2[i], d[i])
             total += n_r
         return total
 

@@ -83,7 +83,7 @@ def test_encode_decode_roundtrip():
     for _ in range(17):
         lb.add(rng.standard_normal(3), rng.standard_normal(1),
                rng.random(), rng.standard_normal(3), 0.0)
-    buf, n = _encode(lb.items, 32, 3, 1, torch.device("cpu"))
+    buf, n = _encode(lb, 32, 3, 1, torch.device("cpu"))
     assert n == 17
     s, a, r, s2, d = _decode(buf, n, 3, 1)
     np.testing.assert_allclose(s[3], lb.items[3][0], rtol=1e-6)
