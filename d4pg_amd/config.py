@@ -74,6 +74,12 @@ def make_parser() -> argparse.ArgumentParser:
     p.add_argument("--eval_trials", default=10, type=int,
                    help="greedy eval rollouts per cycle (reference main.py:309)")
     p.add_argument("--seed", default=0, type=int, help="RNG seed")
+    p.add_argument("--lr_actor", default=1e-4, type=float,
+                   help="actor Adam lr (reference local Adam 1e-4, "
+                        "ddpg.py:67)")
+    p.add_argument("--lr_critic", default=1e-3, type=float,
+                   help="critic Adam lr (reference global Adam 1e-3, "
+                        "main.py:384)")
     p.add_argument("--vector_envs", default=0, type=int,
                    help="MI355X extension: batch M vectorized envs per "
                         "actor rank (one [M,obs] policy forward per tick) "

@@ -165,6 +165,8 @@ class DistributedD4PG:
             self.obs_dim, self.act_dim, env=self.env,
             memory_size=args.rmsize if self.is_learner else 1,
             batch_size=args.bsize, gamma=args.gamma, tau=args.tau,
+            lr_actor=getattr(args, "lr_actor", 1e-4),
+            lr_critic=getattr(args, "lr_critic", 1e-3),
             prioritized_replay=bool(args.p_replay) and self.is_learner,
             critic_dist_info=critic_dist_info(args), n_steps=args.n_steps,
             device=str(self.device) if self.is_learner else "cpu",

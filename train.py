@@ -43,6 +43,7 @@ def main(argv=None):
         backend = "hip" if device == "cuda" else "eager"
     agent = DDPG(obs_dim, act_dim, env=env, memory_size=args.rmsize,
                  batch_size=args.bsize, gamma=args.gamma, tau=args.tau,
+                 lr_actor=args.lr_actor, lr_critic=args.lr_critic,
                  prioritized_replay=bool(args.p_replay),
                  critic_dist_info=critic_dist_info(args),
                  n_steps=args.n_steps, device=device, backend=backend,
