@@ -303,9 +303,11 @@ class DistributedD4PG:
             if self.agent.engine is None:
                 from ..ops import build_fused_engine
                 self.agent._fused = build_fused_engine(self.agent)
-            self.agent.replayBuffer.flush()
-            self.agent.engine.engine.train_steps(n)
-            self.agent.train_steps_done += n
+            # go through the bridge (flush + multi-step + params-dirty flag;
+            # bypassing it once silently broadcast STALE initial actor
+            # params every round — the dirty flag is what makes
+            # sync_params_if_dirty pull the trained weights back)
+            self.agent.engine.step(n=n)
         else:
             for _ in range(n):
                 self.agent.train()
