@@ -54,3 +54,34 @@ def test_worker_cycle_on_gpu(tmp_path):
     cnt = agent.engine.engine.counters()
     assert cnt["adam_t_actor"] == 20
     assert np.isfinite(cnt["loss_critic"])
+
+
+def test_distributed_learner_train_marks_params_dirty():
+    """Regression: the distributed learner's _train must leave the engine
+    bridge params-dirty so the next broadcast ships TRAINED weights (a
+    direct engine.train_steps call once bypassed the flag and silently
+    broadcast the initial actor forever)."""
+    from d4pg_amd.config import configure_env_params, make_parser
+    from d4pg_amd.ops import pack_net
+    from d4pg_amd.parallel.learner import DistributedD4PG
+
+    args = make_parser().parse_args(
+        ["--env", "Pendulum-v1", "--max_steps", "50", "--warmup", "0",
+         "--rmsize", "10000", "--bsize", "64", "--n_steps", "5",
+         "--debug", "0", "--train_steps_per_cycle", "5", "--seed", "1"])
+    configure_env_params(args)
+    node = DistributedD4PG(args, rank=0, world=1, device="cuda",
+                           with_evaluator=False)
+    rng = np.random.default_rng(0)
+    for _ in range(500):
+        node.agent.replayBuffer.add(
+            rng.standard_normal(3).astype("f"),
+            rng.uniform(-1, 1, 1).astype("f"), -rng.random(),
+            rng.standard_normal(3).astype("f"), 0.0)
+    before = pack_net(node.agent.actor).clone()
+    node._train()
+    assert node.agent.engine is not None
+    node.agent.engine.sync_params_if_dirty()
+    after = pack_net(node.agent.actor)
+    assert not torch.equal(before, after), \
+        "broadcast would ship stale params"
