@@ -1551,7 +1551,6 @@ __device__ __noinline__ void p_fwd(float* lds, const float* x1, const float* x2,
                              int B, int in1, int in2, int out, int act_kind,
                              int wg_rel, int nwg) {
     int in_total = in1 + in2;
-    int pad_in = (in_total + 63) & ~63;       // zero-padded K => no tail path
     int nrt = (B + PROWS - 1) / PROWS;
     int nct = (out + 63) / 64;
     int ntiles = nrt * nct;
@@ -1560,22 +1559,21 @@ __device__ __noinline__ void p_fwd(float* lds, const float* x1, const float* x2,
         int r0 = (t / nct) * PROWS, c0 = (t % nct) * 64;
         int kk16 = tid >> 6, cc16 = tid & 63;
         float wreg[16];
-        int nfull = pad_in >> 6;
+        int nfull = in_total >> 6;
         auto preload = [&](int kc) {
 #pragma unroll
             for (int u = 0; u < 16; ++u) {
                 int gk = kc + kk16 + 4 * u, gc = c0 + cc16;
-                wreg[u] = (gk < in_total && gc < out)
+                wreg[u] = (gc < out)
                     ? wt[(long)gk * out + gc] : 0.f;
             }
         };
         // first weight chunk's loads fly WITH the x-stage loads (the
         // x ds_writes only wait on the older x loads)
-        preload(0);
-        // x-stage, register-batched (all loads in flight, then ds_writes);
-        // columns [in_total, pad_in) zero-filled
+        if (nfull > 0) preload(0);
+        // x-stage, register-batched (all loads in flight, then ds_writes)
         {
-            int tot = PROWS * pad_in;
+            int tot = PROWS * in_total;
             for (int base = 0; base < tot; base += 256 * 8) {
                 float tmp[8];
 #pragma unroll
@@ -1583,9 +1581,9 @@ __device__ __noinline__ void p_fwd(float* lds, const float* x1, const float* x2,
                     int e = base + u * 256 + tid;
                     float v = 0.f;
                     if (e < tot) {
-                        int rr_ = e / pad_in, kk = e % pad_in;
+                        int rr_ = e / in_total, kk = e % in_total;
                         int gb = r0 + rr_;
-                        if (gb < B && kk < in_total)
+                        if (gb < B)
                             v = (kk < in1)
                                 ? x1[(long)gb * in1 + kk]
                                 : x2[(long)gb * in2 + (kk - in1)];
@@ -1607,13 +1605,13 @@ __device__ __noinline__ void p_fwd(float* lds, const float* x1, const float* x2,
         // weight slice K-chunked through LDS, register-batched + double-
         // buffered (chunk k+1's loads fly during chunk k's FMAs)
         float* ws = lds + PWOFF;                 // [64][65]
-        const float* xr = lds + rq * pad_in;
+        const float* xr = lds + rq * in_total;
         for (int ch = 0; ch < nfull; ++ch) {
             int kc = ch << 6;
             float wb[16];
 #pragma unroll
             for (int u = 0; u < 16; ++u) wb[u] = wreg[u];
-            if (ch + 1 < nfull) preload(kc + 64);
+            if (kc + 64 < (nfull << 6)) preload(kc + 64);
 #pragma unroll
             for (int u = 0; u < 16; ++u)
                 ws[(kk16 + 4 * u) * 65 + cc16] = wb[u];
@@ -1621,6 +1619,21 @@ __device__ __noinline__ void p_fwd(float* lds, const float* x1, const float* x2,
             if (o < out) {
 #pragma unroll 8
                 for (int k = 0; k < 64; ++k)
+                    acc += xr[kc + k] * ws[k * 65 + c];
+            }
+            __syncthreads();
+        }
+        int kc = nfull << 6, klen = in_total - kc;
+        if (klen > 0) {
+            for (int e = tid; e < (klen << 6); e += 256) {
+                int kk = e >> 6, cc = e & 63;
+                int gc = c0 + cc;
+                ws[kk * 65 + cc] = (gc < out)
+                    ? wt[(long)(kc + kk) * out + gc] : 0.f;
+            }
+            __syncthreads();
+            if (o < out) {
+                for (int k = 0; k < klen; ++k)
                     acc += xr[kc + k] * ws[k * 65 + c];
             }
             __syncthreads();
@@ -1653,7 +1666,6 @@ __device__ inline void p_bwd_dx(float* lds, const float* dz, const float* wt,
                                 const float* hprev, int prev_act, float* dx,
                                 int wg_rel, int nwg) {
     int span = in_hi - in_lo;
-    int pad_out = (out + 63) & ~63;           // zero-padded O => no tail path
     int nrt = (B + PROWS - 1) / PROWS;
     int nit = (span + 63) / 64;
     int ntiles = nrt * nit;
@@ -1664,19 +1676,19 @@ __device__ inline void p_bwd_dx(float* lds, const float* dz, const float* wt,
         int ii16 = tid >> 2;
         float wreg[16];
         int gi_ld = i0 + ii16;
-        int nfull = pad_out >> 6;
+        int nfull = out >> 6;
         auto preload = [&](int oc) {
 #pragma unroll
             for (int u = 0; u < 16; ++u) {
                 int go = oc + oo16 + 4 * u;
-                wreg[u] = (gi_ld < in_hi && go < out)
+                wreg[u] = (gi_ld < in_hi)
                     ? wt[(long)gi_ld * out + go] : 0.f;
             }
         };
-        preload(0);
-        // dz-stage, register-batched; columns [out, pad_out) zero-filled
+        if (nfull > 0) preload(0);
+        // dz-stage, register-batched
         {
-            int tot = PROWS * pad_out;
+            int tot = PROWS * out;
             for (int base = 0; base < tot; base += 256 * 8) {
                 float tmp[8];
 #pragma unroll
@@ -1684,10 +1696,9 @@ __device__ inline void p_bwd_dx(float* lds, const float* dz, const float* wt,
                     int e = base + u * 256 + tid;
                     float v = 0.f;
                     if (e < tot) {
-                        int rr_ = e / pad_out, oo = e % pad_out;
+                        int rr_ = e / out, oo = e % out;
                         int gb = r0 + rr_;
-                        if (gb < B && oo < out)
-                            v = dz[(long)gb * out + oo];
+                        if (gb < B) v = dz[(long)gb * out + oo];
                     }
                     tmp[u] = v;
                 }
@@ -1706,13 +1717,13 @@ __device__ inline void p_bwd_dx(float* lds, const float* dz, const float* wt,
         // weight tile O-chunked through LDS, transposed write, double-
         // buffered register-batched staging
         float* ws = lds + PWOFF;                 // [64][65] = [o][i]
-        const float* zr = lds + rq * pad_out;
+        const float* zr = lds + rq * out;
         for (int ch = 0; ch < nfull; ++ch) {
             int oc = ch << 6;
             float wb[16];
 #pragma unroll
             for (int u = 0; u < 16; ++u) wb[u] = wreg[u];
-            if (ch + 1 < nfull) preload(oc + 64);
+            if (oc + 64 < (nfull << 6)) preload(oc + 64);
 #pragma unroll
             for (int u = 0; u < 16; ++u)
                 ws[(oo16 + 4 * u) * 65 + ii16] = wb[u];
@@ -1720,6 +1731,21 @@ __device__ inline void p_bwd_dx(float* lds, const float* dz, const float* wt,
             if (i < in_hi) {
 #pragma unroll 8
                 for (int o = 0; o < 64; ++o)
+                    acc += zr[oc + o] * ws[o * 65 + c];
+            }
+            __syncthreads();
+        }
+        int oc = nfull << 6, olen = out - oc;
+        if (olen > 0) {
+            for (int e = tid; e < 4096; e += 256) {
+                int ii = e >> 6, oo = e & 63;
+                int gi = i0 + ii;
+                ws[oo * 65 + ii] = (gi < in_hi && oo < olen)
+                    ? wt[(long)gi * out + (oc + oo)] : 0.f;
+            }
+            __syncthreads();
+            if (i < in_hi) {
+                for (int o = 0; o < olen; ++o)
                     acc += zr[oc + o] * ws[o * 65 + c];
             }
             __syncthreads();
