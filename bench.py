@@ -42,7 +42,8 @@ def _dist_init(args):
     local = int(os.environ.get("LOCAL_RANK", str(rank)))
     if world > 1:
         import torch.distributed as dist
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        backend = os.environ.get("BENCH_DIST_BACKEND") or (
+            "nccl" if torch.cuda.is_available() else "gloo")
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29511")
         try:
