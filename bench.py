@@ -71,8 +71,9 @@ def _barrier(dist):
 def _max_over_ranks(dist, x: float) -> float:
     if dist is None:
         return x
-    dev = "cuda" if torch.cuda.is_available() else "cpu"
-    t = torch.tensor([x], dtype=torch.float64, device=dev)
+    use_cuda = torch.cuda.is_available() and dist.get_backend() == "nccl"
+    t = torch.tensor([x], dtype=torch.float64,
+                     device="cuda" if use_cuda else "cpu")
     dist.all_reduce(t, op=dist.ReduceOp.MAX)
     return float(t.item())
 
