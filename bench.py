@@ -187,7 +187,9 @@ def main():
     n_gpus = max(world, args.gpus) if world > 1 else args.gpus
     use_gpu = torch.cuda.is_available()
     if use_gpu:
-        torch.cuda.set_device(local)
+        # modulo is the identity on a full node (one rank per GPU) and
+        # lets single-GPU rehearsals run several ranks on device 0
+        torch.cuda.set_device(local % torch.cuda.device_count())
 
     K, W = args.steps, args.warmup
     if use_gpu:
