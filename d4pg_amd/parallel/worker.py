@@ -58,7 +58,17 @@ class Worker:
                            gamma=self.args.gamma, rng=self.rng)
 
     def train_cycle(self, global_model=None, global_count=None) -> None:
-        for _ in range(self.args.train_steps_per_cycle):
+        n = self.args.train_steps_per_cycle
+        if global_model is None and self.agent.backend == "hip":
+            # fused multi-step: one engine launch for the whole cycle
+            self.agent.train()              # builds the bridge on first use
+            if n > 1:
+                self.agent.engine.step(n=n - 1)
+            self.grad_meter.add(n)
+            if global_count is not None:
+                global_count += n
+            return
+        for _ in range(n):
             self.agent.train(global_model)
             self.grad_meter.add()
             if global_count is not None:
