@@ -354,6 +354,74 @@ __global__ void k_ce_grad(const float* __restrict__ q,
     }
 }
 
+// ---- row softmax (wide path: MFMA computes logits, this normalizes) --------
+__global__ void k_row_softmax(const float* __restrict__ logits,
+                              float* __restrict__ y, int B, int K) {
+    int row = blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
+    int lane = threadIdx.x & 63;
+    if (row >= B) return;
+    float v = (lane < K) ? logits[(long)row * K + lane] : -INFINITY;
+    float mx = v;
+    for (int s = 32; s > 0; s >>= 1) mx = fmaxf(mx, __shfl_xor(mx, s, 64));
+    float e = (lane < K) ? __expf(v - mx) : 0.f;
+    float sum = e;
+    for (int s = 32; s > 0; s >>= 1) sum += __shfl_xor(sum, s, 64);
+    if (lane < K) y[(long)row * K + lane] = e / sum;
+}
+
+// ---- fused C51 projection + CE grad + priorities (wide path) ----------------
+// (the persistent path fuses these in p_project_ce; this is the standalone
+// twin for the per-layer/MFMA path — saves a launch + the m round-trip)
+__global__ void k_project_ce(const float* __restrict__ p_t,
+                             const float* __restrict__ r,
+                             const float* __restrict__ d,
+                             const float* __restrict__ q,
+                             const float* __restrict__ w,
+                             float* __restrict__ m_out,
+                             float* __restrict__ dlogits,
+                             float* __restrict__ pri, Counters* cnt,
+                             int B, int K, float v_min, float v_max,
+                             float gamma_n, float per_eps,
+                             int is_weighting) {
+    int row = blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
+    int lane = threadIdx.x & 63;
+    int wrow = threadIdx.x / 64;
+    extern __shared__ float lm[];
+    float* mrow = lm + wrow * 64;
+    mrow[lane] = 0.f;
+    __builtin_amdgcn_wave_barrier();
+    if (row < B && lane < K) {
+        float delta = (v_max - v_min) / (K - 1);
+        float z = v_min + lane * delta;
+        float tz = r[row] + gamma_n * (1.f - d[row]) * z;
+        tz = fminf(v_max, fmaxf(v_min, tz));
+        float b = (tz - v_min) / delta;
+        int l = (int)floorf(b), u = (int)ceilf(b);
+        if (l == u) { if (u > 0) l -= 1; else u += 1; }
+        float p = p_t[(long)row * K + lane];
+        atomicAdd(&mrow[l], p * ((float)u - b));
+        atomicAdd(&mrow[u], p * (b - (float)l));
+    }
+    __builtin_amdgcn_wave_barrier();
+    if (row >= B) return;
+    float mv = (lane < K) ? mrow[lane] : 0.f;
+    float qv = (lane < K) ? q[(long)row * K + lane] : 0.f;
+    if (lane < K) m_out[(long)row * K + lane] = mv;
+    float dot = mv * qv;
+    float ce = -mv * __logf(qv + 1e-10f);
+    for (int ss = 32; ss > 0; ss >>= 1) {
+        dot += __shfl_xor(dot, ss, 64);
+        ce += __shfl_xor(ce, ss, 64);
+    }
+    float scale = (is_weighting && w) ? w[row] : 1.f;
+    if (lane < K)
+        dlogits[(long)row * K + lane] = scale * (qv - mv) / (float)B;
+    if (lane == 0) {
+        pri[row] = dot + per_eps;
+        atomicAdd(&cnt->loss_critic, scale * ce / (float)B);
+    }
+}
+
 // ---- policy gradient through the softmax head (K6 seed) ---------------------
 // L = -(1/B) sum_b sum_k q_k z_k  =>  dlogits_j = -q_j (z_j - E_q[z]) / B
 __global__ void k_policy_grad(const float* __restrict__ q,
@@ -2850,6 +2918,7 @@ public:
     float *ct_h1, *ct_h2, *ct_h3, *p_t, *m_proj;    // critic_target path
     float *c_h1, *c_h2, *c_h3, *q;                  // critic path
     float *dlog, *d3, *d2, *d1, *da;                // critic backward deltas
+    float *logits;                                  // wide softmax scratch
     float *pa_h1, *pa_h2, *pa_h3, *a_out;           // actor (policy) path
     float *pc_h1, *pc_h2, *pc_h3, *pq;              // critic(s, actor(s))
     float *pd3, *pd2, *pdh1, *pda, *adz;            // policy backward deltas
@@ -2929,6 +2998,7 @@ public:
         c_h3 = carve<float>((long)B * H, off);
         q = carve<float>((long)B * K, off);
         dlog = carve<float>((long)B * K, off);
+        logits = carve<float>((long)B * K, off);
         d3 = carve<float>((long)B * H, off);
         d2 = carve<float>((long)B * H, off);
         d1 = carve<float>((long)B * H, off);
@@ -3008,15 +3078,25 @@ public:
     void launch_fwd(std::initializer_list<FwdJob> jobs) {
         FwdJob a[3] = {};
         int n = 0, wgs = 0;
-        bool all_mfma = cfg.batch >= 512;
-        for (auto& j : jobs)
-            all_mfma = all_mfma && mfma_eligible(j.in1 + j.in2, j.out, j.act);
-        if (all_mfma) {
+        if (cfg.batch >= 512) {
+            // every job goes to the matrix cores; softmax heads compute
+            // logits on MFMA then normalize with a wave-per-row epilogue
             for (auto& j : jobs) {
                 int ntm = ceil_div(j.B, MT_M), ntn = ceil_div(j.out, MT_N);
-                hipLaunchKernelGGL(k_mfma_fwd, dim3(ntm * ntn), dim3(256),
-                                   0, stream, j.x1, j.x2, j.wt, j.bias, j.y,
-                                   j.B, j.in1, j.in2, j.out, j.act);
+                if (j.act == ACT_SOFTMAX) {
+                    hipLaunchKernelGGL(k_mfma_fwd, dim3(ntm * ntn),
+                                       dim3(256), 0, stream, j.x1, j.x2,
+                                       j.wt, j.bias, logits, j.B, j.in1,
+                                       j.in2, j.out, ACT_NONE);
+                    hipLaunchKernelGGL(k_row_softmax,
+                                       dim3(ceil_div(j.B, 4)), dim3(256),
+                                       0, stream, logits, j.y, j.B, j.out);
+                } else {
+                    hipLaunchKernelGGL(k_mfma_fwd, dim3(ntm * ntn),
+                                       dim3(256), 0, stream, j.x1, j.x2,
+                                       j.wt, j.bias, j.y, j.B, j.in1,
+                                       j.in2, j.out, j.act);
+                }
             }
             return;
         }
@@ -3313,15 +3393,12 @@ public:
                                           cnet.l[2], ct_h3, ACT_RELU, wg)}); }
         { int wg = 0; launch_fwd({fwd_job(ct_h3, nullptr, p_critic_t,
                                           cnet.l[3], p_t, ACT_SOFTMAX, wg)}); }
-        // P9: projection
-        hipLaunchKernelGGL(k_project, dim3(row_wgs), dim3(256),
-                           waves_per_wg * K * sizeof(float), stream,
-                           p_t, br, bd, m_proj, B, K, cfg.v_min, cfg.v_max,
-                           cfg.gamma_n);
-        // P10: CE grad + priorities
-        hipLaunchKernelGGL(k_ce_grad, dim3(row_wgs), dim3(256), 0, stream,
-                           q, m_proj, bw, dlog, pri, cnt, B, K, cfg.per_eps,
-                           cfg.is_weighting);
+        // P9+P10 fused: projection + CE grad + priorities
+        hipLaunchKernelGGL(k_project_ce, dim3(row_wgs), dim3(256),
+                           waves_per_wg * 64 * sizeof(float), stream,
+                           p_t, br, bd, q, bw, m_proj, dlog, pri, cnt,
+                           B, K, cfg.v_min, cfg.v_max, cfg.gamma_n,
+                           cfg.per_eps, cfg.is_weighting);
         // P11-14: critic backward L4..L1
         launch_bwd(dlog, c_h3, nullptr, p_critic, g_critic, cnet.l[3],
                    d3, nullptr, c_h3, ACT_RELU, true);
