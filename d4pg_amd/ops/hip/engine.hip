@@ -3078,25 +3078,18 @@ public:
     void launch_fwd(std::initializer_list<FwdJob> jobs) {
         FwdJob a[3] = {};
         int n = 0, wgs = 0;
-        if (cfg.batch >= 512) {
-            // every job goes to the matrix cores; softmax heads compute
-            // logits on MFMA then normalize with a wave-per-row epilogue
+        bool all_mfma = cfg.batch >= 512;
+        for (auto& j : jobs)
+            all_mfma = all_mfma && mfma_eligible(j.in1 + j.in2, j.out, j.act);
+        if (all_mfma) {
+            // measured: a softmax head on MFMA (grid 64 wgs, 60% padded
+            // cols) loses to the wave-per-row k_fwd3 head — keep softmax
+            // groups on the per-layer kernel
             for (auto& j : jobs) {
                 int ntm = ceil_div(j.B, MT_M), ntn = ceil_div(j.out, MT_N);
-                if (j.act == ACT_SOFTMAX) {
-                    hipLaunchKernelGGL(k_mfma_fwd, dim3(ntm * ntn),
-                                       dim3(256), 0, stream, j.x1, j.x2,
-                                       j.wt, j.bias, logits, j.B, j.in1,
-                                       j.in2, j.out, ACT_NONE);
-                    hipLaunchKernelGGL(k_row_softmax,
-                                       dim3(ceil_div(j.B, 4)), dim3(256),
-                                       0, stream, logits, j.y, j.B, j.out);
-                } else {
-                    hipLaunchKernelGGL(k_mfma_fwd, dim3(ntm * ntn),
-                                       dim3(256), 0, stream, j.x1, j.x2,
-                                       j.wt, j.bias, j.y, j.B, j.in1,
-                                       j.in2, j.out, j.act);
-                }
+                hipLaunchKernelGGL(k_mfma_fwd, dim3(ntm * ntn), dim3(256),
+                                   0, stream, j.x1, j.x2, j.wt, j.bias, j.y,
+                                   j.B, j.in1, j.in2, j.out, j.act);
             }
             return;
         }
