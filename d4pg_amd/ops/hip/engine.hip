@@ -1597,7 +1597,9 @@ __device__ inline void p_bar(unsigned long long* gb,
             // safety valve: a barrier logic bug must never hard-hang the
             // GPU — desynchronize and terminate instead (~10 s)
             if (++spins > (1L << 26)) break;
-            if (spins < 8) __builtin_amdgcn_s_sleep(2);
+            // barriers resolve in ~2-5 us: stay on the fast poll long
+            // enough to catch that window before backing off
+            if (spins < 48) __builtin_amdgcn_s_sleep(2);
             else __builtin_amdgcn_s_sleep(32);
         }
         __threadfence();                       // acquire: invalidate L1
