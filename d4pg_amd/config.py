@@ -160,6 +160,9 @@ class D4PGConfig:
     cycles_per_epoch: int = 50
     eval_trials: int = 10
     seed: int = 0
+    lr_actor: float = 1e-4
+    lr_critic: float = 1e-3
+    vector_envs: int = 0
     extra: dict = field(default_factory=dict)
 
     @classmethod

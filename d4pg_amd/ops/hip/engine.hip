@@ -2,22 +2,35 @@
 //
 // The entire D4PG train step of the reference (/root/reference/ddpg.py:200-255
 // — PER sample, target forwards, C51 projection, critic CE backward, Adam,
-// policy backward, Adam, target soft-update, priority write-back) runs here
-// as a fixed sequence of ~34 HIP kernels on one stream, captured into a
-// hipGraph so N steps replay with no host dispatch.  All state is
-// device-resident: parameters in a flat slab (weights stored TRANSPOSED,
-// [in][out], for coalesced forward reads), Adam moments, target slabs, the
-// SoA replay store and the prioritized sum/min segment trees in HBM
-// (SURVEY.md §2c K12), philox4x32 RNG, and the beta/Adam step counters.
+// policy backward, Adam, target soft-update, priority write-back) runs
+// on-device.  All state is device-resident: parameters in a flat slab
+// (weights stored TRANSPOSED, [in][out], for coalesced forward reads),
+// Adam moments, target slabs, the SoA replay store and the prioritized
+// sum/min segment trees in HBM, philox4x32 RNG, and
+// the beta/Adam step counters.
 //
-// Regime note (why this shape): at B=64, H=256 every GEMM is tiny, so the
-// step is kernel-boundary-bound (~1.45us per dependent boundary on MI355X);
-// the design minimizes launches (fused dW+dX+db backward kernels,
-// multi-job forward kernels batching independent layers) rather than
-// chasing per-GEMM MFMA peaks.  The wide-batch benchmark config
-// (B=4096, H=1024) reuses the same kernels with larger grids.
+// Three execution paths, selected by batch size (Engine::enqueue_step):
 //
-// Numerics: fp32 end-to-end, matching the reference's CPU fp32.
+//  * PERSISTENT (B <= 256, the flagship regime): k_step_persistent — ONE
+//    kernel runs the whole step (and an nsteps loop of steps) as ~27
+//    layer-parallel phases over 64 co-resident workgroups separated by a
+//    software grid barrier (tree arrival + write-once go-flag).  Each
+//    GEMM/dX phase tiles 4 rows x 64 cols per workgroup with register-
+//    batched, double-buffered LDS staging; PER write-back and the next
+//    step's tree sampling hide under the dW/Adam phases.  At B=64 the
+//    step is latency-bound (PMC: 3% VALUBusy), so the design maximizes
+//    memory-level parallelism, not FLOPs.  k_step_chain (D4PG_CHAIN=1) is
+//    an experimental row-local chain-fused variant.
+//  * MFMA (B >= 512, the wide-batch config): per-layer f32-in matrix-core
+//    GEMMs (v_mfma_f32_32x32x2_f32 — exact fp32 at the full fp32 rate),
+//    64x128 workgroup tiles, double-buffered LDS K-slices, split-K dW
+//    with fused bias, per-level PER tree repair.
+//  * per-layer VALU kernels + hipGraph capture (fallback / middle sizes,
+//    incl. the row-block megakernels for 256 < B <= 512).
+//
+// Numerics: fp32 end-to-end, matching the reference's CPU fp32; per-thread
+// dot products accumulate in ascending-k order so every path agrees with
+// the eager torch oracle within fp32 tolerance.
 
 #include <hip/hip_runtime.h>
 #include <cstdint>

@@ -55,3 +55,14 @@ def test_run_dir_name_encodes_config():
     # (main.py:59-64)
     assert "Pendulum" in rd
     assert "5N" in rd or "5" in rd
+
+
+def test_d4pg_config_from_args_roundtrip():
+    from d4pg_amd.config import D4PGConfig
+    args = make_parser().parse_args(["--env", "Pendulum-v1", "--bsize",
+                                     "128", "--lr_critic", "5e-4"])
+    cfg = D4PGConfig.from_args(args)
+    assert cfg.bsize == 128
+    assert cfg.lr_critic == 5e-4
+    assert cfg.env == "Pendulum-v1"
+    assert cfg.extra == {}          # every CLI flag has a dataclass field
