@@ -329,8 +329,12 @@ class DistributedD4PG:
                       flush=True)
 
     def run(self, rounds: int, save: bool = False):
+        interval = max(1, int(getattr(self.args, "broadcast_interval", 1)))
         for rnd in range(rounds):
-            self._broadcast_params()
+            # --broadcast_interval K: refresh actor params every K rounds
+            # (bounded staleness knob; K=1 = every round)
+            if rnd % interval == 0:
+                self._broadcast_params()
             lb = self._collect()
             ingested = self._exchange(lb)
             if self.is_learner:
