@@ -99,3 +99,10 @@ def test_two_rank_vector_actors():
     # 8 envs x ~(30 - n_steps + 1) matured transitions per round x 4 rounds
     assert replay_len > 500
     assert step > 0
+
+
+def test_two_rank_broadcast_interval():
+    out = _run_world(2, extra=("--broadcast_interval", "2",
+                               "--max_steps", "30"), port=29619)
+    step, replay_len = out["learner"]
+    assert replay_len > 0 and step > 0
