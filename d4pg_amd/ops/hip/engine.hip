@@ -1207,6 +1207,157 @@ __device__ inline void mfma_pipeline(FA ldA, FB ldB, int nch,
     }
 }
 
+// ---------------------------------------------------------------------------
+// Big-GEMM variant: 128x256 workgroup tiles + split-K.
+// Tile-traffic accounting (NOTES.md): HBM bytes scale with
+// (1/MT_M + 1/MT_N), so 128x256 halves the weight re-streaming of 64x128
+// (402 -> 201 MB per 4096x1024x1024 GEMM) while split-K keeps the grid at
+// >= 256 workgroups for full MFMA issue. Partial sums land in a zeroed
+// scratch via fp32 atomics; a grid-stride epilogue applies bias+activation
+// (or the backward mask).
+#define M2_M 128
+#define M2_N 256
+
+template <typename FA, typename FB>
+__device__ inline void mfma_pipeline2(FA ldA, FB ldB, int nch,
+                                      float* As0, float* As1,
+                                      float* Bs0, float* Bs1,
+                                      f32x16 acc[2][4]) {
+    int tid = threadIdx.x, wid = tid >> 6, lane = tid & 63;
+    int wm0 = (wid >> 1) * 64, wn0 = (wid & 1) * 128;
+    int r = lane & 31, kk2 = lane >> 5;
+    float ta[16], tb[32];
+    ldA(0, ta);
+    ldB(0, tb);
+    auto wr = [&](float* As, float* Bs) {
+#pragma unroll
+        for (int u = 0; u < 16; ++u) {
+            int e = u * 256 + tid;
+            As[(e & 31) * (M2_M + 4) + (e >> 5)] = ta[u];
+        }
+#pragma unroll
+        for (int u = 0; u < 32; ++u) {
+            int e = u * 256 + tid;
+            Bs[(e >> 8) * (M2_N + 4) + (e & 255)] = tb[u];
+        }
+    };
+    wr(As0, Bs0);
+    __syncthreads();
+    for (int ch = 0; ch < nch; ++ch) {
+        float* As = (ch & 1) ? As1 : As0;
+        float* Bs = (ch & 1) ? Bs1 : Bs0;
+        bool more = ch + 1 < nch;
+        if (more) {
+            ldA((ch + 1) * MT_K, ta);
+            ldB((ch + 1) * MT_K, tb);
+        }
+#pragma unroll
+        for (int ks = 0; ks < MT_K; ks += 2) {
+            float a0 = As[(ks + kk2) * (M2_M + 4) + wm0 + r];
+            float a1 = As[(ks + kk2) * (M2_M + 4) + wm0 + 32 + r];
+            float b[4];
+#pragma unroll
+            for (int j = 0; j < 4; ++j)
+                b[j] = Bs[(ks + kk2) * (M2_N + 4) + wn0 + 32 * j + r];
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                acc[0][j] = __builtin_amdgcn_mfma_f32_32x32x2f32(
+                    a0, b[j], acc[0][j], 0, 0, 0);
+                acc[1][j] = __builtin_amdgcn_mfma_f32_32x32x2f32(
+                    a1, b[j], acc[1][j], 0, 0, 0);
+            }
+        }
+        if (more) wr((ch & 1) ? As0 : As1, (ch & 1) ? Bs0 : Bs1);
+        __syncthreads();
+    }
+}
+
+#define MFMA2_LDS_DECL \
+    __shared__ float As2[2][MT_K * (M2_M + 4)]; \
+    __shared__ float Bs2[2][MT_K * (M2_N + 4)]
+
+// partial C[B,out] += X-chunk @ W-chunk (split-K over `ksplit` segments,
+// fp32-atomic accumulation into pre-zeroed `part`)
+__global__ void __launch_bounds__(256, 1)
+k_mfma_fwd2(const float* __restrict__ x1, const float* __restrict__ x2,
+            const float* __restrict__ wt, float* __restrict__ part,
+            int B, int in1, int in2, int out, int ksplit) {
+    MFMA2_LDS_DECL;
+    int in_total = in1 + in2;
+    int ntm = (B + M2_M - 1) / M2_M;
+    int ntn = (out + M2_N - 1) / M2_N;
+    int seg = blockIdx.x / (ntm * ntn);
+    int tile = blockIdx.x % (ntm * ntn);
+    int m0 = (tile / ntn) * M2_M, n0 = (tile % ntn) * M2_N;
+    int nch_total = (in_total + MT_K - 1) / MT_K;
+    int nch_seg = (nch_total + ksplit - 1) / ksplit;
+    int ch0 = seg * nch_seg;
+    int nch = min(nch_seg, nch_total - ch0);
+    if (nch <= 0) return;
+    int k_base = ch0 * MT_K;
+    int tid = threadIdx.x, lane = tid & 63;
+    f32x16 acc[2][4] = {};
+    auto ldA = [&](int k0, float* t) {
+#pragma unroll
+        for (int u = 0; u < 16; ++u) {
+            int e = u * 256 + tid;
+            int gm = m0 + (e >> 5), gk = k_base + k0 + (e & 31);
+            float v = 0.f;
+            if (gm < B && gk < in_total)
+                v = (gk < in1) ? x1[(long)gm * in1 + gk]
+                               : x2[(long)gm * in2 + (gk - in1)];
+            t[u] = v;
+        }
+    };
+    auto ldB = [&](int k0, float* t) {
+#pragma unroll
+        for (int u = 0; u < 32; ++u) {
+            int e = u * 256 + tid;
+            int gk = k_base + k0 + (e >> 8), gn = n0 + (e & 255);
+            t[u] = (gk < in_total && gn < out)
+                ? wt[(long)gk * out + gn] : 0.f;
+        }
+    };
+    mfma_pipeline2(ldA, ldB, nch, As2[0], As2[1], Bs2[0], Bs2[1], acc);
+    int wid = tid >> 6;
+    int wm0 = (wid >> 1) * 64, wn0 = (wid & 1) * 128;
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+#pragma unroll
+            for (int reg = 0; reg < 16; ++reg) {
+                int row = (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+                int col = lane & 31;
+                int gm = m0 + wm0 + i * 32 + row;
+                int gn = n0 + wn0 + j * 32 + col;
+                if (gm < B && gn < out)
+                    atomicAdd(&part[(long)gm * out + gn], acc[i][j][reg]);
+            }
+        }
+    }
+}
+
+// epilogue: y = act(part + bias) (fwd) or dx = part * act_mask (dX)
+__global__ void k_mfma_epilogue(const float* __restrict__ part,
+                                const float* __restrict__ bias,
+                                const float* __restrict__ hprev,
+                                float* __restrict__ y, long n, int out,
+                                int act_kind, int prev_act) {
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += (long)gridDim.x * blockDim.x) {
+        float v = part[i];
+        if (bias) {
+            v += bias[i % out];
+            if (act_kind == ACT_RELU) v = fmaxf(v, 0.f);
+            else if (act_kind == ACT_TANH) v = tanhf(v);
+        } else if (hprev) {
+            v *= act_mask(prev_act, hprev[i]);
+        }
+        y[i] = v;
+    }
+}
+
 #define MFMA_LDS_DECL \
     __shared__ float As2[2][MT_K * (MT_M + 4)]; \
     __shared__ float Bs2[2][MT_K * (MT_N + 4)]
@@ -2934,6 +3085,7 @@ public:
     float *c_h1, *c_h2, *c_h3, *q;                  // critic path
     float *dlog, *d3, *d2, *d1, *da;                // critic backward deltas
     float *logits;                                  // wide softmax scratch
+    float *part_scratch;                            // split-K partial sums
     float *pa_h1, *pa_h2, *pa_h3, *a_out;           // actor (policy) path
     float *pc_h1, *pc_h2, *pc_h3, *pq;              // critic(s, actor(s))
     float *pd3, *pd2, *pdh1, *pda, *adz;            // policy backward deltas
@@ -3014,6 +3166,7 @@ public:
         q = carve<float>((long)B * K, off);
         dlog = carve<float>((long)B * K, off);
         logits = carve<float>((long)B * K, off);
+        part_scratch = carve<float>((long)B * (H + A), off);
         d3 = carve<float>((long)B * H, off);
         d2 = carve<float>((long)B * H, off);
         d1 = carve<float>((long)B * H, off);
@@ -3101,6 +3254,29 @@ public:
             // cols) loses to the wave-per-row k_fwd3 head — keep softmax
             // groups on the per-layer kernel
             for (auto& j : jobs) {
+                int in_total = j.in1 + j.in2;
+                if (j.B >= 2048 && j.out >= 256 && in_total >= 256) {
+                    // big GEMM: 128x256 tiles + split-K (half the weight
+                    // re-streaming of the 64x128 tiles; see NOTES.md)
+                    int ntm = ceil_div(j.B, M2_M);
+                    int ntn = ceil_div(j.out, M2_N);
+                    int ksplit = 1;
+                    while (ntm * ntn * ksplit < 256 &&
+                           ksplit * 2 * MT_K <= in_total)
+                        ksplit *= 2;
+                    HIP_CHECK(hipMemsetAsync(part_scratch, 0,
+                                             (long)j.B * j.out * 4, stream));
+                    hipLaunchKernelGGL(k_mfma_fwd2,
+                                       dim3(ntm * ntn * ksplit), dim3(256),
+                                       0, stream, j.x1, j.x2, j.wt,
+                                       part_scratch, j.B, j.in1, j.in2,
+                                       j.out, ksplit);
+                    hipLaunchKernelGGL(k_mfma_epilogue, dim3(1024),
+                                       dim3(256), 0, stream, part_scratch,
+                                       j.bias, (const float*)nullptr, j.y,
+                                       (long)j.B * j.out, j.out, j.act, 0);
+                    continue;
+                }
                 int ntm = ceil_div(j.B, MT_M), ntn = ceil_div(j.out, MT_N);
                 hipLaunchKernelGGL(k_mfma_fwd, dim3(ntm * ntn), dim3(256),
                                    0, stream, j.x1, j.x2, j.wt, j.bias, j.y,

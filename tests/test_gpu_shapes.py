@@ -66,6 +66,7 @@ def run_one_step(O, A, H, K, B, cap=4096, seed=3):
     (11, 1, 320, 64, 256),     # persistent upper edge, K == wave
     (5, 2, 192, 31, 300),      # row-block path (256 < B <= 512)
     (9, 4, 192, 41, 640),      # MFMA path, dims off the 64/128 tiles
+    (9, 4, 256, 41, 2048),     # big-GEMM path (128x256 tiles + split-K)
 ])
 def test_engine_shape_parity(O, A, H, K, B):
     run_one_step(O, A, H, K, B)
