@@ -3254,8 +3254,13 @@ public:
             // cols) loses to the wave-per-row k_fwd3 head — keep softmax
             // groups on the per-layer kernel
             for (auto& j : jobs) {
+                static const bool no_fwd2 = [] {
+                    const char* e = getenv("D4PG_NO_FWD2");
+                    return e && e[0] == '1';
+                }();
                 int in_total = j.in1 + j.in2;
-                if (j.B >= 2048 && j.out >= 256 && in_total >= 256) {
+                if (!no_fwd2 && j.B >= 2048 && j.out >= 256 &&
+                    in_total >= 256) {
                     // big GEMM: 128x256 tiles + split-K (half the weight
                     // re-streaming of the 64x128 tiles; see NOTES.md)
                     int ntm = ceil_div(j.B, M2_M);
