@@ -3254,12 +3254,17 @@ public:
             // cols) loses to the wave-per-row k_fwd3 head — keep softmax
             // groups on the per-layer kernel
             for (auto& j : jobs) {
-                static const bool no_fwd2 = [] {
-                    const char* e = getenv("D4PG_NO_FWD2");
+                // OPT-IN (D4PG_FWD2=1): measured 7% SLOWER than the
+                // 64x128 tiles (272 vs 294 steps/s same-box) — the weight
+                // operand is L2-resident so the halved "HBM traffic" was
+                // never paid, while the split-K memset+atomics+epilogue
+                // cost is real.  Kept for documentation/experimentation.
+                static const bool fwd2 = [] {
+                    const char* e = getenv("D4PG_FWD2");
                     return e && e[0] == '1';
                 }();
                 int in_total = j.in1 + j.in2;
-                if (!no_fwd2 && j.B >= 2048 && j.out >= 256 &&
+                if (fwd2 && j.B >= 2048 && j.out >= 256 &&
                     in_total >= 256) {
                     // big GEMM: 128x256 tiles + split-K (half the weight
                     // re-streaming of the 64x128 tiles; see NOTES.md)
