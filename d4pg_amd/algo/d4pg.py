@@ -51,7 +51,8 @@ class DDPG:
                  tau=0.001, prioritized_replay=True, critic_dist_info=None,
                  n_steps=1, *, device="cpu", backend="eager", hidden=256,
                  is_weighting=False, true_td_priorities=False, seed=None,
-                 noise_eps=0.3):
+                 noise="gaussian", noise_eps=0.3, ou_theta=0.15,
+                 ou_sigma=0.2, ou_mu=0.0):
         self.gamma = gamma
         self.n_steps = n_steps
         self.n_step_gamma = gamma ** n_steps
@@ -104,7 +105,15 @@ class DDPG:
         self.optimizer_global_actor = None
         self.optimizer_global_critic = None
 
-        self.noise = GaussianNoise(act_dim, eps=noise_eps, rng=rng)
+        # --noise selector: the reference declares OU flags but only ever
+        # constructs GaussianNoise (ddpg.py:74-75); here 'ou' actually
+        # wires --ou_theta/sigma/mu through
+        if noise == "ou":
+            from ..noise import OrnsteinUhlenbeckProcess
+            self.noise = OrnsteinUhlenbeckProcess(
+                act_dim, mu=ou_mu, theta=ou_theta, sigma=ou_sigma, rng=rng)
+        else:
+            self.noise = GaussianNoise(act_dim, eps=noise_eps, rng=rng)
 
         self.prioritized_replay = prioritized_replay
         if prioritized_replay:
@@ -351,6 +360,24 @@ class DDPG:
         return _copy.deepcopy(st)
 
     def load_state_dict(self, st: dict, load_replay: bool = True) -> None:
+        # A fresh backend='hip' agent restoring a GPU-format checkpoint must
+        # build the fused bridge FIRST: the on-HBM replay blob and engine
+        # slabs in `st` have nowhere to land otherwise, and the CPU replay
+        # buffer cannot parse the GPU-format replay dict.
+        if self.backend == "hip" and "engine" in st and self._fused is None:
+            from ..ops import build_fused_engine
+            self._fused = build_fused_engine(self)
+        if "replay" in st and load_replay:
+            gpu_format = "sum_tree" in st["replay"]
+            have_gpu_buf = hasattr(self.replayBuffer, "engine")
+            if gpu_format != have_gpu_buf:
+                raise ValueError(
+                    "checkpoint replay format (%s) does not match this "
+                    "agent's replay buffer (%s) — load with backend=%r or "
+                    "pass load_replay=False"
+                    % ("on-HBM" if gpu_format else "CPU",
+                       "on-HBM" if have_gpu_buf else "CPU",
+                       "hip" if gpu_format else "eager"))
         self.actor.load_state_dict(st["actor"])
         self.critic.load_state_dict(st["critic"])
         self.actor_target.load_state_dict(st["actor_target"])
@@ -372,6 +399,6 @@ class DDPG:
                 eng.load_slab(k, st["engine"][k])
             cnt = st["engine"]["counters"]
             if "seed" in st["engine"]:
-                eng.ext.set_seed(eng.h, int(st["engine"]["seed"]))
+                eng.set_seed(int(st["engine"]["seed"]))
             eng.set_schedule(cnt["adam_t_actor"], cnt["max_priority"])
             self._fused._params_dirty = False

@@ -80,6 +80,15 @@ def make_parser() -> argparse.ArgumentParser:
     p.add_argument("--lr_critic", default=1e-3, type=float,
                    help="critic Adam lr (reference global Adam 1e-3, "
                         "main.py:384)")
+    p.add_argument("--noise", default="gaussian", type=str,
+                   choices=["gaussian", "ou"],
+                   help="exploration noise process.  The reference ships "
+                        "--ou_theta/sigma/mu flags but never constructs the "
+                        "OU process (ddpg.py:74-75 commented out); here "
+                        "'ou' wires them up")
+    p.add_argument("--noise_eps", default=0.3, type=float,
+                   help="Gaussian noise scale epsilon (reference "
+                        "random_process.py:7 hardcodes 0.3)")
     p.add_argument("--vector_envs", default=0, type=int,
                    help="MI355X extension: batch M vectorized envs per "
                         "actor rank (one [M,obs] policy forward per tick) "
@@ -102,6 +111,18 @@ def configure_env_params(args) -> None:
     rng = ENV_VALUE_RANGES.get(args.env)
     if rng is not None:
         args.v_min, args.v_max = rng
+
+
+def noise_kwargs(args) -> dict:
+    """DDPG ctor kwargs selecting/parameterizing the exploration noise
+    from the CLI flags (--noise, --noise_eps, --ou_theta/sigma/mu)."""
+    return {
+        "noise": getattr(args, "noise", "gaussian"),
+        "noise_eps": getattr(args, "noise_eps", 0.3),
+        "ou_theta": getattr(args, "ou_theta", 0.15),
+        "ou_sigma": getattr(args, "ou_sigma", 0.2),
+        "ou_mu": getattr(args, "ou_mu", 0.0),
+    }
 
 
 def critic_dist_info(args) -> dict:
@@ -162,6 +183,8 @@ class D4PGConfig:
     seed: int = 0
     lr_actor: float = 1e-4
     lr_critic: float = 1e-3
+    noise: str = "gaussian"
+    noise_eps: float = 0.3
     vector_envs: int = 0
     extra: dict = field(default_factory=dict)
 

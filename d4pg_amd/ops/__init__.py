@@ -255,8 +255,17 @@ class FusedEngine:
         return self.ext.counters(self.h)
 
     def set_schedule(self, steps_done, max_priority=1.0):
-        """Restore schedule counters to 'steps_done completed' (resume)."""
+        """Restore schedule counters to 'steps_done completed' (resume).
+        Counters live in device memory (read by the kernels each step), so
+        no graph invalidation is needed here."""
         self.ext.set_schedule(self.h, int(steps_done), float(max_priority))
+
+    def set_seed(self, seed):
+        """Reseed the device philox stream.  The seed rides in captured
+        kernel arguments, so this invalidates any captured hipGraph and
+        forces a recapture on the next train_steps()."""
+        self.ext.set_seed(self.h, int(seed))
+        self._captured = 0
 
     def read(self, name):
         return self.ext.read_buffer(self.h, name)

@@ -1756,15 +1756,28 @@ __device__ inline void p_bar(unsigned long long* gb,
                                    __HIP_MEMORY_SCOPE_AGENT);
         }
         volatile unsigned long long* f = &gb[144];
+        volatile unsigned long long* err = &gb[176];  // timeout flag (own line)
         long spins = 0;
         while (*f < round) {
-            // safety valve: a barrier logic bug must never hard-hang the
-            // GPU — desynchronize and terminate instead (~10 s)
-            if (++spins > (1L << 26)) break;
+            // safety valve: a barrier logic bug (or non-co-resident grid)
+            // must never hard-hang the GPU.  On timeout (~10 s) raise the
+            // error flag and bail; other workgroups see the flag on their
+            // next poll-check and drain out instead of each spinning the
+            // full 10 s.  The host checks the flag after every step() /
+            // sync and throws — the step's results are NOT used.
+            if (++spins > (1L << 26)) {
+                atomicAdd((unsigned long long*)err, 1ull);
+                break;
+            }
             // barriers resolve in ~2-5 us: stay on the fast poll long
-            // enough to catch that window before backing off
+            // enough to catch that window before backing off.  The error
+            // check rides the slow path only (every 64th backoff poll, to
+            // keep poll traffic off the shared error line).
             if (spins < 48) __builtin_amdgcn_s_sleep(2);
-            else __builtin_amdgcn_s_sleep(32);
+            else {
+                __builtin_amdgcn_s_sleep(32);
+                if ((spins & 63) == 0 && *err) break;
+            }
         }
         __threadfence();                       // acquire: invalidate L1
     }
@@ -3104,7 +3117,36 @@ public:
         cnet = make_critic_net(c.obs, c.act, c.hidden, c.atoms);
         tree_cap = next_pow2(c.capacity);
         HIP_CHECK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
+        // co-residency gate for the persistent path: all PNWG workgroups
+        // must fit on the device at once or the software grid barrier
+        // cannot make progress
+        int dev = 0, per_cu = 0;
+        HIP_CHECK(hipGetDevice(&dev));
+        hipDeviceProp_t prop;
+        HIP_CHECK(hipGetDeviceProperties(&prop, dev));
+        HIP_CHECK(hipOccupancyMaxActiveBlocksPerMultiprocessor(
+            &per_cu, k_step_persistent, 256, 0));
+        persistent_fits_ = (long)per_cu * prop.multiProcessorCount >= PNWG;
+        HIP_CHECK(hipOccupancyMaxActiveBlocksPerMultiprocessor(
+            &per_cu, k_step_chain, 256, 0));
+        chain_fits_ = (long)per_cu * prop.multiProcessorCount >= PNWG;
         alloc();
+    }
+
+    bool persistent_fits_ = false;
+    bool chain_fits_ = false;
+
+    // the persistent grid barrier's timeout flag (gbar[176]); raised on
+    // device when a p_bar spins out, fatal host-side — the engine state
+    // must be considered corrupt after it
+    void check_bar_error() {
+        unsigned long long e = 0;
+        HIP_CHECK(hipMemcpy(&e, gbar + 176, 8, hipMemcpyDeviceToHost));
+        if (e)
+            throw std::runtime_error(
+                "persistent grid barrier timed out (desynchronized "
+                "workgroups; device state is corrupt) — occupancy check "
+                "passed but the grid did not make progress");
     }
 
     ~Engine() {
@@ -3369,8 +3411,12 @@ public:
     // ---------------- the train step (one launch sequence) ----------------
     bool use_persistent() const {
         // the persistent megakernel assumes: probe/row waves cover the batch
-        // (B <= PNWG*4), fan-ins fit the LDS x-stage, softmax <= one wave
-        return cfg.batch <= PNWG * 4 &&
+        // (B <= PNWG*4), fan-ins fit the LDS x-stage, softmax <= one wave,
+        // AND all PNWG workgroups co-resident (the software grid barrier
+        // deadlocks otherwise — persistent_fits_ is the ctor's occupancy
+        // check; on an occupancy-limited device we fall back to the
+        // row-block + hipGraph path instead of corrupting state)
+        return persistent_fits_ && cfg.batch <= PNWG * 4 &&
                cfg.obs <= PXMAX && cfg.hidden + cfg.act <= PXMAX &&
                cfg.hidden <= PXMAX && cfg.atoms <= 64;
     }
@@ -3418,7 +3464,7 @@ public:
         // dominates), so opt-in via D4PG_CHAIN=1 for further tuning.
         const char* e = getenv("D4PG_CHAIN");
         if (!e || e[0] != '1') return false;
-        return use_persistent() && cfg.hidden % 64 == 0 &&
+        return use_persistent() && chain_fits_ && cfg.hidden % 64 == 0 &&
                3 * ceil_div(cfg.batch, PR) <= PNWG && cfg.act <= 64;
     }
 
@@ -3681,7 +3727,8 @@ public:
     }
 
     void step(int n) {
-        if (use_persistent() && n > 0) {
+        bool persistent = use_persistent() && n > 0;
+        if (persistent) {
             // whole multi-step run in ONE launch — zero host dispatch
             // between steps (numerically identical to n single launches)
             enqueue_persistent(n);
@@ -3689,11 +3736,17 @@ public:
             for (int i = 0; i < n; ++i) enqueue_step();
         }
         HIP_CHECK(hipStreamSynchronize(stream));
+        if (persistent) check_bar_error();
+    }
+
+    void invalidate_graph() {
+        if (graph_exec) { hipGraphExecDestroy(graph_exec); graph_exec = nullptr; }
+        if (graph) { hipGraphDestroy(graph); graph = nullptr; }
+        graph_steps = 0;
     }
 
     void capture(int steps_per_graph) {
-        if (graph_exec) { hipGraphExecDestroy(graph_exec); graph_exec = nullptr; }
-        if (graph) { hipGraphDestroy(graph); graph = nullptr; }
+        invalidate_graph();
         HIP_CHECK(hipStreamBeginCapture(stream, hipStreamCaptureModeThreadLocal));
         for (int i = 0; i < steps_per_graph; ++i) enqueue_step();
         HIP_CHECK(hipStreamEndCapture(stream, &graph));

@@ -128,7 +128,14 @@ static void ingest(int64_t h, torch::Tensor s, torch::Tensor a,
 // completed" (adam/rng pre-advanced to steps_done+1, beta_t = steps_done —
 // the same convention Engine::alloc establishes at step 0)
 static void set_seed(int64_t h, int64_t seed) {
-    get(h).cfg.seed = (uint64_t)seed;
+    Engine& e = get(h);
+    e.cfg.seed = (uint64_t)seed;
+    // the seed is baked into captured kernel ARGUMENTS (k_per_sample /
+    // k_step_persistent take it by value), so a previously captured
+    // hipGraph would silently keep sampling with the stale seed —
+    // invalidate it; FusedEngine.set_seed resets _captured so the next
+    // train_steps() recaptures with the new seed.
+    e.invalidate_graph();
 }
 
 static void set_schedule(int64_t h, int64_t steps_done, double max_priority) {

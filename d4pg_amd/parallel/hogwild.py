@@ -26,11 +26,12 @@ from .worker import Worker, global_model_eval
 
 def _build_agent(args, env, seed):
     obs_dim, act_dim = obs_act_dims(env, her=bool(args.her))
+    from ..config import noise_kwargs
     return DDPG(obs_dim, act_dim, env=env, memory_size=args.rmsize,
                 batch_size=args.bsize, gamma=args.gamma, tau=args.tau,
                 prioritized_replay=bool(args.p_replay),
                 critic_dist_info=critic_dist_info(args),
-                n_steps=args.n_steps, seed=seed)
+                n_steps=args.n_steps, seed=seed, **noise_kwargs(args))
 
 
 def _worker_main(name, args, global_model, global_count, opt_actor,

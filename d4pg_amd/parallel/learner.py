@@ -43,7 +43,7 @@ import torch
 import torch.distributed as dist
 
 from ..algo.d4pg import DDPG
-from ..config import critic_dist_info, run_dir_name
+from ..config import critic_dist_info, noise_kwargs, run_dir_name
 from ..envs import make, obs_act_dims
 from ..her import add_experience, rollout_episode
 from ..ops import pack_net, unpack_net
@@ -170,7 +170,7 @@ class DistributedD4PG:
             prioritized_replay=bool(args.p_replay) and self.is_learner,
             critic_dist_info=critic_dist_info(args), n_steps=args.n_steps,
             device=str(self.device) if self.is_learner else "cpu",
-            backend=backend, seed=seed)
+            backend=backend, seed=seed, **noise_kwargs(args))
         self.rng = np.random.default_rng(seed)
 
         # wire geometry: per-round per-rank transition cap.  HER can add up
@@ -191,6 +191,18 @@ class DistributedD4PG:
                                          horizon=args.max_steps)
             self.vfold = VecNStep(m, self.obs_dim, self.act_dim,
                                   args.n_steps, args.gamma)
+        # vector-actor exploration noise honors --noise/--noise_eps/--ou_*
+        # (this path used to hardcode eps=0.3).  OU
+        # runs batched over [M, act]; Gaussian stays a one-liner below.
+        self.noise_eps = float(getattr(args, "noise_eps", 0.3))
+        self.vec_noise = None
+        if (self.vector is not None
+                and getattr(args, "noise", "gaussian") == "ou"):
+            from ..noise import OrnsteinUhlenbeckProcess
+            self.vec_noise = OrnsteinUhlenbeckProcess(
+                (self.vector.n, self.act_dim), mu=args.ou_mu,
+                theta=args.ou_theta, sigma=args.ou_sigma,
+                rng=np.random.default_rng(seed + 29))
         if int(getattr(args, "vector_envs", 0)) > 0 and not args.her \
                 and args.env.startswith("Pendulum"):
             # every rank must agree on the wire size
@@ -209,6 +221,8 @@ class DistributedD4PG:
         # consecutive rounds is flagged).
         self.last_seen = {r: 0 for r in range(1, self.world)}
         self.heartbeat_warn_rounds = 10
+        # wire-cap overflow accounting (see _exchange)
+        self.dropped_transitions = 0
 
     # -- round phases -----------------------------------------------------
 
@@ -243,11 +257,17 @@ class DistributedD4PG:
             obs = self.vector.reset()
             self.vfold.reset()
             actor = self.agent.actor
+            if self.vec_noise is not None:
+                self.vec_noise.reset()
             with _t.no_grad():
                 for t in range(self.vector.horizon):
                     a = actor(_t.from_numpy(obs)).numpy()
-                    a = np.clip(a + 0.3 * self.rng.standard_normal(a.shape),
-                                -1.0, 1.0).astype(np.float32)
+                    if self.vec_noise is not None:   # batched OU [M, act]
+                        a = a + self.vec_noise.sample()
+                    else:
+                        a = a + (self.noise_eps
+                                 * self.rng.standard_normal(a.shape))
+                    a = np.clip(a, -1.0, 1.0).astype(np.float32)
                     obs2, r, done = self.vector.step(a)
                     out = self.vfold.push(obs, a, r, obs2, done)
                     if out is not None:
@@ -265,12 +285,24 @@ class DistributedD4PG:
 
     def _exchange(self, lb):
         dev = self.comm_device
-        cnt = torch.tensor([float(len(lb))], device=dev)
+        buf, n = _encode(lb, self.push_cap, self.obs_dim,
+                         self.act_dim, dev)
+        # the advertised count must match what _encode actually shipped:
+        # len(lb) > push_cap means the wire buffer truncated (e.g. an env /
+        # HER change emitting more transitions per round than the cap was
+        # sized for) — count the drop and say so instead of letting decode
+        # and counts silently disagree.
+        if len(lb) > n:
+            self.dropped_transitions += len(lb) - n
+            print(f"[rank {self.rank}] WARNING: push_cap {self.push_cap} "
+                  f"truncated {len(lb) - n} of {len(lb)} transitions this "
+                  f"round ({self.dropped_transitions} dropped total) — "
+                  f"raise --episodes_per_cycle sizing or vector_envs cap",
+                  flush=True)
+        cnt = torch.tensor([float(n)], device=dev)
         counts = [torch.zeros_like(cnt) for _ in range(self.world)]
         dist.all_gather(counts, cnt)
         self._last_counts = counts
-        buf, n = _encode(lb, self.push_cap, self.obs_dim,
-                         self.act_dim, dev)
         blocks = [torch.zeros_like(buf) for _ in range(self.world)]
         dist.all_gather(blocks, buf)
         if not self.is_learner:

@@ -10,8 +10,9 @@ New design, not a translation: the trees are flat numpy arrays updated in
 tree level instead of per-element Python recursion) and sampling is a
 *batched prefix descent* — all B probes walk the tree together, one
 vectorized level per iteration.  This is the exact algorithm the on-HBM HIP
-tree (ops/hip/sumtree.hip, K12 in SURVEY.md §2c) implements with one wave
-per probe, so the CPU path doubles as its parity oracle.
+tree (k_per_sample / k_per_update / k_per_leaves in ops/hip/engine.hip, K12
+in SURVEY.md §2c) implements with one wave per probe, so the CPU path
+doubles as its parity oracle.
 
 Deviation (documented): the reference samples mass in
 ``random() * sum(0, len-1)`` whose exclusive ``end`` drops the newest element

@@ -4,8 +4,9 @@ Capability parity with /root/reference/replay_memory.py:4-80 (ring buffer,
 random minibatch, random-policy n-step prefill via ``initialize``), but a new
 design: structure-of-arrays storage with preallocated float32 numpy arrays
 and fully vectorized gather — the same SoA layout the on-HBM GPU replay
-(ops/hip/sumtree.hip + replay/hbm.py) uses, so transitions move host<->device
-as flat contiguous slabs.
+(the rs/ra/rr/rs2/rd slabs in ops/hip/engine.hip, fronted by
+ops.GPUReplayAdapter) uses, so transitions move host<->device as flat
+contiguous slabs.
 
 API surface kept: ``Replay(max_size, env=None, n_steps=1, gamma=0.99)`` with
 ``add(state, action, reward, next_state, done)``, ``initialize(init_length)``

@@ -41,13 +41,14 @@ def main(argv=None):
     backend = args.backend
     if backend == "auto":
         backend = "hip" if device == "cuda" else "eager"
+    from d4pg_amd.config import noise_kwargs
     agent = DDPG(obs_dim, act_dim, env=env, memory_size=args.rmsize,
                  batch_size=args.bsize, gamma=args.gamma, tau=args.tau,
                  lr_actor=args.lr_actor, lr_critic=args.lr_critic,
                  prioritized_replay=bool(args.p_replay),
                  critic_dist_info=critic_dist_info(args),
                  n_steps=args.n_steps, device=device, backend=backend,
-                 seed=args.seed)
+                 seed=args.seed, **noise_kwargs(args))
     rd = run_dir_name(args)
     writer = SummaryWriter(rd)
     worker = Worker("1", args, agent, env, writer=writer, run_dir=rd)
