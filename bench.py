@@ -5,11 +5,28 @@
 training step — the Pendulum-v1 D4PG learner (51-atom C51 critic,
 n-step=5, prioritized replay, batch 64) of BASELINE.json — on N GPUs of
 one node, one rank per GPU (launched via torch.distributed.run for N>1),
-on synthetic transitions and random-init weights.  Each rank owns a full
-fused HIP learner engine (weak scaling: per-GPU work fixed); the value
-reported is the WHOLE-JOB aggregate learner grad-steps/sec over all
-ranks, using the MAX elapsed time over ranks for EXACTLY K timed steps
-bracketed by barrier + torch.cuda.synchronize on both sides.
+on synthetic transitions and random-init weights.
+
+N>1 is COMMUNICATION-BEARING weak scaling (not independent engines): the
+default mode is the HogWild re-expression over RCCL — every rank owns a
+full fused HIP learner engine and its own replay shard (the reference's
+replay is per-worker too, main.py:188-197) and all ranks start from the
+same broadcast parameters; every --sync_every local steps the parameter
+slabs (actor, critic, both targets) are all-reduce-AVERAGED over xGMI and
+a block of synthetic transitions is all_gathered and ingested by every
+rank (the DistributedD4PG wire path, parallel/learner.py:_exchange).
+Both collectives sit INSIDE the timed region.  This replaces the
+reference's unbounded HogWild staleness (shared-memory params,
+ddpg.py:104-120) with a bounded sync_every-step window; per-step gradient
+sync (--mode dp, parallel/dp.py) is also available but serializes the
+0.3 ms flagship step behind two collective latencies, so the bounded-
+staleness mode is the default — see SURVEY.md §2b's xGMI latency note.
+
+The value reported is the WHOLE-JOB aggregate learner grad-steps/sec:
+each rank performs K real optimizer steps (HogWild accounting, same as
+the reference's shared global_count, main.py:307), timed over the MAX
+elapsed across ranks for EXACTLY K steps bracketed by barrier +
+torch.cuda.synchronize on both sides.
 
 Reference parity: the train step is the full D4PG update of
 /root/reference/ddpg.py:200-255 (PER sample + IS weights, target
@@ -28,6 +45,7 @@ import time
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
+import numpy as np
 import torch
 
 FLAGSHIP = dict(obs_dim=3, act_dim=1, hidden=256, n_atoms=51, batch=64,
@@ -42,16 +60,20 @@ def _dist_init(args):
     local = int(os.environ.get("LOCAL_RANK", str(rank)))
     if world > 1:
         import torch.distributed as dist
-        backend = os.environ.get("BENCH_DIST_BACKEND") or (
-            "nccl" if torch.cuda.is_available() else "gloo")
+        backend = os.environ.get("BENCH_DIST_BACKEND")
+        if backend is None:
+            # nccl(=RCCL) needs one DISTINCT GPU per rank; a single-GPU
+            # box rehearsing world>1 exchanges over gloo instead (engines
+            # still run on the GPU)
+            n_gpu = (torch.cuda.device_count()
+                     if torch.cuda.is_available() else 0)
+            backend = "nccl" if n_gpu >= world else "gloo"
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29511")
         try:
             dist.init_process_group(backend=backend, rank=rank,
                                     world_size=world)
         except Exception:
-            # the collectives here only synchronize timing (weak scaling,
-            # independent learners) — gloo is a safe fallback
             if backend != "gloo":
                 dist.init_process_group(backend="gloo", rank=rank,
                                         world_size=world)
@@ -103,7 +125,6 @@ def make_gpu_engine(seed: int):
 def make_cpu_agent(seed: int):
     """Eager CPU fallback so `python bench.py` runs in a GPU-less
     container (numbers are then CPU numbers, not the MI355X headline)."""
-    import numpy as np
     from d4pg_amd.algo.d4pg import DDPG
     cfg = FLAGSHIP
     agent = DDPG(cfg["obs_dim"], cfg["act_dim"],
@@ -125,11 +146,111 @@ def make_cpu_agent(seed: int):
     return agent
 
 
+def _synth_block(rng, n, obs_dim, act_dim):
+    """One rank's synthetic transition block for the wire exchange
+    (flat [n, 2*obs+act+2] rows, the _encode layout of parallel/learner)."""
+    w = 2 * obs_dim + act_dim + 2
+    arr = rng.standard_normal((n, w)).astype(np.float32)
+    arr[:, obs_dim + act_dim] = -rng.random(n)          # rewards
+    arr[:, -1] = 0.0                                    # dones
+    return arr
+
+
+class _WireExchange:
+    """The actor->replay wire path inside the timed region: all_gather a
+    fixed-size synthetic transition block from every rank, ingest all of
+    them (mirrors DistributedD4PG._exchange over the same collectives)."""
+
+    def __init__(self, dist_mod, world, rank, eng, block=512):
+        self.dist = dist_mod
+        self.world = world
+        self.eng = eng
+        cfg = FLAGSHIP
+        self.o, self.a = cfg["obs_dim"], cfg["act_dim"]
+        self.block = block
+        self.rng = np.random.default_rng(10_000 + rank)
+        use_cuda = (torch.cuda.is_available()
+                    and dist_mod.get_backend() == "nccl")
+        self.dev = torch.device("cuda" if use_cuda else "cpu")
+
+    def round(self):
+        arr = _synth_block(self.rng, self.block, self.o, self.a)
+        buf = torch.from_numpy(arr).to(self.dev)
+        blocks = [torch.empty_like(buf) for _ in range(self.world)]
+        self.dist.all_gather(blocks, buf)
+        o, a = self.o, self.a
+        for b in blocks:
+            rows = b.cpu().numpy()
+            self.eng.ingest(torch.from_numpy(rows[:, :o]),
+                            torch.from_numpy(rows[:, o:o + a]),
+                            torch.from_numpy(rows[:, o + a]),
+                            torch.from_numpy(rows[:, o + a + 1:2 * o + a + 1]),
+                            torch.from_numpy(rows[:, 2 * o + a + 1]))
+
+
+def run_gpu_multirank(dist_mod, world, rank, args):
+    """Communication-bearing weak scaling: local-SGD parameter averaging +
+    transition all_gather every --sync_every steps (module docstring)."""
+    from d4pg_amd.parallel.dp import DPEngine, LocalSGDSync
+    eng = make_gpu_engine(seed=1000 + rank)
+    K, W, S = args.steps, args.warmup, max(1, args.sync_every)
+    mode = args.mode
+    if mode == "auto":
+        mode = "localsgd"
+
+    if mode == "dp":
+        dp = DPEngine(eng)
+        dp.train_steps(max(1, W))
+        _barrier(dist_mod)
+        t0 = time.perf_counter()
+        dp.train_steps(K)
+        torch.cuda.synchronize()
+        elapsed = time.perf_counter() - t0
+        _barrier(dist_mod)
+        return elapsed, "dp%d (sync grad all-reduce per step)" % world
+
+    if mode == "indep":
+        spg = args.steps_per_graph
+        eng.train_steps(max(W, spg), steps_per_graph=spg)
+        _barrier(dist_mod)
+        t0 = time.perf_counter()
+        eng.train_steps(K, steps_per_graph=spg)
+        torch.cuda.synchronize()
+        elapsed = time.perf_counter() - t0
+        _barrier(dist_mod)
+        return elapsed, "indep%d (no communication; A/B baseline)" % world
+
+    # localsgd (default): bounded-staleness HogWild over RCCL
+    sync = LocalSGDSync(eng)
+    wire = _WireExchange(dist_mod, world, rank, eng) \
+        if args.xfer_transitions else None
+    sync.broadcast_initial(src=0)
+
+    def run_rounds(nsteps):
+        done = 0
+        while done < nsteps:
+            n = min(S, nsteps - done)
+            eng.step(n)                      # n local steps, one launch
+            sync.average()                   # param all-reduce over xGMI
+            if wire is not None:
+                wire.round()                 # transition all_gather+ingest
+            done += n
+
+    run_rounds(max(W, S))
+    _barrier(dist_mod)
+    t0 = time.perf_counter()
+    run_rounds(K)
+    torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    _barrier(dist_mod)
+    return elapsed, ("hogwild-rccl x%d (param avg + transition all_gather "
+                     "every %d steps)" % (world, S))
+
+
 def measure_env_steps_per_sec(seed: int, n: int = 3000) -> float:
     """Auxiliary (untimed-region) metric: actor-side env-steps/sec — native
     Pendulum dynamics + B=1 actor inference + exploration noise, the per-env-
     step path of /root/reference/main.py:142-152."""
-    import numpy as np
     from d4pg_amd.envs import make
     from d4pg_amd.models import actor
     from d4pg_amd.noise import GaussianNoise
@@ -154,7 +275,6 @@ def measure_env_steps_vector(seed: int, m: int = 64, ticks: int = 1500):
     """Aux metric: vectorized-actor env-steps/sec — M batched native
     Pendulums + one [M,obs] policy forward per tick (the MI355X-native
     actor mode, parallel/learner.py --vector_envs)."""
-    import numpy as np
     from d4pg_amd.envs.vector import VectorPendulum
     from d4pg_amd.models import actor
     env = VectorPendulum(m, seed=seed)
@@ -181,6 +301,16 @@ def main():
     ap.add_argument("--steps", type=int, default=4000)
     ap.add_argument("--warmup", type=int, default=400)
     ap.add_argument("--steps-per-graph", type=int, default=32)
+    ap.add_argument("--mode", default="auto",
+                    choices=["auto", "localsgd", "dp", "indep"],
+                    help="N>1 coupling: localsgd = bounded-staleness "
+                         "HogWild over RCCL (default), dp = per-step grad "
+                         "all-reduce, indep = no communication (A/B only)")
+    ap.add_argument("--sync_every", type=int, default=16,
+                    help="localsgd: local steps between param averages")
+    ap.add_argument("--xfer_transitions", type=int, default=1,
+                    help="localsgd: all_gather+ingest a synthetic "
+                         "transition block every sync round")
     args = ap.parse_args()
 
     dist, world, rank, local = _dist_init(args)
@@ -192,7 +322,10 @@ def main():
         torch.cuda.set_device(local % torch.cuda.device_count())
 
     K, W = args.steps, args.warmup
-    if use_gpu:
+    parallelism = "dp1 (one learner per GPU)"
+    if use_gpu and world > 1:
+        elapsed, parallelism = run_gpu_multirank(dist, world, rank, args)
+    elif use_gpu:
         eng = make_gpu_engine(seed=1000 + rank)
         spg = args.steps_per_graph
         eng.train_steps(max(W, spg), steps_per_graph=spg)  # warmup+capture
@@ -206,6 +339,12 @@ def main():
         K = min(K, 400)
         W = min(W, 20)
         agent = make_cpu_agent(seed=1000 + rank)
+        if world > 1:
+            # CPU rehearsal of the multi-rank topology: eager agents with
+            # per-step gradient averaging over gloo (parallel/dp.py hook)
+            from d4pg_amd.parallel.dp import eager_grad_sync
+            agent.grad_sync = eager_grad_sync()
+            parallelism = "dp%d (eager, gloo rehearsal)" % world
         for _ in range(W):
             agent.train()
         _barrier(dist)
@@ -220,9 +359,16 @@ def main():
     vec_sps = measure_env_steps_vector(seed=4321 + rank) if rank == 0 else 0.0
 
     if rank == 0:
+        # In every mode each rank performs K real optimizer steps, so the
+        # whole-job aggregate is N*K grad steps (HogWild accounting — the
+        # reference's shared global_count sums worker steps, main.py:307).
+        # EXCEPT dp mode: there the N per-step gradients merge into ONE
+        # global step at global batch N*B, so the job made K steps.
+        is_dp = parallelism.startswith("dp") and world > 1
+        total_steps = K if is_dp else n_gpus * K
         out = {
             "metric": "learner grad-steps/sec (Pendulum-v1 D4PG)",
-            "value": n_gpus * K / elapsed,
+            "value": total_steps / elapsed,
             "unit": "grad_steps/s",
             "n_gpus": n_gpus,
             "steps": K,
@@ -236,9 +382,10 @@ def main():
                     "transitions; reference publishes no numbers)",
             "config": {"model": "D4PG Pendulum-v1 (obs3/act1, 4x256 MLP "
                                 "actor+critic, 51 atoms)",
-                       "global_batch": FLAGSHIP["batch"] * n_gpus,
+                       "global_batch": FLAGSHIP["batch"] *
+                                       (n_gpus if is_dp else 1),
                        "seq_len": None,
-                       "parallelism": f"dp{n_gpus} (one learner per GPU)",
+                       "parallelism": parallelism,
                        "n_step": FLAGSHIP["n_steps"],
                        "prioritized_replay": True,
                        "replay_capacity": FLAGSHIP["capacity"],

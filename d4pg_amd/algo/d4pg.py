@@ -128,6 +128,12 @@ class DDPG:
 
         self._fused = None   # lazily-built HIP fused-step engine bridge
         self.train_steps_done = 0
+        # learner-DP hook (parallel/dp.py): called with the module whose
+        # .grad tensors are about to be consumed by the optimizer, right
+        # after backward — the DP wrapper all-reduce-averages them there,
+        # preserving the reference's update order (critic Adam BEFORE the
+        # policy forward, ddpg.py:232/236).
+        self.grad_sync = None
 
     @property
     def engine(self):
@@ -276,6 +282,8 @@ class DDPG:
             self.copy_gradients(self.critic, global_model.critic)
             self.optimizer_global_critic.step()
         else:
+            if self.grad_sync is not None:
+                self.grad_sync(self.critic)
             self.optimizer_critic.step()
 
         # -- actor update --
@@ -291,6 +299,8 @@ class DDPG:
             self.optimizer_global_actor.step()
             self.sync_local_global(global_model)
         else:
+            if self.grad_sync is not None:
+                self.grad_sync(self.actor)
             self.optimizer_actor.step()
 
         self.update_target_parameters()

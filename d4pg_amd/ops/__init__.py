@@ -193,6 +193,14 @@ class FusedEngine:
              "g_actor": 4, "g_critic": 5, "m_actor": 6, "v_actor": 7,
              "m_critic": 8, "v_critic": 9}
 
+    # split-step phase masks (engine.hip enum): the learner-DP path runs
+    # GRADS, all-reduces the grad slab over RCCL, then runs APPLY.
+    PH_CRITIC_GRADS = 1
+    PH_CRITIC_APPLY = 2
+    PH_ACTOR_GRADS = 4
+    PH_ACTOR_APPLY = 8
+    PH_ALL = 15
+
     def load_from_modules(self, actor, actor_target, critic, critic_target):
         self.ext.load_slab(self.h, 0, pack_net(actor))
         self.ext.load_slab(self.h, 1, pack_net(actor_target))
@@ -218,10 +226,21 @@ class FusedEngine:
     def ingest(self, s, a, r, s2, d):
         self.ext.ingest(self.h, s, a, r, s2, d)
 
+    def device_slab(self, name) -> torch.Tensor:
+        """Zero-copy torch CUDA view of an engine slab — feed directly to
+        torch.distributed collectives (RCCL).  Aliases engine memory."""
+        return self.ext.device_tensor(self.h, self.SLABS[name])
+
     # -- stepping --
     def step(self, n=1):
         """Uncaptured (eager-launch) steps — used by parity tests."""
         self.ext.step(self.h, int(n))
+
+    def step_part(self, mask):
+        """Run only the PH_* phases of one train step (synchronizing), so
+        the caller can all-reduce gradient slabs between GRADS and APPLY
+        (learner data parallelism, SURVEY §2c collectives list)."""
+        self.ext.step_part(self.h, int(mask))
 
     def train_steps(self, n=1, steps_per_graph=8):
         """Graph-replayed steps: captures once (steps_per_graph per replay),

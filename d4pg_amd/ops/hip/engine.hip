@@ -3130,8 +3130,11 @@ public:
         HIP_CHECK(hipOccupancyMaxActiveBlocksPerMultiprocessor(
             &per_cu, k_step_chain, 256, 0));
         chain_fits_ = (long)per_cu * prop.multiProcessorCount >= PNWG;
+        device_ = dev;
         alloc();
     }
+
+    int device_ = 0;
 
     bool persistent_fits_ = false;
     bool chain_fits_ = false;
@@ -3522,17 +3525,24 @@ public:
         return j;
     }
 
-    void enqueue_step_rowblock() {
+    // Phase mask for split stepping (learner data parallelism: the host
+    // all-reduces g_critic / g_actor between the GRADS and APPLY phases,
+    // SURVEY §2c collectives list).  PH_ALL == a whole train step.
+    enum { PH_CRITIC_GRADS = 1, PH_CRITIC_APPLY = 2,
+           PH_ACTOR_GRADS = 4, PH_ACTOR_APPLY = 8, PH_ALL = 15 };
+
+    void enqueue_step_rowblock(int mask = PH_ALL) {
         const int B = cfg.batch, H = cfg.hidden;
         int row_wgs = ceil_div(B, 4);
         long smem = rb_lds_bytes(H);
 
+        RowBlockArgs g = rb_args();
+        if (mask & PH_CRITIC_GRADS) {
         hipLaunchKernelGGL(k_per_sample, dim3(row_wgs), dim3(256), 0, stream,
             sum_tree, min_tree, tree_cap, rs, ra, rr, rs2, rd,
             cfg.obs, cfg.act, bs, ba, br, bs2, bd, bw, bidx, B, cnt,
             cfg.per_beta0, (float)cfg.per_beta_iters, cfg.seed);
 
-        RowBlockArgs g = rb_args();
         hipLaunchKernelGGL(k_critic_rowblock, dim3(row_wgs), dim3(256),
                            smem, stream, g);
         {
@@ -3548,10 +3558,13 @@ public:
             hipLaunchKernelGGL(k_bwd4, dim3(wg), dim3(256), 0, stream,
                                j0, j1, j2, j3);
         }
+        }
+        if (mask & PH_CRITIC_APPLY)
         hipLaunchKernelGGL(k_adam_lerp, dim3(256), dim3(256), 0, stream,
                            p_critic, g_critic, m_critic, v_critic,
                            p_critic_t, cnet.n_params, cfg.lr_critic,
                            0.9f, 0.999f, 1e-8f, cfg.tau, cnt, 0);
+        if (mask & PH_ACTOR_GRADS) {
         hipLaunchKernelGGL(k_policy_rowblock, dim3(row_wgs), dim3(256),
                            smem, stream, g);
         {
@@ -3567,6 +3580,8 @@ public:
             hipLaunchKernelGGL(k_bwd4, dim3(wg), dim3(256), 0, stream,
                                j0, j1, j2, j3);
         }
+        }
+        if (mask & PH_ACTOR_APPLY) {
         hipLaunchKernelGGL(k_adam_lerp, dim3(256), dim3(256), 0, stream,
                            p_actor, g_actor, m_actor, v_actor,
                            p_actor_t, anet.n_params, cfg.lr_actor,
@@ -3574,15 +3589,19 @@ public:
         hipLaunchKernelGGL(k_per_update, dim3(1), dim3(256), 0, stream,
                            sum_tree, min_tree, tree_cap, bidx, pri, B,
                            cfg.per_alpha, cnt);
+        }
     }
 
-    void enqueue_step() {
-        if (use_persistent()) { enqueue_persistent(1); return; }
-        if (use_row_block()) { enqueue_step_rowblock(); return; }
+    void enqueue_step(int mask = PH_ALL) {
+        if (use_persistent() && mask == PH_ALL) {
+            enqueue_persistent(1); return;
+        }
+        if (use_row_block()) { enqueue_step_rowblock(mask); return; }
         const int B = cfg.batch, K = cfg.atoms, H = cfg.hidden;
         const int waves_per_wg = 4;
         int row_wgs = ceil_div(B, waves_per_wg);
 
+        if (mask & PH_CRITIC_GRADS) {
         hipLaunchKernelGGL(k_per_sample, dim3(row_wgs), dim3(256), 0, stream,
             sum_tree, min_tree, tree_cap, rs, ra, rr, rs2, rd,
             cfg.obs, cfg.act, bs, ba, br, bs2, bd, bw, bidx, B, cnt,
@@ -3648,11 +3667,14 @@ public:
                    d1, nullptr, c_h1, ACT_RELU, true);
         launch_bwd(d1, bs, nullptr, p_critic, g_critic, cnet.l[0],
                    nullptr, nullptr, nullptr, ACT_NONE, true);
+        }
         // P15: Adam critic + target soft-update fused
+        if (mask & PH_CRITIC_APPLY)
         hipLaunchKernelGGL(k_adam_lerp, dim3(256), dim3(256), 0, stream,
                            p_critic, g_critic, m_critic, v_critic,
                            p_critic_t, cnet.n_params, cfg.lr_critic,
                            0.9f, 0.999f, 1e-8f, cfg.tau, cnt, 0);
+        if (mask & PH_ACTOR_GRADS) {
         // P16: actor.L1(s) | critic'.L1(s)
         {
             int wg = 0;
@@ -3699,6 +3721,8 @@ public:
                    pd2, nullptr, pa_h1, ACT_RELU, true);
         launch_bwd(pd2, bs, nullptr, p_actor, g_actor, anet.l[0],
                    nullptr, nullptr, nullptr, ACT_NONE, true);
+        }
+        if (!(mask & PH_ACTOR_APPLY)) return;
         // P32: Adam actor + target soft-update fused (critic target was
         // lerped in P15; ordering matches the row-block/persistent paths)
         hipLaunchKernelGGL(k_adam_lerp, dim3(256), dim3(256), 0, stream,
@@ -3737,6 +3761,16 @@ public:
         }
         HIP_CHECK(hipStreamSynchronize(stream));
         if (persistent) check_bar_error();
+    }
+
+    // Partial step for learner data parallelism: run only the phases in
+    // `mask` (PH_* above) and synchronize, so the host can all-reduce the
+    // gradient slabs between GRADS and APPLY.  Never takes the persistent
+    // megakernel (it fuses all phases); flagship shapes go through the
+    // row-block path, wide shapes through the per-layer MFMA path.
+    void step_part(int mask) {
+        enqueue_step(mask);
+        HIP_CHECK(hipStreamSynchronize(stream));
     }
 
     void invalidate_graph() {

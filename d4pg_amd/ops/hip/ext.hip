@@ -218,6 +218,22 @@ static void load_replay_state(int64_t h, torch::Tensor s, torch::Tensor a,
 }
 
 static void step(int64_t h, int64_t n) { get(h).step((int)n); }
+static void step_part(int64_t h, int64_t mask) {
+    get(h).step_part((int)mask);
+}
+
+// Zero-copy torch view of an engine slab (params/grads/moments) ON DEVICE,
+// so torch.distributed collectives (RCCL over xGMI) can all-reduce /
+// broadcast engine state in place — the learner-DP and local-SGD paths.
+// The tensor aliases engine memory: keep the engine alive while using it.
+static torch::Tensor device_tensor(int64_t h, int64_t which) {
+    Engine& e = get(h);
+    long n;
+    float* p = slab_ptr(e, which, n);
+    return torch::from_blob(
+        p, {n}, torch::TensorOptions().dtype(torch::kFloat32)
+                    .device(torch::kCUDA, e.device_));
+}
 static void capture(int64_t h, int64_t n) { get(h).capture((int)n); }
 static void replay(int64_t h, int64_t iters) { get(h).replay((int)iters); }
 static void replay_async(int64_t h, int64_t iters) {
@@ -315,6 +331,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
     mod.def("replay_state", &d4pg::replay_state);
     mod.def("load_replay_state", &d4pg::load_replay_state);
     mod.def("step", &d4pg::step);
+    mod.def("step_part", &d4pg::step_part);
+    mod.def("device_tensor", &d4pg::device_tensor);
     mod.def("capture", &d4pg::capture);
     mod.def("replay", &d4pg::replay);
     mod.def("replay_async", &d4pg::replay_async);
