@@ -357,6 +357,18 @@ def main():
     elapsed = _max_over_ranks(dist, elapsed)
     env_sps = measure_env_steps_per_sec(seed=1234 + rank) if rank == 0 else 0.0
     vec_sps = measure_env_steps_vector(seed=4321 + rank) if rank == 0 else 0.0
+    gpu_sps = None
+    if rank == 0 and use_gpu:
+        # device-resident actor serving (env + policy + noise + fold all
+        # on-GPU, engine.hip rollout section): M=1024 envs, 200-tick
+        # episodes, one hipGraph per episode
+        eng2 = make_gpu_engine(seed=999)
+        eng2.rollout_alloc(1024, FLAGSHIP["n_steps"], horizon=200,
+                           gamma=FLAGSHIP["gamma"], eps=0.3, seed=555)
+        eng2.rollout_run(1)                       # capture + warm
+        t0 = time.perf_counter()
+        steps_done, _ = eng2.rollout_run(10)
+        gpu_sps = steps_done / (time.perf_counter() - t0)
 
     if rank == 0:
         # In every mode each rank performs K real optimizer steps, so the
@@ -392,6 +404,7 @@ def main():
                        "device": "cuda" if use_gpu else "cpu-fallback"},
             "env_steps_per_sec_1actor": env_sps,
             "env_steps_per_sec_vector64": vec_sps,
+            "env_steps_per_sec_gpu_rollout1024": gpu_sps,
         }
         print(json.dumps(out), flush=True)
     if dist is not None:

@@ -94,6 +94,11 @@ def make_parser() -> argparse.ArgumentParser:
                         "actor rank (one [M,obs] policy forward per tick) "
                         "instead of one env per process; 0 = scalar parity "
                         "mode (distributed learner only)")
+    p.add_argument("--gpu_actors", default=-1, type=int,
+                   help="device-resident rollout for vector actor ranks "
+                        "(env dynamics + policy + noise + n-step fold as "
+                        "HIP kernels): 1 = force, 0 = off, -1 = auto "
+                        "(on when the rank sees a GPU)")
     return p
 
 
@@ -186,6 +191,7 @@ class D4PGConfig:
     noise: str = "gaussian"
     noise_eps: float = 0.3
     vector_envs: int = 0
+    gpu_actors: int = -1
     extra: dict = field(default_factory=dict)
 
     @classmethod

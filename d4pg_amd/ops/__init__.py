@@ -292,6 +292,43 @@ class FusedEngine:
     def actor_forward(self, x):
         return self.ext.actor_forward(self.h, x)
 
+    # -- GPU-resident rollout (device Pendulum env + K11 noise + K15 fold,
+    # transitions written straight into this engine's on-HBM replay;
+    # see engine.hip "GPU-resident actor rollout") --
+    def rollout_alloc(self, n_envs, n_steps, horizon=200, gamma=0.99,
+                      noise="gaussian", eps=0.3, ou_theta=0.15,
+                      ou_sigma=0.2, ou_mu=0.0, seed=0):
+        kind = {"gaussian": 0, "ou": 1}[noise]
+        self.ext.rollout_alloc(self.h, int(n_envs), int(n_steps),
+                               int(horizon), float(gamma), kind, float(eps),
+                               float(ou_theta), float(ou_sigma),
+                               float(ou_mu), int(seed))
+
+    def rollout_run(self, episodes=1, reset=True, use_graph=True):
+        """Run whole episodes on-device (one hipGraph per episode when
+        reset=True); returns (env_steps, transitions_emitted)."""
+        self.ext.rollout_run(self.h, int(episodes), bool(reset),
+                             bool(use_graph))
+        info = self.ext.rollout_info(self.h)
+        return (episodes * info["env_steps_per_episode"],
+                episodes * info["emits_per_episode"])
+
+    def rollout_set_state(self, th, thdot):
+        self.ext.rollout_set_state(self.h, torch.as_tensor(th),
+                                   torch.as_tensor(thdot))
+
+    def rollout_info(self):
+        return self.ext.rollout_info(self.h)
+
+    def replay_rows(self):
+        """Host copy of the occupied replay rows (s, a, r, s2, d) — used
+        by GPU actor ranks to ship device-generated transitions to the
+        learner, and by tests."""
+        st = self.ext.replay_state(self.h)
+        return (st["s"].numpy(), st["a"].numpy(),
+                st["r"].numpy().ravel(), st["s2"].numpy(),
+                st["d"].numpy().ravel())
+
     def info(self):
         return self.ext.info(self.h)
 

@@ -1130,6 +1130,172 @@ __global__ void k_tree_build_level(double* sum_tree, double* min_tree,
 }
 
 // ===========================================================================
+// GPU-resident actor rollout: M vectorized Pendulum envs + exploration
+// noise (K11) + n-step fold (K15) entirely on device, transitions written
+// straight into the on-HBM replay ring.  Re-expresses the reference's
+// per-env-step actor loop (/root/reference/main.py:142-152 +
+// random_process.py:4-45) and the host twins VectorPendulum / VecNStep
+// (d4pg_amd/envs/vector.py), which serve as the parity oracles.
+//
+// Per tick: [fwd chain obs->act over M rows] then ONE k_roll_tick kernel
+// (noise + dynamics + ring write + n-step emit).  Tree leaves are written
+// at emit; internal nodes are rebuilt ONCE per episode by the full
+// per-level sweep (k_tree_build_level — ~2M f64 nodes, bandwidth-cheap),
+// which beats per-tick path repair by orders of magnitude.  The whole
+// episode is one hipGraph; per-episode variability (philox epoch, replay
+// base position) lives in device memory so graph replays stay fresh.
+// ===========================================================================
+
+__device__ inline float d4pg_angle_norm(float x) {
+    const float PI = 3.14159265358979323846f;
+    float y = fmodf(x + PI, 2.0f * PI);
+    if (y < 0.0f) y += 2.0f * PI;
+    return y - PI;
+}
+
+// standard normal via Box-Muller on two philox lanes
+__device__ inline float n01(uint32_t a, uint32_t b) {
+    float u1 = u01(a), u2 = u01(b);
+    u1 = fmaxf(u1, 1e-12f);
+    return sqrtf(-2.0f * logf(u1)) *
+           cosf(6.283185307179586f * u2);
+}
+
+struct RollArgs {
+    int M, O, A, n, horizon;
+    float gamma, eps, ou_theta, ou_sigma, ou_mu, ou_dt;
+    int noise_kind;                    // 0 = gaussian, 1 = OU
+    uint64_t seed;
+    float per_alpha;
+    // env + noise state
+    float *th, *thdot, *obs, *act, *ou_x;
+    // n-step rings [n][M][*]
+    float *s_ring, *a_ring, *r_ring;
+    // per-episode device state: [0] philox epoch, [1] replay base pos
+    long long* ep_state;
+    // replay
+    float *rs, *ra, *rr, *rs2, *rd;
+    double *sum_tree, *min_tree;
+    long tree_cap, capacity;
+    Counters* cnt;
+};
+
+// episode prologue: bump the philox epoch, latch the replay write base
+__global__ void k_roll_begin(RollArgs a) {
+    if (threadIdx.x == 0 && blockIdx.x == 0) {
+        a.ep_state[0] += 1;
+        a.ep_state[1] = a.cnt->pos;
+    }
+}
+
+// reset all M envs (uniform th in [-pi,pi], thdot in [-1,1] — Pendulum-v1
+// reset distribution) + OU state to mu, and emit the first observation
+__global__ void k_roll_reset(RollArgs a) {
+    long long ep = a.ep_state[0];
+    for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < a.M;
+         i += gridDim.x * blockDim.x) {
+        Philox4 r = philox4(a.seed ^ 0xD011Eull, ((uint64_t)ep << 20) | 1u,
+                            (uint64_t)i);
+        float th = (2.0f * u01(r.v[0]) - 1.0f) * 3.14159265358979f;
+        float td = 2.0f * u01(r.v[1]) - 1.0f;
+        a.th[i] = th;
+        a.thdot[i] = td;
+        a.obs[i * 3 + 0] = cosf(th);
+        a.obs[i * 3 + 1] = sinf(th);
+        a.obs[i * 3 + 2] = td;
+        for (int k = 0; k < a.A; ++k) a.ou_x[i * a.A + k] = a.ou_mu;
+    }
+}
+
+// one synchronized tick for all M envs: noise -> clip -> dynamics ->
+// n-step ring write -> (optional) matured-transition emit into the replay.
+// slot/emit_k/done are per-tick constants baked into the episode graph.
+__global__ void k_roll_tick(RollArgs a, int tick, int slot, int emit_k,
+                            int done) {
+    long long ep = a.ep_state[0];
+    long long base = a.ep_state[1];
+    double pa = pow((double)a.cnt->max_priority, (double)a.per_alpha);
+    for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < a.M;
+         i += gridDim.x * blockDim.x) {
+        // --- exploration noise (K11): per-env per-tick philox stream ---
+        float u_n = a.act[i * a.A];        // policy output, tanh in (-1,1)
+        if (a.noise_kind == 0 && a.eps != 0.0f) {
+            Philox4 r = philox4(a.seed ^ 0x2F01Eull,
+                                ((uint64_t)ep << 20) | (uint64_t)(tick + 2),
+                                (uint64_t)i);
+            u_n += a.eps * n01(r.v[0], r.v[1]);
+        } else if (a.noise_kind == 1) {
+            Philox4 r = philox4(a.seed ^ 0x2F01Eull,
+                                ((uint64_t)ep << 20) | (uint64_t)(tick + 2),
+                                (uint64_t)i);
+            float x = a.ou_x[i * a.A];
+            x += a.ou_theta * (a.ou_mu - x) * a.ou_dt
+                 + a.ou_sigma * sqrtf(a.ou_dt) * n01(r.v[0], r.v[1]);
+            a.ou_x[i * a.A] = x;
+            u_n += a.eps * x;
+        }
+        u_n = fminf(fmaxf(u_n, -1.0f), 1.0f);
+
+        // --- n-step ring: store s_t, a_t BEFORE stepping ---
+        a.s_ring[(slot * a.M + i) * 3 + 0] = a.obs[i * 3 + 0];
+        a.s_ring[(slot * a.M + i) * 3 + 1] = a.obs[i * 3 + 1];
+        a.s_ring[(slot * a.M + i) * 3 + 2] = a.obs[i * 3 + 2];
+        a.a_ring[slot * a.M + i] = u_n;
+
+        // --- Pendulum dynamics (VectorPendulum.step parity) ---
+        const float max_torque = 2.0f, max_speed = 8.0f, dt = 0.05f,
+                    g = 10.0f;
+        float th = a.th[i], td = a.thdot[i];
+        float u = u_n * max_torque;
+        float an = d4pg_angle_norm(th);
+        float cost = an * an + 0.1f * td * td + 0.001f * u * u;
+        float newtd = td + (1.5f * g * sinf(th) + 3.0f * u) * dt;
+        newtd = fminf(fmaxf(newtd, -max_speed), max_speed);
+        th = th + newtd * dt;
+        a.th[i] = th;
+        a.thdot[i] = newtd;
+        a.r_ring[slot * a.M + i] = -cost;
+        float o0 = cosf(th), o1 = sinf(th), o2 = newtd;
+        a.obs[i * 3 + 0] = o0;
+        a.obs[i * 3 + 1] = o1;
+        a.obs[i * 3 + 2] = o2;
+
+        // --- matured n-step emit (VecNStep parity): transition
+        // (s_{t-n+1}, a_{t-n+1}, sum gamma^k r, s_{t+1}, done) ---
+        if (emit_k >= 0) {
+            int start = (slot + 1) % a.n;       // oldest ring entry
+            float R = 0.0f, gk = 1.0f;
+            for (int k = 0; k < a.n; ++k) {
+                R += gk * a.r_ring[((start + k) % a.n) * a.M + i];
+                gk *= a.gamma;
+            }
+            long dst = (base + (long)emit_k * a.M + i) % a.capacity;
+            a.rs[dst * 3 + 0] = a.s_ring[(start * a.M + i) * 3 + 0];
+            a.rs[dst * 3 + 1] = a.s_ring[(start * a.M + i) * 3 + 1];
+            a.rs[dst * 3 + 2] = a.s_ring[(start * a.M + i) * 3 + 2];
+            a.ra[dst] = a.a_ring[start * a.M + i];
+            a.rr[dst] = R;
+            a.rs2[dst * 3 + 0] = o0;
+            a.rs2[dst * 3 + 1] = o1;
+            a.rs2[dst * 3 + 2] = o2;
+            a.rd[dst] = (float)done;
+            a.sum_tree[a.tree_cap + dst] = pa;
+            a.min_tree[a.tree_cap + dst] = pa;
+        }
+    }
+}
+
+// episode epilogue: advance the replay counters by the emitted block
+__global__ void k_roll_end(RollArgs a, long emitted) {
+    if (threadIdx.x == 0 && blockIdx.x == 0) {
+        long long base = a.ep_state[1];
+        a.cnt->pos = (base + emitted) % a.capacity;
+        long long sz = a.cnt->size + emitted;
+        a.cnt->size = sz > a.capacity ? a.capacity : sz;
+    }
+}
+
+// ===========================================================================
 // MFMA tiled GEMM kernels (wide-batch learner path, f32-in matrix cores)
 // ===========================================================================
 // For the large-batch config (BASELINE config 5: B=4096, H=1024) the GEMMs
@@ -3153,6 +3319,7 @@ public:
     }
 
     ~Engine() {
+        rollout_free();
         if (graph_exec) hipGraphExecDestroy(graph_exec);
         if (graph) hipGraphDestroy(graph);
         hipStreamDestroy(stream);
@@ -3861,6 +4028,165 @@ public:
         h.adam_t_critic -= 1;
         h.rng_epoch -= 1;
         return h;
+    }
+
+    // ---------------- GPU-resident rollout (device env + K11 + K15) -----
+    // See the kernel-section comment above k_roll_begin.  State lives in a
+    // separate pool so rollout capacity is independent of the learner
+    // batch; the actor weights and the replay are SHARED with the learner
+    // (same p_actor slab, same rs/../sum_tree), which is the point: the
+    // whole actor->replay->learner loop stays in HBM.
+    RollArgs roll_{};
+    void* roll_pool_ = nullptr;
+    float *r_h1 = nullptr, *r_h2 = nullptr, *r_h3 = nullptr;
+    hipGraph_t roll_graph = nullptr;
+    hipGraphExec_t roll_graph_exec = nullptr;
+    int roll_emits_per_ep = 0;
+
+    void rollout_free() {
+        if (roll_graph_exec) { hipGraphExecDestroy(roll_graph_exec);
+                               roll_graph_exec = nullptr; }
+        if (roll_graph) { hipGraphDestroy(roll_graph); roll_graph = nullptr; }
+        if (roll_pool_) { hipFree(roll_pool_); roll_pool_ = nullptr; }
+    }
+
+    void rollout_alloc(int M, int nsteps, int horizon, float gamma,
+                       int noise_kind, float eps, float ou_theta,
+                       float ou_sigma, float ou_mu, uint64_t seed) {
+        if (cfg.obs != 3 || cfg.act != 1)
+            throw std::runtime_error(
+                "device rollout models Pendulum dynamics (obs=3, act=1)");
+        if (horizon < nsteps)
+            throw std::runtime_error("rollout horizon < n_steps");
+        rollout_free();
+        RollArgs a{};
+        a.M = M; a.O = 3; a.A = 1; a.n = nsteps; a.horizon = horizon;
+        a.gamma = gamma; a.eps = eps; a.ou_theta = ou_theta;
+        a.ou_sigma = ou_sigma; a.ou_mu = ou_mu; a.ou_dt = 1e-2f;
+        a.noise_kind = noise_kind; a.seed = seed;
+        a.per_alpha = cfg.per_alpha;
+        const int H = cfg.hidden;
+        long off = 0;
+        auto sub = [&](long nelem) {
+            long o = off;
+            off += (nelem * 4 + 255) & ~255L;
+            return o;
+        };
+        long o_th = sub(M), o_td = sub(M), o_obs = sub((long)M * 3),
+             o_act = sub(M), o_ou = sub(M),
+             o_sr = sub((long)nsteps * M * 3), o_ar = sub((long)nsteps * M),
+             o_rr = sub((long)nsteps * M), o_ep = sub(4),
+             o_h1 = sub((long)M * H), o_h2 = sub((long)M * H),
+             o_h3 = sub((long)M * H);
+        HIP_CHECK(hipMalloc(&roll_pool_, off));
+        HIP_CHECK(hipMemset(roll_pool_, 0, off));
+        char* base = (char*)roll_pool_;
+        a.th = (float*)(base + o_th); a.thdot = (float*)(base + o_td);
+        a.obs = (float*)(base + o_obs); a.act = (float*)(base + o_act);
+        a.ou_x = (float*)(base + o_ou);
+        a.s_ring = (float*)(base + o_sr); a.a_ring = (float*)(base + o_ar);
+        a.r_ring = (float*)(base + o_rr);
+        a.ep_state = (long long*)(base + o_ep);
+        r_h1 = (float*)(base + o_h1); r_h2 = (float*)(base + o_h2);
+        r_h3 = (float*)(base + o_h3);
+        a.rs = rs; a.ra = ra; a.rr = rr; a.rs2 = rs2; a.rd = rd;
+        a.sum_tree = sum_tree; a.min_tree = min_tree;
+        a.tree_cap = tree_cap; a.capacity = cfg.capacity;
+        a.cnt = cnt;
+        roll_ = a;
+        roll_emits_per_ep = horizon - nsteps + 1;
+    }
+
+    void roll_fwd_chain() {
+        const int M = roll_.M;
+        auto run = [&](const float* x, const LayerDesc& l, float* y,
+                       int actk) {
+            if (M >= 512) {
+                int ntm = ceil_div(M, MT_M), ntn = ceil_div(l.out, MT_N);
+                hipLaunchKernelGGL(k_mfma_fwd, dim3(ntm * ntn), dim3(256),
+                                   0, stream, x, (const float*)nullptr,
+                                   p_actor + l.w_off, p_actor + l.b_off, y,
+                                   M, l.in1, l.in2, l.out, actk);
+            } else {
+                FwdJob j{};
+                j.x1 = x; j.x2 = nullptr;
+                j.wt = p_actor + l.w_off; j.bias = p_actor + l.b_off;
+                j.y = y; j.B = M; j.in1 = l.in1; j.in2 = l.in2;
+                j.out = l.out; j.act = actk;
+                j.wg0 = 0; j.nwg_b = ceil_div(M, TB);
+                j.nwg_o = ceil_div(l.out, TO);
+                FwdJob e{};
+                hipLaunchKernelGGL(k_fwd3, dim3(j.nwg_b * j.nwg_o),
+                                   dim3(256), 0, stream, j, e, e, 1);
+            }
+        };
+        run(roll_.obs, anet.l[0], r_h1, ACT_RELU);
+        run(r_h1, anet.l[1], r_h2, ACT_NONE);   // fc2->fc2_2 no-act quirk
+        run(r_h2, anet.l[2], r_h3, ACT_RELU);
+        run(r_h3, anet.l[3], roll_.act, ACT_TANH);
+    }
+
+    void rollout_enqueue_episode(bool reset) {
+        RollArgs& a = roll_;
+        int wgs = ceil_div(a.M, 256);
+        hipLaunchKernelGGL(k_roll_begin, dim3(1), dim3(64), 0, stream, a);
+        if (reset)
+            hipLaunchKernelGGL(k_roll_reset, dim3(wgs), dim3(256), 0,
+                               stream, a);
+        int emit_k = 0;
+        for (int t = 0; t < a.horizon; ++t) {
+            roll_fwd_chain();
+            int ek = (t >= a.n - 1) ? emit_k : -1;
+            int done = (t == a.horizon - 1) ? 1 : 0;
+            hipLaunchKernelGGL(k_roll_tick, dim3(wgs), dim3(256), 0, stream,
+                               a, t, t % a.n, ek, done);
+            if (ek >= 0) ++emit_k;
+        }
+        hipLaunchKernelGGL(k_roll_end, dim3(1), dim3(64), 0, stream, a,
+                           (long)emit_k * a.M);
+        // internal tree nodes: one bandwidth-bound full sweep per episode
+        // (vs horizon x per-path repair) — leaves were set at emit
+        for (long lo = tree_cap / 2; lo >= 1; lo >>= 1)
+            hipLaunchKernelGGL(k_tree_build_level, dim3(1024), dim3(256), 0,
+                               stream, sum_tree, min_tree, lo, 2 * lo);
+    }
+
+    void rollout_run(int episodes, bool reset, bool use_graph) {
+        if (!roll_pool_)
+            throw std::runtime_error("rollout_alloc first");
+        if (use_graph && reset) {
+            if (!roll_graph_exec) {
+                HIP_CHECK(hipStreamBeginCapture(
+                    stream, hipStreamCaptureModeThreadLocal));
+                rollout_enqueue_episode(true);
+                HIP_CHECK(hipStreamEndCapture(stream, &roll_graph));
+                HIP_CHECK(hipGraphInstantiate(&roll_graph_exec, roll_graph,
+                                              nullptr, nullptr, 0));
+            }
+            for (int e = 0; e < episodes; ++e)
+                HIP_CHECK(hipGraphLaunch(roll_graph_exec, stream));
+        } else {
+            for (int e = 0; e < episodes; ++e)
+                rollout_enqueue_episode(reset);
+        }
+        HIP_CHECK(hipStreamSynchronize(stream));
+    }
+
+    // test hook: inject exact env state (parity vs the numpy oracle)
+    void rollout_set_state(const float* th, const float* td, int M) {
+        if (!roll_pool_ || M != roll_.M)
+            throw std::runtime_error("rollout_set_state: bad M");
+        HIP_CHECK(hipMemcpy(roll_.th, th, M * 4, hipMemcpyHostToDevice));
+        HIP_CHECK(hipMemcpy(roll_.thdot, td, M * 4, hipMemcpyHostToDevice));
+        std::vector<float> obs(3 * M);
+        for (int i = 0; i < M; ++i) {
+            obs[i * 3 + 0] = cosf(th[i]);
+            obs[i * 3 + 1] = sinf(th[i]);
+            obs[i * 3 + 2] = td[i];
+        }
+        HIP_CHECK(hipMemcpy(roll_.obs, obs.data(), 3L * M * 4,
+                            hipMemcpyHostToDevice));
+        HIP_CHECK(hipMemset(roll_.ou_x, 0, M * 4));
     }
 
     // actor forward for eval/smoke: x[B,obs] (device via staging) -> a[B,act]

@@ -255,6 +255,41 @@ static py::dict counters(int64_t h) {
     return d;
 }
 
+// ---- GPU-resident rollout (device Pendulum + K11 noise + K15 fold) ----
+static void rollout_alloc(int64_t h, int64_t M, int64_t n_steps,
+                          int64_t horizon, double gamma, int64_t noise_kind,
+                          double eps, double ou_theta, double ou_sigma,
+                          double ou_mu, int64_t seed) {
+    get(h).rollout_alloc((int)M, (int)n_steps, (int)horizon, (float)gamma,
+                         (int)noise_kind, (float)eps, (float)ou_theta,
+                         (float)ou_sigma, (float)ou_mu, (uint64_t)seed);
+}
+
+static void rollout_run(int64_t h, int64_t episodes, bool reset,
+                        bool use_graph) {
+    get(h).rollout_run((int)episodes, reset, use_graph);
+}
+
+static void rollout_set_state(int64_t h, torch::Tensor th,
+                              torch::Tensor thdot) {
+    Engine& e = get(h);
+    auto t1 = f32_contig(th), t2 = f32_contig(thdot);
+    TORCH_CHECK(t1.numel() == t2.numel(), "th/thdot size mismatch");
+    e.rollout_set_state(t1.data_ptr<float>(), t2.data_ptr<float>(),
+                        (int)t1.numel());
+}
+
+static py::dict rollout_info(int64_t h) {
+    Engine& e = get(h);
+    py::dict d;
+    d["M"] = e.roll_.M;
+    d["n"] = e.roll_.n;
+    d["horizon"] = e.roll_.horizon;
+    d["emits_per_episode"] = (long)e.roll_emits_per_ep * e.roll_.M;
+    d["env_steps_per_episode"] = (long)e.roll_.horizon * e.roll_.M;
+    return d;
+}
+
 static torch::Tensor actor_forward(int64_t h, torch::Tensor x) {
     Engine& e = get(h);
     auto t = f32_contig(x);
@@ -338,6 +373,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
     mod.def("replay_async", &d4pg::replay_async);
     mod.def("sync", &d4pg::sync);
     mod.def("counters", &d4pg::counters);
+    mod.def("rollout_alloc", &d4pg::rollout_alloc);
+    mod.def("rollout_run", &d4pg::rollout_run);
+    mod.def("rollout_set_state", &d4pg::rollout_set_state);
+    mod.def("rollout_info", &d4pg::rollout_info);
     mod.def("actor_forward", &d4pg::actor_forward);
     mod.def("read_buffer", &d4pg::read_buffer);
 }

@@ -191,6 +191,31 @@ class DistributedD4PG:
                                          horizon=args.max_steps)
             self.vfold = VecNStep(m, self.obs_dim, self.act_dim,
                                   args.n_steps, args.gamma)
+        # GPU-resident actor serving (SURVEY K11 + §2c): when this actor
+        # rank has a GPU, the whole collection loop — vectorized Pendulum
+        # dynamics, fused actor forward, exploration noise, n-step fold —
+        # runs as device kernels (engine.hip rollout section), one hipGraph
+        # per episode; the host only ships the matured transitions to the
+        # learner.  Re-expresses /root/reference/main.py:142-152.
+        self.roll_engine = None
+        ga = int(getattr(args, "gpu_actors", -1))
+        if (self.vector is not None and ga != 0
+                and torch.cuda.is_available()):
+            from ..ops import FusedEngine
+            m = int(args.vector_envs)
+            cap = m * max(1, args.max_steps - args.n_steps + 1)
+            self.roll_engine = FusedEngine(
+                obs_dim=self.obs_dim, act_dim=self.act_dim, hidden=256,
+                n_atoms=args.n_atoms, batch=64, capacity=cap,
+                v_min=args.v_min, v_max=args.v_max,
+                gamma_n=args.gamma ** args.n_steps, tau=args.tau,
+                lr_actor=1e-4, lr_critic=1e-4, seed=seed + 31)
+            self.roll_engine.rollout_alloc(
+                m, args.n_steps, horizon=args.max_steps, gamma=args.gamma,
+                noise=getattr(args, "noise", "gaussian"),
+                eps=getattr(args, "noise_eps", 0.3),
+                ou_theta=args.ou_theta, ou_sigma=args.ou_sigma,
+                ou_mu=args.ou_mu, seed=seed + 37)
         # vector-actor exploration noise honors --noise/--noise_eps/--ou_*
         # (this path used to hardcode eps=0.3).  OU
         # runs batched over [M, act]; Gaussian stays a one-liner below.
@@ -251,6 +276,17 @@ class DistributedD4PG:
                 0.95 * self.ewma + 0.05 * R
             print(f"[eval] step {self.global_step} return {R:.2f} "
                   f"ewma {self.ewma:.2f}", flush=True)
+            return lb
+        if self.roll_engine is not None:
+            # device rollout: load the freshly broadcast policy into the
+            # rollout engine's actor slab, run one on-device episode, ship
+            # the already-n-step-folded transitions
+            from ..ops import pack_net as _pack
+            self.roll_engine.load_slab("actor", _pack(self.agent.actor))
+            env_steps, _ = self.roll_engine.rollout_run(1)
+            s, a, r, s2, d = self.roll_engine.replay_rows()
+            lb.add_batch(s, a, r, s2, d)
+            self.env_meter.add(env_steps)
             return lb
         if self.vector is not None:
             import torch as _t

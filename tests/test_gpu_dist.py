@@ -30,12 +30,20 @@ O, A, H, K, B = 3, 1, 256, 51, 64
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
-def make_engine(seed=0, capacity=4096):
+def make_engine(seed=0, capacity=4096, init_params=True):
+    from d4pg_amd.models import actor, critic
     from d4pg_amd.ops import FusedEngine
-    return FusedEngine(obs_dim=O, act_dim=A, hidden=H, n_atoms=K, batch=B,
-                       capacity=capacity, v_min=-300.0, v_max=0.0,
-                       gamma_n=0.99 ** 5, tau=0.001, lr_actor=1e-4,
-                       lr_critic=1e-4, seed=seed)
+    eng = FusedEngine(obs_dim=O, act_dim=A, hidden=H, n_atoms=K, batch=B,
+                      capacity=capacity, v_min=-300.0, v_max=0.0,
+                      gamma_n=0.99 ** 5, tau=0.001, lr_actor=1e-4,
+                      lr_critic=1e-4, seed=seed)
+    if init_params:
+        torch.manual_seed(seed)
+        a = actor(O, A, hidden=H)
+        c = critic(O, A, {"type": "categorical", "v_min": -300.0,
+                          "v_max": 0.0, "n_atoms": K}, hidden=H)
+        eng.load_from_modules(a, a, c, c)
+    return eng
 
 
 def fill(eng, seed=3, n=1024):
@@ -138,7 +146,8 @@ def test_manual_two_engine_grad_dp_consistency():
                                    err_msg=f"slab {s} diverged across ranks")
     # and training actually moved the params
     assert not np.allclose(ea.store_slab("actor").numpy(),
-                           make_engine(seed=11).store_slab("actor").numpy())
+                           make_engine(seed=11).store_slab("actor").numpy()), \
+        "3 DP steps left the actor slab at its initial values"
 
 
 def _localsgd_rank(rank, world, port, q):
