@@ -2349,27 +2349,32 @@ __device__ inline void p_adam_lerp(float* __restrict__ p,
 
 // PER sample + gather phase (k_per_sample semantics; one wave per probe)
 __device__ inline void p_sample(const PStepArgs& g, int wg0 = 0,
-                                int sched_off = 0) {
+                                int sched_off = 0,
+                                long long epoch_ovr = -1,
+                                long long beta_ovr = -1) {
     // sched_off = +1 when pre-sampling the NEXT step's batch before the
-    // counter tick (overlapped under the actor-Adam phase)
+    // counter tick (overlapped under the actor-Adam phase).  epoch_ovr /
+    // beta_ovr: explicit schedule values from the hoisted-counter path
+    // (k_step_persistent reads the counters ONCE per launch and threads
+    // base+s through; loss zeroing then lives in the ct.L3 phase).
     int probe = ((int)blockIdx.x - wg0) * 4 + (threadIdx.x >> 6);
     int lane = threadIdx.x & 63;
-    if (sched_off == 0 && probe == 0 && lane == 0) {
+    if (epoch_ovr < 0 && sched_off == 0 && probe == 0 && lane == 0) {
         g.cnt->loss_critic = 0.f;
         g.cnt->loss_actor = 0.f;
     }
     if (probe >= g.B) return;
     long long n = g.cnt->size;
     double total = g.sum_tree[1];
-    float frac = fminf(
-        (float)((double)(g.cnt->beta_t + sched_off) / g.per_beta_iters),
-        1.0f);
+    long long beta_t = beta_ovr >= 0 ? beta_ovr
+                                     : g.cnt->beta_t + sched_off;
+    float frac = fminf((float)((double)beta_t / g.per_beta_iters), 1.0f);
     float beta = g.per_beta0 + frac * (1.0f - g.per_beta0);
     long idx;
     if (lane == 0) {
-        Philox4 r = philox4(g.seed,
-                            (uint64_t)(g.cnt->rng_epoch + sched_off),
-                            (uint64_t)probe);
+        long long epoch = epoch_ovr >= 0 ? epoch_ovr
+                                         : g.cnt->rng_epoch + sched_off;
+        Philox4 r = philox4(g.seed, (uint64_t)epoch, (uint64_t)probe);
         double mass = (double)u01(r.v[0]) * total;
         long node = 1;
         while (node < g.tree_cap) {
@@ -3015,12 +3020,20 @@ k_step_persistent(PStepArgs g, int nsteps) {
     NetPtrs ct = net_ptrs(g.p_critic_t, g.cl);
     const int O = g.O, A = g.A, H = g.H, K = g.K, B = g.B;
 
+    // Schedule counters are read ONCE per launch and threaded through as
+    // base+s (saves the per-step tick phase + its grid barrier, ~7 us);
+    // wg 0 writes the advanced values back after the last step.
+    const long long bt0 = g.cnt->beta_t;
+    const long long ep0 = g.cnt->rng_epoch;
+    const long long ta0 = g.cnt->adam_t_actor;
+    const long long tc0 = g.cnt->adam_t_critic;
+
     for (int s = 0; s < nsteps; ++s) {
                PTIME(g, s, 0);
         // PH0: PER sample + batch gather (only on a launch's first step —
         // later steps were pre-sampled under the previous actor-Adam)
         if (s == 0 || B > 64) {      // 16 overlap wgs cover 64 probes
-            p_sample(g);
+            p_sample(g, 0, 0, ep0 + s, bt0 + s);
             p_bar(ctr, tgt);
         }
         PTIME(g, s, 1);
@@ -3078,13 +3091,23 @@ k_step_persistent(PStepArgs g, int nsteps) {
         p_bar(ctr, tgt); PTIME(g, s, 6);
         p_fwd(lds, g.ct_h2, nullptr, ct.w3, ct.b3, g.ct_h3, B, H, 0, H,
               ACT_RELU, wg, PNWG);
+        // per-step loss accumulators zeroed here, the last full-grid phase
+        // before this step's proj-CE (loss_critic) and pgrad (loss_actor)
+        // add into them — replaces the removed tick phase's zeroing
+        if (wg == PNWG - 1 && threadIdx.x == 0) {
+            g.cnt->loss_critic = 0.f;
+            g.cnt->loss_actor = 0.f;
+        }
         p_bar(ctr, tgt); PTIME(g, s, 7);
+        // PH7+8 fused: ct.L4 softmax, then the SAME wgs run the C51
+        // projection + CE grad + priorities on the rows they just wrote
+        // (same-thread p_t reuse, wave==row layout matches p_project_ce;
+        // saves the projection phase's grid barrier)
         p_fwd(lds, g.ct_h3, nullptr, ct.w4, ct.b4, g.p_t, B, H, 0, K,
               ACT_SOFTMAX, wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 8);
-        // PH8: C51 projection + CE grad + priorities (fused)
-        p_project_ce(g, lds);
-        p_bar(ctr, tgt); PTIME(g, s, 9);
+        if (wg < (B + 3) / 4)
+            p_project_ce(g, lds);
+        p_bar(ctr, tgt); PTIME(g, s, 8); PTIME(g, s, 9);
         // PH10-12: critic dX chain (pre-update weights)
         p_bwd_dx(lds, g.dlog, c.w4, 0, H, K, B, g.c_h3, ACT_RELU, g.d3,
                  wg, PNWG);
@@ -3112,7 +3135,7 @@ k_step_persistent(PStepArgs g, int nsteps) {
         // PH14: Adam + target soft-update, critic
         p_adam_lerp(g.p_critic, g.g_critic, g.m_critic, g.v_critic,
                     g.p_critic_t, g.n_critic, g.lr_critic, g.tau,
-                    g.cnt->adam_t_critic);
+                    tc0 + s);
         p_bar(ctr, tgt); PTIME(g, s, 14);
         // PH15-18: critic(s, a_out) with UPDATED critic params
         p_fwd(lds, g.bs, nullptr, c.w1, c.b1, g.pc_h1, B, O, 0, H,
@@ -3124,12 +3147,13 @@ k_step_persistent(PStepArgs g, int nsteps) {
         p_fwd(lds, g.pc_h2, nullptr, c.w3, c.b3, g.pc_h3, B, H, 0, H,
               ACT_RELU, wg, PNWG);
         p_bar(ctr, tgt); PTIME(g, s, 17);
+        // PH17+19 fused: pc.L4 softmax, then the same wgs compute the
+        // policy head gradient on their freshly written pq rows
         p_fwd(lds, g.pc_h3, nullptr, c.w4, c.b4, g.pq, B, H, 0, K,
               ACT_SOFTMAX, wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 18);
-        // PH19: policy head gradient
-        p_policy_grad(g);
-        p_bar(ctr, tgt); PTIME(g, s, 19);
+        if (wg < (B + 3) / 4)
+            p_policy_grad(g);
+        p_bar(ctr, tgt); PTIME(g, s, 18); PTIME(g, s, 19);
         // PH20-22: dX back through the critic, ending at the tanh-masked
         // action slice (adz)
         p_bwd_dx(lds, g.pd3, c.w4, 0, H, K, B, g.pc_h3, ACT_RELU, g.pd2,
@@ -3175,24 +3199,20 @@ k_step_persistent(PStepArgs g, int nsteps) {
         if (wg < 48)
             p_adam_lerp(g.p_actor, g.g_actor, g.m_actor, g.v_actor,
                         g.p_actor_t, g.n_actor, g.lr_actor, g.tau,
-                        g.cnt->adam_t_actor, 48);
+                        ta0 + s, 48);
         else if (s + 1 < nsteps && B <= 64)
-            p_sample(g, 48, 1);
-        p_bar(ctr, tgt); PTIME(g, s, 27);
-        // PH28: schedule-counter tick (tree write-back ran in PH26)
-        if (wg == 0 && threadIdx.x == 0) {
-            g.cnt->beta_t += 1;
-            g.cnt->adam_t_actor += 1;
-            g.cnt->adam_t_critic += 1;
-            g.cnt->rng_epoch += 1;
-            if (s + 1 < nsteps) {
-                g.cnt->loss_critic = 0.f;
-                g.cnt->loss_actor = 0.f;
-            }
-        }
-        p_bar(ctr, tgt); PTIME(g, s, 28);
+            p_sample(g, 48, 1, ep0 + s + 1, bt0 + s + 1);
+        p_bar(ctr, tgt); PTIME(g, s, 27); PTIME(g, s, 28);
+        // (the per-step schedule-counter tick phase is gone: counters were
+        // hoisted to base+s registers; final values land below)
     }
-    if (wg == 0 && threadIdx.x == 0) g.gbar[160] = tgt;
+    if (wg == 0 && threadIdx.x == 0) {
+        g.cnt->beta_t = bt0 + nsteps;
+        g.cnt->rng_epoch = ep0 + nsteps;
+        g.cnt->adam_t_actor = ta0 + nsteps;
+        g.cnt->adam_t_critic = tc0 + nsteps;
+        g.gbar[160] = tgt;
+    }
 }
 
 // ===========================================================================

@@ -14,7 +14,7 @@ PHASES = [
     "PH9 c.dX4", "PH10 c.dX3", "PH11 c.dX2", "PH12 c.dW", "PH13 c.adam",
     "PH14 pc.L1", "PH15 pc.L2", "PH16 pc.L3", "PH17 pc.L4sm", "PH18 pgrad",
     "PH19 p.dX4", "PH20 p.dX3", "PH21 p.dX2a", "PH22 a.dX4", "PH23 a.dX3",
-    "PH24 a.dX2", "PH25 a.dW", "PH26 a.adam", "PH27 per_upd",
+    "PH24 a.dX2", "PH25 a.dW", "PH26 a.adam+nx", "PH27 (fused)",
 ]
 
 eng = make_engine()
