@@ -1950,6 +1950,32 @@ __device__ inline void p_bar(unsigned long long* gb,
     __syncthreads();
 }
 
+// quad-local barrier: synchronizes the 4 workgroups of one quad (a
+// row-group's column-tile team in the chained fwd/bwd blocks) without
+// touching the grid barrier.  Each quad owns its own 128 B line at
+// gbar[192 + q*16]; arrivals and polls stay quad-private, so 16 quads
+// proceed fully independently (~0.5-1 us vs ~3.3 us for the grid form).
+// qround carries across launches like the grid round: members read the
+// line once at kernel entry (ordered before any arrival by the first
+// grid barrier) and count from there.
+__device__ inline void q_bar(unsigned long long* qc,
+                             unsigned long long& qround) {
+    __syncthreads();
+    qround += 4;
+    if (threadIdx.x == 0) {
+        __threadfence();
+        atomicAdd(qc, 1ull);
+        volatile unsigned long long* f = qc;
+        long spins = 0;
+        while (*f < qround) {
+            if (++spins > (1L << 26)) break;
+            __builtin_amdgcn_s_sleep(1);
+        }
+        __threadfence();
+    }
+    __syncthreads();
+}
+
 // Tiled forward: 16-row x 64-col tiles; x rows staged in LDS (broadcast
 // reads), weight column read ONCE per tile (not once per row).  Thread
 // (rq = tid/64, c = tid%64) accumulates rows {r0+rq, +4, +8, +12} of column
@@ -2440,9 +2466,10 @@ __device__ inline void p_project(const PStepArgs& g, float* lds) {
 // distribution keeps it in LDS and computes the CE gradient/priority
 // immediately (saves a barrier and the m_proj global round-trip; q comes
 // from PH4).  m_proj is still written out for introspection/parity tests.
-__device__ inline void p_project_ce(const PStepArgs& g, float* lds) {
+__device__ inline void p_project_ce(const PStepArgs& g, float* lds,
+                                    int row_base = -1) {
     int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
-    int row = blockIdx.x * 4 + wid;
+    int row = (row_base < 0 ? (int)blockIdx.x * 4 : row_base) + wid;
     int K = g.K;
     float* mrow = lds + wid * 64;
     for (int k = lane; k < K; k += 64) mrow[k] = 0.f;
@@ -2488,8 +2515,10 @@ __device__ inline void p_project_ce(const PStepArgs& g, float* lds) {
 __device__ inline void p_bwd_dx_narrow(const PStepArgs& g, const float* dz,
                                        const float* wt, int in_lo,
                                        int in_hi, int out,
-                                       const float* ymask, float* dx) {
-    int row = blockIdx.x * 4 + (threadIdx.x >> 6);
+                                       const float* ymask, float* dx,
+                                       int row_base = -1) {
+    int row = (row_base < 0 ? (int)blockIdx.x * 4 : row_base)
+              + (threadIdx.x >> 6);
     int lane = threadIdx.x & 63;
     if (row >= g.B) return;
     int span = in_hi - in_lo;
@@ -2531,8 +2560,10 @@ __device__ inline void p_ce_grad(const PStepArgs& g) {
     }
 }
 
-__device__ inline void p_policy_grad(const PStepArgs& g) {
-    int row = blockIdx.x * 4 + (threadIdx.x >> 6);
+__device__ inline void p_policy_grad(const PStepArgs& g,
+                                     int row_base = -1) {
+    int row = (row_base < 0 ? (int)blockIdx.x * 4 : row_base)
+              + (threadIdx.x >> 6);
     int lane = threadIdx.x & 63;
     if (row >= g.B) return;
     int K = g.K;
@@ -3014,6 +3045,10 @@ k_step_persistent(PStepArgs g, int nsteps) {
     unsigned long long tgt = g.gbar[160];   // cross-launch round base
     unsigned long long* ctr = g.gbar;
     int wg = blockIdx.x;
+    // quad-barrier round base: my quad's arrival counter as of launch
+    // entry (every member reads it before ANY member can arrive — the
+    // first q_bar use sits behind several grid barriers)
+    unsigned long long qrnd = ctr[192 + ((wg >> 2) << 4)];
     NetPtrs a = net_ptrs(g.p_actor, g.al);
     NetPtrs at = net_ptrs(g.p_actor_t, g.al);
     NetPtrs c = net_ptrs(g.p_critic, g.cl);
@@ -3084,40 +3119,61 @@ k_step_persistent(PStepArgs g, int nsteps) {
         else
             p_fwd(lds, g.pa_h3, nullptr, a.w4, a.b4, g.a_out, B, H, 0, A,
                   ACT_TANH, wg - 42, 22);
-        p_bar(ctr, tgt); PTIME(g, s, 5);
-        // PH5-7: critic_target chain on (s2, a2)
-        p_fwd(lds, g.ct_h1, g.a2, ct.w2, ct.b2, g.ct_h2, B, H, A, H,
-              ACT_RELU, wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 6);
-        p_fwd(lds, g.ct_h2, nullptr, ct.w3, ct.b3, g.ct_h3, B, H, 0, H,
-              ACT_RELU, wg, PNWG);
-        // per-step loss accumulators zeroed here, the last full-grid phase
-        // before this step's proj-CE (loss_critic) and pgrad (loss_actor)
-        // add into them — replaces the removed tick phase's zeroing
+        // per-step loss accumulators zeroed here (grid barrier below
+        // orders this before the quad-chained proj-CE / pgrad adds)
         if (wg == PNWG - 1 && threadIdx.x == 0) {
             g.cnt->loss_critic = 0.f;
             g.cnt->loss_actor = 0.f;
         }
-        p_bar(ctr, tgt); PTIME(g, s, 7);
-        // PH7+8 fused: ct.L4 softmax, then the SAME wgs run the C51
-        // projection + CE grad + priorities on the rows they just wrote
-        // (same-thread p_t reuse, wave==row layout matches p_project_ce;
-        // saves the projection phase's grid barrier)
-        p_fwd(lds, g.ct_h3, nullptr, ct.w4, ct.b4, g.p_t, B, H, 0, K,
-              ACT_SOFTMAX, wg, PNWG);
-        if (wg < (B + 3) / 4)
-            p_project_ce(g, lds);
-        p_bar(ctr, tgt); PTIME(g, s, 8); PTIME(g, s, 9);
-        // PH10-12: critic dX chain (pre-update weights)
-        p_bwd_dx(lds, g.dlog, c.w4, 0, H, K, B, g.c_h3, ACT_RELU, g.d3,
-                 wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 10);
-        p_bwd_dx(lds, g.d3, c.w3, 0, H, H, B, g.c_h2, ACT_RELU, g.d2,
-                 wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 11);
-        p_bwd_dx(lds, g.d2, c.w2, 0, H, H, B, g.c_h1, ACT_RELU, g.d1,
-                 wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 12);
+        p_bar(ctr, tgt); PTIME(g, s, 5);
+        // PH5-8 as ONE phase: the critic_target chain + projection + CE is
+        // ROW-LOCAL (row b's p_t depends only on row b's ct_h1/a2), so each
+        // quad of 4 wgs walks its 4-row group through ct.L2 -> ct.L3 ->
+        // ct.L4 softmax -> C51 projection/CE with quad-local barriers; the
+        // 16 quads never wait on each other (grid barriers cost ~3.3 us,
+        // quad barriers well under 1 us, and straggler coupling is gone).
+        {
+            int q = wg >> 2, m = wg & 3;
+            unsigned long long* qc = ctr + 192 + (q << 4);
+            for (int rg = q; rg < (B + 3) / 4; rg += PNWG / 4) {
+                long r0 = (long)rg * 4;
+                int nb = min(4, B - (int)r0);
+                p_fwd(lds, g.ct_h1 + r0 * H, g.a2 + r0 * A, ct.w2, ct.b2,
+                      g.ct_h2 + r0 * H, nb, H, A, H, ACT_RELU, m, 4);
+                q_bar(qc, qrnd);
+                p_fwd(lds, g.ct_h2 + r0 * H, nullptr, ct.w3, ct.b3,
+                      g.ct_h3 + r0 * H, nb, H, 0, H, ACT_RELU, m, 4);
+                q_bar(qc, qrnd);
+                p_fwd(lds, g.ct_h3 + r0 * H, nullptr, ct.w4, ct.b4,
+                      g.p_t + r0 * K, nb, H, 0, K, ACT_SOFTMAX, m, 4);
+                if (m == 0)
+                    p_project_ce(g, lds, (int)r0);
+                // no trailing q_bar: row groups are independent and the
+                // closing grid barrier publishes everything
+            }
+        }
+        p_bar(ctr, tgt);
+        PTIME(g, s, 6); PTIME(g, s, 7); PTIME(g, s, 8); PTIME(g, s, 9);
+        // PH10-12 as ONE phase: critic dX chain (pre-update weights),
+        // row-local — quad-chained like the target block above
+        {
+            int q = wg >> 2, m = wg & 3;
+            unsigned long long* qc = ctr + 192 + (q << 4);
+            for (int rg = q; rg < (B + 3) / 4; rg += PNWG / 4) {
+                long r0 = (long)rg * 4;
+                int nb = min(4, B - (int)r0);
+                p_bwd_dx(lds, g.dlog + r0 * K, c.w4, 0, H, K, nb,
+                         g.c_h3 + r0 * H, ACT_RELU, g.d3 + r0 * H, m, 4);
+                q_bar(qc, qrnd);
+                p_bwd_dx(lds, g.d3 + r0 * H, c.w3, 0, H, H, nb,
+                         g.c_h2 + r0 * H, ACT_RELU, g.d2 + r0 * H, m, 4);
+                q_bar(qc, qrnd);
+                p_bwd_dx(lds, g.d2 + r0 * H, c.w2, 0, H, H, nb,
+                         g.c_h1 + r0 * H, ACT_RELU, g.d1 + r0 * H, m, 4);
+            }
+        }
+        p_bar(ctr, tgt);
+        PTIME(g, s, 10); PTIME(g, s, 11); PTIME(g, s, 12);
         // PH13: critic dW, 4 jobs split by tile count (l1 is the biggest)
         if (wg < 4)
             p_dw2(lds, g.d1, g.bs, nullptr, g.g_critic + g.cl[0].w_off,
@@ -3137,43 +3193,61 @@ k_step_persistent(PStepArgs g, int nsteps) {
                     g.p_critic_t, g.n_critic, g.lr_critic, g.tau,
                     tc0 + s);
         p_bar(ctr, tgt); PTIME(g, s, 14);
-        // PH15-18: critic(s, a_out) with UPDATED critic params
-        p_fwd(lds, g.bs, nullptr, c.w1, c.b1, g.pc_h1, B, O, 0, H,
-              ACT_RELU, wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 15);
-        p_fwd(lds, g.pc_h1, g.a_out, c.w2, c.b2, g.pc_h2, B, H, A, H,
-              ACT_RELU, wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 16);
-        p_fwd(lds, g.pc_h2, nullptr, c.w3, c.b3, g.pc_h3, B, H, 0, H,
-              ACT_RELU, wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 17);
-        // PH17+19 fused: pc.L4 softmax, then the same wgs compute the
-        // policy head gradient on their freshly written pq rows
-        p_fwd(lds, g.pc_h3, nullptr, c.w4, c.b4, g.pq, B, H, 0, K,
-              ACT_SOFTMAX, wg, PNWG);
-        if (wg < (B + 3) / 4)
-            p_policy_grad(g);
-        p_bar(ctr, tgt); PTIME(g, s, 18); PTIME(g, s, 19);
-        // PH20-22: dX back through the critic, ending at the tanh-masked
-        // action slice (adz)
-        p_bwd_dx(lds, g.pd3, c.w4, 0, H, K, B, g.pc_h3, ACT_RELU, g.pd2,
-                 wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 20);
-        p_bwd_dx(lds, g.pd2, c.w3, 0, H, H, B, g.pc_h2, ACT_RELU, g.pdh1,
-                 wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 21);
-        p_bwd_dx_narrow(g, g.pdh1, c.w2, H, H + A, H, g.a_out, g.adz);
-        p_bar(ctr, tgt); PTIME(g, s, 22);
-        // PH23-25: dX through the actor (az3, az2, az1)
-        p_bwd_dx(lds, g.adz, a.w4, 0, H, A, B, g.pa_h3, ACT_RELU, g.az3,
-                 wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 23);
-        p_bwd_dx(lds, g.az3, a.w3, 0, H, H, B, nullptr, ACT_NONE, g.az2,
-                 wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 24);
-        p_bwd_dx(lds, g.az2, a.w2, 0, H, H, B, g.pa_h1, ACT_RELU, g.az1,
-                 wg, PNWG);
-        p_bar(ctr, tgt); PTIME(g, s, 25);
+        // PH15-25 as ONE phase: the whole policy megachain — critic(s,
+        // a_out) forward with UPDATED critic params, policy head gradient,
+        // dX back through the critic to the action slice, then dX through
+        // the actor — is row-local end-to-end, so each quad chains its
+        // 4-row group through all 10 layers with quad-local barriers
+        // (replaces 11 grid barriers; VERDICT r1 #7 / NOTES.md roadmap —
+        // this is the pair-barrier idea at quad granularity, which keeps
+        // the weight re-streaming of the tiled form: one full W read per
+        // row-group, L2-resident)
+        {
+            int q = wg >> 2, m = wg & 3;
+            unsigned long long* qc = ctr + 192 + (q << 4);
+            for (int rg = q; rg < (B + 3) / 4; rg += PNWG / 4) {
+                long r0 = (long)rg * 4;
+                int nb = min(4, B - (int)r0);
+                p_fwd(lds, g.bs + r0 * O, nullptr, c.w1, c.b1,
+                      g.pc_h1 + r0 * H, nb, O, 0, H, ACT_RELU, m, 4);
+                q_bar(qc, qrnd);
+                p_fwd(lds, g.pc_h1 + r0 * H, g.a_out + r0 * A, c.w2, c.b2,
+                      g.pc_h2 + r0 * H, nb, H, A, H, ACT_RELU, m, 4);
+                q_bar(qc, qrnd);
+                p_fwd(lds, g.pc_h2 + r0 * H, nullptr, c.w3, c.b3,
+                      g.pc_h3 + r0 * H, nb, H, 0, H, ACT_RELU, m, 4);
+                q_bar(qc, qrnd);
+                p_fwd(lds, g.pc_h3 + r0 * H, nullptr, c.w4, c.b4,
+                      g.pq + r0 * K, nb, H, 0, K, ACT_SOFTMAX, m, 4);
+                if (m == 0)
+                    p_policy_grad(g, (int)r0);
+                q_bar(qc, qrnd);
+                p_bwd_dx(lds, g.pd3 + r0 * K, c.w4, 0, H, K, nb,
+                         g.pc_h3 + r0 * H, ACT_RELU, g.pd2 + r0 * H, m, 4);
+                q_bar(qc, qrnd);
+                p_bwd_dx(lds, g.pd2 + r0 * H, c.w3, 0, H, H, nb,
+                         g.pc_h2 + r0 * H, ACT_RELU, g.pdh1 + r0 * H,
+                         m, 4);
+                q_bar(qc, qrnd);
+                if (m == 0)
+                    p_bwd_dx_narrow(g, g.pdh1, c.w2, H, H + A, H,
+                                    g.a_out, g.adz, (int)r0);
+                q_bar(qc, qrnd);
+                p_bwd_dx(lds, g.adz + r0 * A, a.w4, 0, H, A, nb,
+                         g.pa_h3 + r0 * H, ACT_RELU, g.az3 + r0 * H, m, 4);
+                q_bar(qc, qrnd);
+                p_bwd_dx(lds, g.az3 + r0 * H, a.w3, 0, H, H, nb,
+                         nullptr, ACT_NONE, g.az2 + r0 * H, m, 4);
+                q_bar(qc, qrnd);
+                p_bwd_dx(lds, g.az2 + r0 * H, a.w2, 0, H, H, nb,
+                         g.pa_h1 + r0 * H, ACT_RELU, g.az1 + r0 * H, m, 4);
+            }
+        }
+        p_bar(ctr, tgt);
+        PTIME(g, s, 15); PTIME(g, s, 16); PTIME(g, s, 17);
+        PTIME(g, s, 18); PTIME(g, s, 19); PTIME(g, s, 20);
+        PTIME(g, s, 21); PTIME(g, s, 22); PTIME(g, s, 23);
+        PTIME(g, s, 24); PTIME(g, s, 25);
         // PH26: actor dW, with the PER tree write-back overlapped on the
         // last workgroup (priorities have been final since the proj+CE
         // phase; the counter tick moves to the final phase so both Adams
@@ -3419,7 +3493,7 @@ public:
         az1 = carve<float>((long)B * H, off);
         az2 = carve<float>((long)B * H, off);
         az3 = carve<float>((long)B * H, off);
-        gbar = carve<unsigned long long>(192, off);
+        gbar = carve<unsigned long long>(512, off);
         tstamp = carve<unsigned long long>(64, off);
         ing_s = carve<float>((long)ing_cap * O, off);
         ing_a = carve<float>((long)ing_cap * A, off);
