@@ -198,7 +198,26 @@ __global__ void k_per_sample(
         Philox4 r = philox4(seed, (uint64_t)cnt->rng_epoch, (uint64_t)probe);
         double mass = (double)u01(r.v[0]) * total;
         long node = 1;
-        while (node < tree_cap) {
+        // 4-ary descent over the binary layout: a stored parent is
+        // bit-exactly the double-sum of its children, so comparing
+        // against grandchildren sums reproduces two binary steps with
+        // ONE contiguous 32 B load — half the dependent-load chain
+        // (20 levels -> 10 loads at 1e6 leaves).
+        while (node < tree_cap / 2) {
+            long gb = 4 * node;
+            double g0 = sum_tree[gb], g1 = sum_tree[gb + 1],
+                   g2 = sum_tree[gb + 2];
+            double s01 = g0 + g1;
+            if (mass <= s01) {
+                if (mass <= g0) node = gb;
+                else { mass -= g0; node = gb + 1; }
+            } else {
+                mass -= s01;
+                if (mass <= g2) node = gb + 2;
+                else { mass -= g2; node = gb + 3; }
+            }
+        }
+        if (node < tree_cap) {                 // odd final level
             double ls = sum_tree[2 * node];
             if (mass > ls) { mass -= ls; node = 2 * node + 1; }
             else           { node = 2 * node; }
@@ -2403,9 +2422,33 @@ __device__ inline void p_sample(const PStepArgs& g, int wg0 = 0,
         Philox4 r = philox4(g.seed, (uint64_t)epoch, (uint64_t)probe);
         double mass = (double)u01(r.v[0]) * total;
         long node = 1;
-        while (node < g.tree_cap) {
-            // deep tree levels are random-access, once-per-step: bypass L2
-            // for them, keep the hot top of the tree cached
+        // 4-ary descent over the binary layout (see k_per_sample): two
+        // binary levels per contiguous grandchildren load, bit-exact leaf
+        // choice.  Deep levels are random-access, once-per-step: bypass
+        // L2 for them, keep the hot top of the tree cached.
+        while (node < g.tree_cap / 2) {
+            long gb = 4 * node;
+            double g0, g1, g2;
+            if (gb >= 16384) {
+                g0 = __builtin_nontemporal_load(&g.sum_tree[gb]);
+                g1 = __builtin_nontemporal_load(&g.sum_tree[gb + 1]);
+                g2 = __builtin_nontemporal_load(&g.sum_tree[gb + 2]);
+            } else {
+                g0 = g.sum_tree[gb];
+                g1 = g.sum_tree[gb + 1];
+                g2 = g.sum_tree[gb + 2];
+            }
+            double s01 = g0 + g1;
+            if (mass <= s01) {
+                if (mass <= g0) node = gb;
+                else { mass -= g0; node = gb + 1; }
+            } else {
+                mass -= s01;
+                if (mass <= g2) node = gb + 2;
+                else { mass -= g2; node = gb + 3; }
+            }
+        }
+        if (node < g.tree_cap) {               // odd final level
             double ls = (node >= 16384)
                 ? __builtin_nontemporal_load(&g.sum_tree[2 * node])
                 : g.sum_tree[2 * node];
