@@ -25,7 +25,7 @@ def main():
     start = col("start")
     end = col("end")
     try:
-        gx = col("grid_size", "grid_x", "workgroup")
+        gx = col("grid_size_x", "grid_size", "grid_x")
     except KeyError:
         gx = None
     agg = defaultdict(lambda: [0, 0.0])
