@@ -48,6 +48,78 @@ __global__ void k_mfma_peak(float* out, int iters) {
     out[blockIdx.x * blockDim.x + threadIdx.x] = s;
 }
 
+// MFMA with operands streamed from LDS exactly like mfma_pipeline's ks
+// loop (3 ds_read_b32 per 2 MFMAs, padded rows).  PIPE=0: naive reads
+// right before use (what the compiler makes of it); PIPE=1: 2-stage
+// manual operand prefetch (read ks+2's operands before ks's MFMAs issue).
+template <int PIPE>
+__global__ void k_mfma_lds(float* out, int iters) {
+    __shared__ float As[32 * 68];
+    __shared__ float Bs[32 * 132];
+    int tid = threadIdx.x, wid = tid >> 6, lane = tid & 63;
+    for (int e = tid; e < 32 * 68; e += blockDim.x) As[e] = 1e-9f;
+    for (int e = tid; e < 32 * 132; e += blockDim.x) Bs[e] = 1e-9f;
+    __syncthreads();
+    int wm0 = (wid >> 1) * 32, wn0 = (wid & 1) * 64;
+    int r = lane & 31, kk2 = lane >> 5;
+    f32x16 a00{}, a01{};
+    for (int it = 0; it < iters; ++it) {
+        if (PIPE == 0) {
+#pragma unroll
+            for (int ks = 0; ks < 32; ks += 2) {
+                float a0 = As[(ks + kk2) * 68 + wm0 + r];
+                float b0 = Bs[(ks + kk2) * 132 + wn0 + r];
+                float b1 = Bs[(ks + kk2) * 132 + wn0 + 32 + r];
+                a00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, a00,
+                                                           0, 0, 0);
+                a01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, a01,
+                                                           0, 0, 0);
+            }
+        } else {
+            float a0 = As[kk2 * 68 + wm0 + r];
+            float b0 = Bs[kk2 * 132 + wn0 + r];
+            float b1 = Bs[kk2 * 132 + wn0 + 32 + r];
+#pragma unroll
+            for (int ks = 0; ks < 32; ks += 2) {
+                float na = 0.f, nb0 = 0.f, nb1 = 0.f;
+                if (ks + 2 < 32) {
+                    na = As[(ks + 2 + kk2) * 68 + wm0 + r];
+                    nb0 = Bs[(ks + 2 + kk2) * 132 + wn0 + r];
+                    nb1 = Bs[(ks + 2 + kk2) * 132 + wn0 + 32 + r];
+                }
+                a00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, a00,
+                                                           0, 0, 0);
+                a01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, a01,
+                                                           0, 0, 0);
+                a0 = na; b0 = nb0; b1 = nb1;
+            }
+        }
+    }
+    float s = a00[lane & 15] + a01[lane & 15];
+    out[blockIdx.x * blockDim.x + tid] = s;
+}
+
+template <int PIPE>
+static int run_lds(int wgs, int iters, const char* tag) {
+    float* out;
+    CHECK(hipMalloc(&out, (size_t)wgs * 256 * 4));
+    dim3 grid(wgs), block(256);
+    hipLaunchKernelGGL((k_mfma_lds<PIPE>), grid, block, 0, 0, out, iters);
+    CHECK(hipDeviceSynchronize());
+    hipEvent_t e0, e1;
+    hipEventCreate(&e0); hipEventCreate(&e1);
+    hipEventRecord(e0);
+    hipLaunchKernelGGL((k_mfma_lds<PIPE>), grid, block, 0, 0, out, iters);
+    hipEventRecord(e1);
+    CHECK(hipDeviceSynchronize());
+    float ms = 0; hipEventElapsedTime(&ms, e0, e1);
+    double flop = (double)wgs * 4 * 32.0 * (double)iters * 4096.0;
+    printf("%-34s %4d wgs: %7.1f TF/s (%.2f ms)\n", tag, wgs,
+           flop / ms / 1e9, ms);
+    hipFree(out);
+    return 0;
+}
+
 template <int ACCS>
 static int run_mfma(int wgs, int waves_per_wg, int iters, const char* tag) {
     float* out;
@@ -96,5 +168,7 @@ int main() {
     run_mfma<4>(256, 4, 30000, "1 wg/CU, 4 chains");
     run_mfma<4>(512, 4, 15000, "2 wgs/CU, 4 chains");
     run_mfma<8>(512, 4, 8000, "2 wgs/CU, 8 chains");
+    run_lds<0>(512, 2000, "LDS operands, naive reads");
+    run_lds<1>(512, 2000, "LDS operands, 2-stage prefetch");
     return 0;
 }
