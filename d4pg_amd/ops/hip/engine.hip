@@ -1392,6 +1392,54 @@ __device__ inline void mfma_pipeline(FA ldA, FB ldB, int nch,
     }
 }
 
+// Variant of mfma_pipeline with caller-supplied register->LDS writer and
+// VECTORIZED staging: the staging instruction stream (scalar loads +
+// per-element bounds checks + scattered ds_writes) of the original
+// closures was ~350 VALU-pipe instructions per chunk — comparable to the
+// chunk's 32 MFMAs' ~1400 MAI cycles — capping the H-GEMMs at ~44% of
+// the measured 156 TF/s f32 MFMA peak (scripts/mfma_peak,
+// profiles/gemm_lab).  Interior tiles load float4 and write b128 rows,
+// cutting staging to ~100 instructions; edge tiles keep the guarded
+// scalar path.  LDS contents are bit-identical to mfma_pipeline's.
+template <typename FA, typename FB, typename FW, typename FWS>
+__device__ inline void mfma_pipeline_w(FA ldA, FB ldB, FW wr, FWS wrs,
+                                       int nch, float* As0, float* As1,
+                                       float* Bs0, float* Bs1, f32x16& a00,
+                                       f32x16& a01) {
+    int tid = threadIdx.x, lane = tid & 63, wid = tid >> 6;
+    int wm0 = (wid >> 1) * 32, wn0 = (wid & 1) * 64;
+    int r = lane & 31, kk2 = lane >> 5;
+    ldA(0);
+    ldB(0);
+    wr(As0, Bs0);
+    __syncthreads();
+    for (int ch = 0; ch < nch; ++ch) {
+        float* As = (ch & 1) ? As1 : As0;
+        float* Bs = (ch & 1) ? Bs1 : Bs0;
+        float* Asn = (ch & 1) ? As0 : As1;
+        float* Bsn = (ch & 1) ? Bs0 : Bs1;
+        bool more = ch + 1 < nch;
+        if (more) {
+            ldA((ch + 1) * MT_K);
+            ldB((ch + 1) * MT_K);
+        }
+        // next chunk's LDS writes ride the MFMA shadows as slices 6..15
+        // (slice 6 trails the loads by ~500 MFMA-pipe cycles, past L2
+        // latency) instead of bunching at the barrier, where the two
+        // co-resident workgroups' bursts convoy and expose the staging
+#pragma unroll
+        for (int ks = 0; ks < MT_K; ks += 2) {
+            float a0 = As[(ks + kk2) * (MT_M + 4) + wm0 + r];
+            float b0 = Bs[(ks + kk2) * (MT_N + 4) + wn0 + r];
+            float b1 = Bs[(ks + kk2) * (MT_N + 4) + wn0 + 32 + r];
+            a00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, a00, 0, 0, 0);
+            a01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, a01, 0, 0, 0);
+            if (more) wrs(ks >> 1, Asn, Bsn);
+        }
+        __syncthreads();
+    }
+}
+
 // ---------------------------------------------------------------------------
 // Big-GEMM variant: 128x256 workgroup tiles + split-K.
 // Tile-traffic accounting (NOTES.md): HBM bytes scale with
@@ -1559,30 +1607,74 @@ k_mfma_fwd(const float* __restrict__ x1, const float* __restrict__ x2,
     int m0 = (blockIdx.x / ntn) * MT_M, n0 = (blockIdx.x % ntn) * MT_N;
     int tid = threadIdx.x, lane = tid & 63;
     f32x16 acc00 = {}, acc01 = {};
-    auto ldA = [&](int k0, float* t) {
+    // vectorized staging geometry (see mfma_pipeline_w):
+    //   A: thread owns row m0+tid/4, k-segment (tid%4)*8 -> 2x float4
+    //      loads, scalar transposed LDS writes
+    //   B: thread owns k-row tid/8, n-segment (tid%8)*16 -> 4x float4
+    //      loads AND 4x b128 LDS writes (row-major, no transpose)
+    const int am = tid >> 2, ak = (tid & 3) * 8;
+    const int bk = tid >> 3, bn = (tid & 7) * 16;
+    const bool vecA = (in1 % 4 == 0) && (m0 + MT_M <= B);
+    const bool vecB = (out % 4 == 0) && (n0 + MT_N <= out);
+    float ta[8];
+    float4 tb[4];
+    auto ldA = [&](int k0) {
+        if (vecA && k0 + MT_K <= in1) {
+            const float* src = x1 + (long)(m0 + am) * in1 + k0 + ak;
+            float4 v0 = *reinterpret_cast<const float4*>(src);
+            float4 v1 = *reinterpret_cast<const float4*>(src + 4);
+            ta[0] = v0.x; ta[1] = v0.y; ta[2] = v0.z; ta[3] = v0.w;
+            ta[4] = v1.x; ta[5] = v1.y; ta[6] = v1.z; ta[7] = v1.w;
+        } else {
+            int gm = m0 + am;
 #pragma unroll
-        for (int u = 0; u < 8; ++u) {
-            int e = u * 256 + tid;
-            int gm = m0 + (e >> 5), gk = k0 + (e & 31);
-            float v = 0.f;
-            if (gm < B && gk < in_total)
-                v = (gk < in1) ? x1[(long)gm * in1 + gk]
-                               : x2[(long)gm * in2 + (gk - in1)];
-            t[u] = v;
+            for (int u = 0; u < 8; ++u) {
+                int gk = k0 + ak + u;
+                float v = 0.f;
+                if (gm < B && gk < in_total)
+                    v = (gk < in1) ? x1[(long)gm * in1 + gk]
+                                   : x2[(long)gm * in2 + (gk - in1)];
+                ta[u] = v;
+            }
         }
     };
-    auto ldB = [&](int k0, float* t) {
+    auto ldB = [&](int k0) {
+        int gk = k0 + bk;
+        if (vecB && k0 + MT_K <= in_total) {
+            const float4* src = reinterpret_cast<const float4*>(
+                wt + (long)gk * out + n0 + bn);
 #pragma unroll
-        for (int u = 0; u < 16; ++u) {
-            int e = u * 256 + tid;
-            int gk = k0 + (e >> 7), gn = n0 + (e & 127);
-            t[u] = (gk < in_total && gn < out)
-                ? wt[(long)gk * out + gn] : 0.f;
+            for (int u = 0; u < 4; ++u) tb[u] = src[u];
+        } else {
+#pragma unroll
+            for (int u = 0; u < 16; ++u) {
+                int gn = n0 + bn + u;
+                float v = (gk < in_total && gn < out)
+                    ? wt[(long)gk * out + gn] : 0.f;
+                reinterpret_cast<float*>(tb)[u] = v;
+            }
         }
+    };
+    auto wr = [&](float* As, float* Bs) {
+#pragma unroll
+        for (int u = 0; u < 8; ++u)
+            As[(ak + u) * (MT_M + 4) + am] = ta[u];
+        float4* dst = reinterpret_cast<float4*>(&Bs[bk * (MT_N + 4) + bn]);
+#pragma unroll
+        for (int u = 0; u < 4; ++u) dst[u] = tb[u];
+    };
+    // sliced form for the in-loop pipeline: A scalar writes at slices
+    // 6..13, B b128 writes at 12..15 (two streams overlap harmlessly)
+    auto wrs = [&](int sl, float* As, float* Bs) {
+        if (sl >= 6 && sl < 14)
+            As[(ak + sl - 6) * (MT_M + 4) + am] = ta[sl - 6];
+        if (sl >= 12)
+            reinterpret_cast<float4*>(
+                &Bs[bk * (MT_N + 4) + bn])[sl - 12] = tb[sl - 12];
     };
     int nch = (in_total + MT_K - 1) / MT_K;
-    mfma_pipeline(ldA, ldB, nch, As2[0], As2[1], Bs2[0], Bs2[1],
-                  acc00, acc01, nullptr);
+    mfma_pipeline_w(ldA, ldB, wr, wrs, nch, As2[0], As2[1], Bs2[0], Bs2[1],
+                    acc00, acc01);
     int wid = tid >> 6;
     int wm0 = (wid >> 1) * 32, wn0 = (wid & 1) * 64;
     const f32x16* accs[2] = {&acc00, &acc01};
@@ -1714,11 +1806,36 @@ k_mfma_dx(const float* __restrict__ dz, const float* __restrict__ wt,
 // epilogue when ksplit > 1 (dwt/db must be pre-zeroed by the caller in
 // that case); bias db[o] = sum_b dz[b][o] accumulates for free out of the
 // B-fragment during the MFMA loop (wm0==0 waves of the m0==0 tiles).
+// Split-K partial reduce: dwt[i,o] = sum_seg parts[seg][i,o] (+ db tail).
+// Replaces the fp32-atomicAdd epilogue of the split-K dW: 2.1M atomics per
+// H-layer dW cost ~80 us of L2 read-modify-write serialization, while
+// disjoint partials + this bandwidth-bound sweep cost ~3 us
+// (profiles/gemm_lab evidence).
+__global__ void k_dw_reduce(const float* __restrict__ parts,
+                            float* __restrict__ dwt, float* __restrict__ db,
+                            long nw, int out, int ksplit) {
+    long total = nw + (db ? out : 0);
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+         i += (long)gridDim.x * blockDim.x) {
+        float s = 0.f;
+        if (i < nw) {
+            for (int g = 0; g < ksplit; ++g)
+                s += parts[(long)g * nw + i];
+            dwt[i] = s;
+        } else {
+            long j = i - nw;
+            for (int g = 0; g < ksplit; ++g)
+                s += parts[(long)ksplit * nw + (long)g * out + j];
+            db[j] = s;
+        }
+    }
+}
+
 __global__ void __launch_bounds__(256, 2)
 k_mfma_dw(const float* __restrict__ dz, const float* __restrict__ x1,
           const float* __restrict__ x2, float* __restrict__ dwt,
           float* __restrict__ db, int B, int in1, int in2, int out,
-          int ksplit) {
+          int ksplit, float* __restrict__ parts) {
     MFMA_LDS_DECL;
     int in_total = in1 + in2;
     int ntn = (out + MT_N - 1) / MT_N;
@@ -1822,7 +1939,9 @@ k_mfma_dw(const float* __restrict__ dz, const float* __restrict__ x1,
             int go = n0 + wn0 + j * 32 + col;
             if (gi < in_total && go < out) {
                 if (ksplit > 1)
-                    atomicAdd(&dwt[(long)gi * out + go], (*accs[t])[reg]);
+                    // disjoint per-segment partials (k_dw_reduce sums)
+                    parts[((long)seg * in_total + gi) * out + go] =
+                        (*accs[t])[reg];
                 else
                     dwt[(long)gi * out + go] = (*accs[t])[reg];
             }
@@ -1836,8 +1955,9 @@ k_mfma_dw(const float* __restrict__ dz, const float* __restrict__ x1,
         if ((lane >> 5) == 0) {
             int go0 = n0 + wn0 + r, go1 = n0 + wn0 + 32 + r;
             if (ksplit > 1) {
-                if (go0 < out) atomicAdd(&db[go0], bias0);
-                if (go1 < out) atomicAdd(&db[go1], bias1);
+                long off = (long)ksplit * in_total * out + (long)seg * out;
+                if (go0 < out) parts[off + go0] = bias0;
+                if (go1 < out) parts[off + go1] = bias1;
             } else {
                 if (go0 < out) db[go0] = bias0;
                 if (go1 < out) db[go1] = bias1;
@@ -3407,6 +3527,7 @@ public:
     float *pd3, *pd2, *pdh1, *pda, *adz;            // policy backward deltas
     float *az1, *az2, *az3;                         // actor per-layer dz rows
     float *ing_s, *ing_a, *ing_r, *ing_s2, *ing_d;  // ingestion staging
+    float* dw_parts;                                // split-K dW partials
     int ing_cap;
     unsigned long long* gbar;                       // persistent grid barrier
     unsigned long long* tstamp;                     // phase timing probe
@@ -3536,6 +3657,22 @@ public:
         az1 = carve<float>((long)B * H, off);
         az2 = carve<float>((long)B * H, off);
         az3 = carve<float>((long)B * H, off);
+        // split-K dW disjoint partials (wide path only): max over layers
+        // of ksplit * (weights + bias); the ksplit formula mirrors
+        // launch_bwd's and guarantees every segment is non-empty
+        long dwp = 0;
+        if (B >= 512) {
+            auto need = [&](const LayerDesc& l) {
+                int it = l.in1 + l.in2;
+                int ntm = ceil_div(it, MT_M), ntn = ceil_div(l.out, MT_N);
+                int ks = 1;
+                while (ntm * ntn * ks < 256 && ks * 2 * MT_K <= B) ks *= 2;
+                long v = (long)ks * ((long)it * l.out + l.out);
+                if (v > dwp) dwp = v;
+            };
+            for (int i = 0; i < 4; ++i) { need(anet.l[i]); need(cnet.l[i]); }
+        }
+        dw_parts = carve<float>(dwp, off);
         gbar = carve<unsigned long long>(512, off);
         tstamp = carve<unsigned long long>(64, off);
         ing_s = carve<float>((long)ing_cap * O, off);
@@ -3659,19 +3796,24 @@ public:
         if (cfg.batch >= 512) {
             if (want_dw) {
                 int ntm = ceil_div(in_total, MT_M), ntn = ceil_div(l.out, MT_N);
-                // split K (= batch) until the grid covers the chip
+                // split K (= batch) until the grid covers the chip;
+                // segments write disjoint partials, k_dw_reduce sums
+                // (atomic epilogue retired — see k_dw_reduce comment)
                 int ksplit = 1;
                 while (ntm * ntn * ksplit < 256 &&
                        ksplit * 2 * MT_K <= cfg.batch)
                     ksplit *= 2;
-                if (ksplit > 1)
-                    HIP_CHECK(hipMemsetAsync(
-                        gslab + l.w_off, 0,
-                        ((long)in_total * l.out + l.out) * 4, stream));
                 hipLaunchKernelGGL(k_mfma_dw, dim3(ntm * ntn * ksplit),
                                    dim3(256), 0, stream, dz, x1, x2,
                                    gslab + l.w_off, gslab + l.b_off,
-                                   cfg.batch, l.in1, l.in2, l.out, ksplit);
+                                   cfg.batch, l.in1, l.in2, l.out, ksplit,
+                                   dw_parts);
+                if (ksplit > 1)
+                    hipLaunchKernelGGL(k_dw_reduce, dim3(1024), dim3(256),
+                                       0, stream, dw_parts,
+                                       gslab + l.w_off, gslab + l.b_off,
+                                       (long)in_total * l.out, l.out,
+                                       ksplit);
             }
             if (dx1) {
                 int ntm = ceil_div(cfg.batch, MT_M);

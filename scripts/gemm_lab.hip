@@ -84,13 +84,18 @@ int main() {
         int ksplit = 1;
         while (ntm * ntn * ksplit < 256 && ksplit * 2 * MT_K <= B)
             ksplit *= 2;
+        float* parts;
+        CHECK(hipMalloc(&parts,
+                        (size_t)ksplit * (in_total * out + out) * 4));
         hipEvent_t e0, e1; hipEventCreate(&e0); hipEventCreate(&e1);
         auto dw = [&]() {
-            if (ksplit > 1)
-                hipMemsetAsync(gw, 0, ((size_t)in_total * out + out) * 4, 0);
             hipLaunchKernelGGL(k_mfma_dw, dim3(ntm * ntn * ksplit),
                                dim3(256), 0, 0, dz, x1, (const float*)nullptr,
-                               gw, gb, B, in_total, 0, out, ksplit);
+                               gw, gb, B, in_total, 0, out, ksplit, parts);
+            if (ksplit > 1)
+                hipLaunchKernelGGL(k_dw_reduce, dim3(1024), dim3(256), 0, 0,
+                                   parts, gw, gb, (long)in_total * out, out,
+                                   ksplit);
         };
         dw(); CHECK(hipDeviceSynchronize());
         hipEventRecord(e0);
