@@ -1710,48 +1710,64 @@ k_mfma_dx(const float* __restrict__ dz, const float* __restrict__ wt,
     int n0 = (blockIdx.x % ntn) * MT_N;             // relative i tile
     int tid = threadIdx.x, lane = tid & 63;
     f32x16 acc00 = {}, acc01 = {};
-    auto ldA = [&](int k0, float* t) {
-#pragma unroll
-        for (int u = 0; u < 8; ++u) {
-            int e = u * 256 + tid;
-            int gm = m0 + (e >> 5), gk = k0 + (e & 31);
-            t[u] = (gm < B && gk < out) ? dz[(long)gm * out + gk] : 0.f;
-        }
-    };
-    // B tile = Wt^T: t holds Bs[k(=o)][n(=i rel)] = wt[i][o]; the load
-    // geometry keeps 32 consecutive o per thread-sweep (128 B segments)
-    auto ldB = [&](int k0, float* t) {
-#pragma unroll
-        for (int u = 0; u < 16; ++u) {
-            int e = u * 256 + tid;
-            int nn = e >> 5, kk = e & 31;         // [n][k] load order
-            int gi = in_lo + n0 + nn, gk = k0 + kk;
-            t[u] = (gi < in_hi && gk < out)
-                ? wt[(long)gi * out + gk] : 0.f;
-        }
-    };
-    // ldB writes must land at Bs[kk][nn] but wr() assumes [k][n] from the
-    // (e>>7, e&127) mapping; so remap via a custom pipeline here:
-    {
-        float ta[8], tb[16];
-        ldA(0, ta);
-        ldB(0, tb);
-        auto wrA = [&](float* As) {
+    // Vectorized staging (see k_mfma_fwd):
+    //   A[k(=o)][m(=b)] = dz[b][o]: thread owns dz-row m0+tid/4,
+    //     o-segment (tid%4)*8 -> 2x float4 loads, scalar transposed
+    //     LDS writes
+    //   B[k(=o)][n(=i)] = wt[i][o]: thread owns wt-row in_lo+n0+tid/2,
+    //     o-segment (tid%2)*16 -> 4x float4 loads, scalar transposed
+    //     LDS writes
+    const int am = tid >> 2, ak = (tid & 3) * 8;
+    const int bnn = tid >> 1, bkk = (tid & 1) * 16;
+    const bool vecA = (out % 4 == 0) && (m0 + MT_M <= B);
+    const bool vecB = (out % 4 == 0) && (n0 + MT_N <= span);
+    float ta[8], tb[16];
+    auto ldA = [&](int k0) {
+        if (vecA && k0 + MT_K <= out) {
+            const float* src = dz + (long)(m0 + am) * out + k0 + ak;
+            float4 v0 = *reinterpret_cast<const float4*>(src);
+            float4 v1 = *reinterpret_cast<const float4*>(src + 4);
+            ta[0] = v0.x; ta[1] = v0.y; ta[2] = v0.z; ta[3] = v0.w;
+            ta[4] = v1.x; ta[5] = v1.y; ta[6] = v1.z; ta[7] = v1.w;
+        } else {
+            int gm = m0 + am;
 #pragma unroll
             for (int u = 0; u < 8; ++u) {
-                int e = u * 256 + tid;
-                As[(e & 31) * (MT_M + 4) + (e >> 5)] = ta[u];
+                int gk = k0 + ak + u;
+                ta[u] = (gm < B && gk < out)
+                    ? dz[(long)gm * out + gk] : 0.f;
             }
-        };
-        auto wrB = [&](float* Bs) {
+        }
+    };
+    auto ldB = [&](int k0) {
+        int gi = in_lo + n0 + bnn;
+        if (vecB && k0 + MT_K <= out) {
+            const float4* src = reinterpret_cast<const float4*>(
+                wt + (long)gi * out + k0 + bkk);
+#pragma unroll
+            for (int u = 0; u < 4; ++u)
+                reinterpret_cast<float4*>(tb)[u] = src[u];
+        } else {
 #pragma unroll
             for (int u = 0; u < 16; ++u) {
-                int e = u * 256 + tid;
-                Bs[(e & 31) * (MT_N + 4) + (e >> 5)] = tb[u];
+                int gk = k0 + bkk + u;
+                tb[u] = (gi < in_hi && gk < out)
+                    ? wt[(long)gi * out + gk] : 0.f;
             }
+        }
+    };
+    {
+        ldA(0);
+        ldB(0);
+        auto wr = [&](float* As, float* Bs) {
+#pragma unroll
+            for (int u = 0; u < 8; ++u)
+                As[(ak + u) * (MT_M + 4) + am] = ta[u];
+#pragma unroll
+            for (int u = 0; u < 16; ++u)
+                Bs[(bkk + u) * (MT_N + 4) + bnn] = tb[u];
         };
-        wrA(As2[0]);
-        wrB(Bs2[0]);
+        wr(As2[0], Bs2[0]);
         __syncthreads();
         int wid = tid >> 6;
         int wm0 = (wid >> 1) * 32, wn0 = (wid & 1) * 64;
@@ -1760,10 +1776,12 @@ k_mfma_dx(const float* __restrict__ dz, const float* __restrict__ wt,
         for (int ch = 0; ch < nch; ++ch) {
             float* As = As2[ch & 1];
             float* Bs = Bs2[ch & 1];
+            float* Asn = As2[(ch + 1) & 1];
+            float* Bsn = Bs2[(ch + 1) & 1];
             bool more = ch + 1 < nch;
             if (more) {
-                ldA((ch + 1) * MT_K, ta);
-                ldB((ch + 1) * MT_K, tb);
+                ldA((ch + 1) * MT_K);
+                ldB((ch + 1) * MT_K);
             }
 #pragma unroll
             for (int ks = 0; ks < MT_K; ks += 2) {
@@ -1772,10 +1790,18 @@ k_mfma_dx(const float* __restrict__ dz, const float* __restrict__ wt,
                 float b1 = Bs[(ks + kk2) * (MT_N + 4) + wn0 + 32 + r];
                 acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
                 acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
-            }
-            if (more) {
-                wrA(As2[(ch + 1) & 1]);
-                wrB(Bs2[(ch + 1) & 1]);
+                // next chunk's LDS writes ride the MFMA shadows: one A
+                // element + two B elements per slice over slices 6..13
+                if (more) {
+                    int sl = ks >> 1;
+                    if (sl >= 6 && sl < 14) {
+                        int u = sl - 6;
+                        Asn[(ak + u) * (MT_M + 4) + am] = ta[u];
+                        Bsn[(bkk + 2 * u) * (MT_N + 4) + bnn] = tb[2 * u];
+                        Bsn[(bkk + 2 * u + 1) * (MT_N + 4) + bnn] =
+                            tb[2 * u + 1];
+                    }
+                }
             }
             __syncthreads();
         }
@@ -1852,62 +1878,79 @@ k_mfma_dw(const float* __restrict__ dz, const float* __restrict__ x1,
     long b_base = (long)ch0 * MT_K;
     int tid = threadIdx.x, lane = tid & 63;
     f32x16 acc00 = {}, acc01 = {};
-    // As[k(=b)][m(=i)] = X[b][i], i fastest on the read side
-    auto ldA = [&](int k0, float* t) {
+    // Vectorized staging (same rationale as k_mfma_fwd):
+    //   A[k(=b)][m(=i)]: thread owns batch-row b_base+k0+tid/8,
+    //     i-segment (tid%8)*8 -> 2x float4 loads AND b128 LDS writes
+    //     (no transpose: global and LDS are both i-fastest)
+    //   B[k(=b)][n(=o)]: thread owns batch-row tid/8, o-segment
+    //     (tid%8)*16 -> 4x float4 loads and b128 writes
+    const int akb = tid >> 3, ai = (tid & 7) * 8;
+    const int bo = (tid & 7) * 16;
+    const bool vecA = (in1 % 4 == 0) && (m0 + MT_M <= in1);
+    const bool vecB = (out % 4 == 0) && (n0 + MT_N <= out);
+    float4 ta[2], tb[4];
+    auto ldA = [&](int k0) {
+        long gb = b_base + k0 + akb;
+        if (vecA && b_base + k0 + MT_K <= B) {
+            const float* src = x1 + gb * in1 + m0 + ai;
+            ta[0] = *reinterpret_cast<const float4*>(src);
+            ta[1] = *reinterpret_cast<const float4*>(src + 4);
+        } else {
 #pragma unroll
-        for (int u = 0; u < 8; ++u) {
-            int e = u * 256 + tid;
-            long gb = b_base + k0 + (e >> 6);
-            int gi = m0 + (e & 63);
-            float v = 0.f;
-            if (gb < B && gi < in_total)
-                v = (gi < in1) ? x1[gb * in1 + gi]
-                               : x2[gb * in2 + (gi - in1)];
-            t[u] = v;
+            for (int u = 0; u < 8; ++u) {
+                int gi = m0 + ai + u;
+                float v = 0.f;
+                if (gb < B && gi < in_total)
+                    v = (gi < in1) ? x1[gb * in1 + gi]
+                                   : x2[gb * in2 + (gi - in1)];
+                reinterpret_cast<float*>(ta)[u] = v;
+            }
         }
     };
-    auto ldB = [&](int k0, float* t) {
+    auto ldB = [&](int k0) {
+        long gb = b_base + k0 + akb;
+        if (vecB && b_base + k0 + MT_K <= B) {
+            const float4* src =
+                reinterpret_cast<const float4*>(dz + gb * out + n0 + bo);
 #pragma unroll
-        for (int u = 0; u < 16; ++u) {
-            int e = u * 256 + tid;
-            long gb = b_base + k0 + (e >> 7);
-            int go = n0 + (e & 127);
-            t[u] = (gb < B && go < out) ? dz[gb * out + go] : 0.f;
+            for (int u = 0; u < 4; ++u) tb[u] = src[u];
+        } else {
+#pragma unroll
+            for (int u = 0; u < 16; ++u) {
+                int go = n0 + bo + u;
+                reinterpret_cast<float*>(tb)[u] =
+                    (gb < B && go < out) ? dz[gb * out + go] : 0.f;
+            }
         }
     };
-    // custom pipeline (A write geometry differs from mfma_pipeline's)
     float bias0 = 0.f, bias1 = 0.f;
     {
         int wid = tid >> 6;
         int wm0 = (wid >> 1) * 32, wn0 = (wid & 1) * 64;
         int r = lane & 31, kk2 = lane >> 5;
-        float ta[8], tb[16];
-        ldA(0, ta);
-        ldB(0, tb);
-        auto wrA = [&](float* As) {
+        ldA(0);
+        ldB(0);
+        auto wr = [&](float* As, float* Bs) {
+            float4* da =
+                reinterpret_cast<float4*>(&As[akb * (MT_M + 4) + ai]);
+            da[0] = ta[0];
+            da[1] = ta[1];
+            float4* dbp =
+                reinterpret_cast<float4*>(&Bs[akb * (MT_N + 4) + bo]);
 #pragma unroll
-            for (int u = 0; u < 8; ++u) {
-                int e = u * 256 + tid;
-                As[(e >> 6) * (MT_M + 4) + (e & 63)] = ta[u];
-            }
+            for (int u = 0; u < 4; ++u) dbp[u] = tb[u];
         };
-        auto wrB = [&](float* Bs) {
-#pragma unroll
-            for (int u = 0; u < 16; ++u) {
-                int e = u * 256 + tid;
-                Bs[(e >> 7) * (MT_N + 4) + (e & 127)] = tb[u];
-            }
-        };
-        wrA(As2[0]);
-        wrB(Bs2[0]);
+        wr(As2[0], Bs2[0]);
         __syncthreads();
         for (int ch = 0; ch < nch; ++ch) {
             float* As = As2[ch & 1];
             float* Bs = Bs2[ch & 1];
+            float* Asn = As2[(ch + 1) & 1];
+            float* Bsn = Bs2[(ch + 1) & 1];
             bool more = ch + 1 < nch;
             if (more) {
-                ldA((ch + 1) * MT_K, ta);
-                ldB((ch + 1) * MT_K, tb);
+                ldA((ch + 1) * MT_K);
+                ldB((ch + 1) * MT_K);
             }
 #pragma unroll
             for (int ks = 0; ks < MT_K; ks += 2) {
@@ -1917,10 +1960,19 @@ k_mfma_dw(const float* __restrict__ dz, const float* __restrict__ x1,
                 acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
                 acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
                 if (wm0 == 0) { bias0 += b0; bias1 += b1; }
-            }
-            if (more) {
-                wrA(As2[(ch + 1) & 1]);
-                wrB(Bs2[(ch + 1) & 1]);
+                // next chunk's LDS writes ride the MFMA shadows
+                if (more) {
+                    int sl = ks >> 1;
+                    if (sl == 6)
+                        reinterpret_cast<float4*>(
+                            &Asn[akb * (MT_M + 4) + ai])[0] = ta[0];
+                    else if (sl == 7)
+                        reinterpret_cast<float4*>(
+                            &Asn[akb * (MT_M + 4) + ai])[1] = ta[1];
+                    else if (sl >= 8 && sl < 12)
+                        reinterpret_cast<float4*>(
+                            &Bsn[akb * (MT_N + 4) + bo])[sl - 8] = tb[sl - 8];
+                }
             }
             __syncthreads();
         }
