@@ -415,11 +415,18 @@ __global__ void k_project_ce(const float* __restrict__ p_t,
                              int B, int K, float v_min, float v_max,
                              float gamma_n, float per_eps,
                              int is_weighting) {
-    int row = blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
     int lane = threadIdx.x & 63;
     int wrow = threadIdx.x / 64;
+    int wpb = blockDim.x / 64;
     extern __shared__ float lm[];
     float* mrow = lm + wrow * 64;
+    __shared__ float loss_red[8];
+    float ce_acc = 0.f;
+    // grid-stride over rows with a single per-wg loss atomic at the end:
+    // one atomicAdd per ROW to the shared loss scalar serialized ~50 us
+    // at B=4096 (4096 same-address RMWs)
+    for (int row = blockIdx.x * wpb + wrow; row < B;
+         row += gridDim.x * wpb) {
     mrow[lane] = 0.f;
     __builtin_amdgcn_wave_barrier();
     if (row < B && lane < K) {
@@ -435,7 +442,6 @@ __global__ void k_project_ce(const float* __restrict__ p_t,
         atomicAdd(&mrow[u], p * (b - (float)l));
     }
     __builtin_amdgcn_wave_barrier();
-    if (row >= B) return;
     float mv = (lane < K) ? mrow[lane] : 0.f;
     float qv = (lane < K) ? q[(long)row * K + lane] : 0.f;
     if (lane < K) m_out[(long)row * K + lane] = mv;
@@ -450,7 +456,17 @@ __global__ void k_project_ce(const float* __restrict__ p_t,
         dlogits[(long)row * K + lane] = scale * (qv - mv) / (float)B;
     if (lane == 0) {
         pri[row] = dot + per_eps;
-        atomicAdd(&cnt->loss_critic, scale * ce / (float)B);
+        ce_acc += scale * ce / (float)B;
+    }
+    __builtin_amdgcn_wave_barrier();
+    }
+    __syncthreads();
+    if (lane == 0) loss_red[wrow] = ce_acc;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        float s = 0.f;
+        for (int wv = 0; wv < wpb; ++wv) s += loss_red[wv];
+        atomicAdd(&cnt->loss_critic, s);
     }
 }
 
@@ -460,17 +476,31 @@ __global__ void k_policy_grad(const float* __restrict__ q,
                               float* __restrict__ dlogits,
                               Counters* cnt,
                               int B, int K, float v_min, float v_max) {
-    int row = blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
     int lane = threadIdx.x & 63;
-    if (row >= B) return;
+    int wrow = threadIdx.x / 64;
+    int wpb = blockDim.x / 64;
+    __shared__ float loss_red[8];
+    float loss_acc = 0.f;
     float delta = (v_max - v_min) / (K - 1);
     float z = v_min + lane * delta;
-    float qv = (lane < K) ? q[(long)row * K + lane] : 0.f;
-    float e = qv * z;
-    for (int s = 32; s > 0; s >>= 1) e += __shfl_xor(e, s, 64);
-    if (lane < K)
-        dlogits[(long)row * K + lane] = -qv * (z - e) / (float)B;
-    if (lane == 0) atomicAdd(&cnt->loss_actor, -e / (float)B);
+    // grid-stride rows + one per-wg loss atomic (see k_project_ce)
+    for (int row = blockIdx.x * wpb + wrow; row < B;
+         row += gridDim.x * wpb) {
+        float qv = (lane < K) ? q[(long)row * K + lane] : 0.f;
+        float e = qv * z;
+        for (int s = 32; s > 0; s >>= 1) e += __shfl_xor(e, s, 64);
+        if (lane < K)
+            dlogits[(long)row * K + lane] = -qv * (z - e) / (float)B;
+        if (lane == 0) loss_acc += -e / (float)B;
+    }
+    __syncthreads();
+    if (lane == 0) loss_red[wrow] = loss_acc;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        float s = 0.f;
+        for (int wv = 0; wv < wpb; ++wv) s += loss_red[wv];
+        atomicAdd(&cnt->loss_actor, s);
+    }
 }
 
 __device__ inline float act_mask(int act, float h) {
@@ -1595,16 +1625,27 @@ __global__ void k_mfma_epilogue(const float* __restrict__ part,
     __shared__ float As2[2][MT_K * (MT_M + 4)]; \
     __shared__ float Bs2[2][MT_K * (MT_N + 4)]
 
-// C[B,out] = act(X[B,in1 (++ concat in2)] @ Wt[in,out] + bias)
+// C[B,out] = act(X[B,in1 (++ concat in2)] @ Wt[in,out] + bias).
+// ksplit > 1: split-K over `parts` (disjoint per-segment partials summed
+// by k_head_finish) — used for the narrow heads (out <= 64), whose
+// M-tile-only grids (64 wgs) otherwise leave 3/4 of the chip idle.
 __global__ void __launch_bounds__(256, 2)
 k_mfma_fwd(const float* __restrict__ x1, const float* __restrict__ x2,
            const float* __restrict__ wt, const float* __restrict__ bias,
            float* __restrict__ y, int B, int in1, int in2, int out,
-           int act_kind) {
+           int act_kind, int ksplit, float* __restrict__ parts) {
     MFMA_LDS_DECL;
     int in_total = in1 + in2;
     int ntn = (out + MT_N - 1) / MT_N;
-    int m0 = (blockIdx.x / ntn) * MT_M, n0 = (blockIdx.x % ntn) * MT_N;
+    int ntm = (B + MT_M - 1) / MT_M;
+    int seg = blockIdx.x / (ntm * ntn);
+    int tile = blockIdx.x % (ntm * ntn);
+    int m0 = (tile / ntn) * MT_M, n0 = (tile % ntn) * MT_N;
+    int nch_total = (in_total + MT_K - 1) / MT_K;
+    int nch_seg = (nch_total + ksplit - 1) / ksplit;
+    int ch0 = seg * nch_seg;
+    int nch = min(nch_seg, nch_total - ch0);
+    int kbase = ch0 * MT_K;
     int tid = threadIdx.x, lane = tid & 63;
     f32x16 acc00 = {}, acc01 = {};
     // vectorized staging geometry (see mfma_pipeline_w):
@@ -1618,7 +1659,8 @@ k_mfma_fwd(const float* __restrict__ x1, const float* __restrict__ x2,
     const bool vecB = (out % 4 == 0) && (n0 + MT_N <= out);
     float ta[8];
     float4 tb[4];
-    auto ldA = [&](int k0) {
+    auto ldA = [&](int k0r) {
+        int k0 = kbase + k0r;
         if (vecA && k0 + MT_K <= in1) {
             const float* src = x1 + (long)(m0 + am) * in1 + k0 + ak;
             float4 v0 = *reinterpret_cast<const float4*>(src);
@@ -1638,7 +1680,8 @@ k_mfma_fwd(const float* __restrict__ x1, const float* __restrict__ x2,
             }
         }
     };
-    auto ldB = [&](int k0) {
+    auto ldB = [&](int k0r) {
+        int k0 = kbase + k0r;
         int gk = k0 + bk;
         if (vecB && k0 + MT_K <= in_total) {
             const float4* src = reinterpret_cast<const float4*>(
@@ -1672,7 +1715,6 @@ k_mfma_fwd(const float* __restrict__ x1, const float* __restrict__ x2,
             reinterpret_cast<float4*>(
                 &Bs[bk * (MT_N + 4) + bn])[sl - 12] = tb[sl - 12];
     };
-    int nch = (in_total + MT_K - 1) / MT_K;
     mfma_pipeline_w(ldA, ldB, wr, wrs, nch, As2[0], As2[1], Bs2[0], Bs2[1],
                     acc00, acc01);
     int wid = tid >> 6;
@@ -1688,11 +1730,49 @@ k_mfma_fwd(const float* __restrict__ x1, const float* __restrict__ x2,
             int gm = m0 + wm0 + i * 32 + row;
             int gn = n0 + wn0 + j * 32 + col;
             if (gm < B && gn < out) {
-                float v = (*accs[t])[reg] + bias[gn];
-                if (act_kind == ACT_RELU) v = fmaxf(v, 0.f);
-                else if (act_kind == ACT_TANH) v = tanhf(v);
-                y[(long)gm * out + gn] = v;
+                if (ksplit > 1) {
+                    parts[((long)seg * B + gm) * out + gn] =
+                        (*accs[t])[reg];
+                } else {
+                    float v = (*accs[t])[reg] + bias[gn];
+                    if (act_kind == ACT_RELU) v = fmaxf(v, 0.f);
+                    else if (act_kind == ACT_TANH) v = tanhf(v);
+                    y[(long)gm * out + gn] = v;
+                }
             }
+        }
+    }
+}
+
+// Head finish: y[row] = act(sum_seg parts[seg][row] + bias), wave == row
+// (out <= 64); softmax reduces across the wave like k_fwd3's epilogue.
+__global__ void k_head_finish(const float* __restrict__ parts,
+                              const float* __restrict__ bias,
+                              float* __restrict__ y, int B, int out,
+                              int ksplit, int act_kind) {
+    int lane = threadIdx.x & 63, wrow = threadIdx.x / 64;
+    int wpb = blockDim.x / 64;
+    for (int row = blockIdx.x * wpb + wrow; row < B;
+         row += gridDim.x * wpb) {
+        float v = 0.f;
+        if (lane < out) {
+            for (int g = 0; g < ksplit; ++g)
+                v += parts[((long)g * B + row) * out + lane];
+            v += bias[lane];
+        }
+        if (act_kind == ACT_SOFTMAX) {
+            float mx = (lane < out) ? v : -INFINITY;
+            for (int s = 32; s > 0; s >>= 1)
+                mx = fmaxf(mx, __shfl_xor(mx, s, 64));
+            float e = (lane < out) ? __expf(v - mx) : 0.f;
+            float sum = e;
+            for (int s = 32; s > 0; s >>= 1)
+                sum += __shfl_xor(sum, s, 64);
+            if (lane < out) y[(long)row * out + lane] = e / sum;
+        } else if (lane < out) {
+            if (act_kind == ACT_RELU) v = fmaxf(v, 0.f);
+            else if (act_kind == ACT_TANH) v = tanhf(v);
+            y[(long)row * out + lane] = v;
         }
     }
 }
@@ -3721,6 +3801,13 @@ public:
                 while (ntm * ntn * ks < 256 && ks * 2 * MT_K <= B) ks *= 2;
                 long v = (long)ks * ((long)it * l.out + l.out);
                 if (v > dwp) dwp = v;
+                if (l.out <= 64) {      // split-K head fwd partials
+                    int ntmh = ceil_div(B, MT_M), ksh = 1;
+                    while (ntmh * ksh < 512 && ksh * 2 * MT_K <= it)
+                        ksh *= 2;
+                    long v2 = (long)ksh * (long)B * l.out;
+                    if (v2 > dwp) dwp = v2;
+                }
             };
             for (int i = 0; i < 4; ++i) { need(anet.l[i]); need(cnet.l[i]); }
         }
@@ -3774,11 +3861,12 @@ public:
     }
 
     bool mfma_eligible(int in_total, int out, int act_kind) const {
-        // wide-batch GEMMs go to the matrix cores (small dims pad into
-        // the 128x128 tile — still far faster than the VALU fallback);
-        // only the wave-wide softmax head stays on k_fwd3
-        (void)out; (void)in_total;
-        return cfg.batch >= 512 && act_kind != ACT_SOFTMAX;
+        // wide-batch GEMMs go to the matrix cores; narrow heads
+        // (out <= 64, incl. the softmax ones) take the split-K
+        // k_mfma_fwd + k_head_finish pair — their M-tile-only grids
+        // were chip-starved on both k_fwd3 (~60 us) and plain MFMA
+        (void)out; (void)in_total; (void)act_kind;
+        return cfg.batch >= 512;
     }
 
     void launch_fwd(std::initializer_list<FwdJob> jobs) {
@@ -3788,10 +3876,25 @@ public:
         for (auto& j : jobs)
             all_mfma = all_mfma && mfma_eligible(j.in1 + j.in2, j.out, j.act);
         if (all_mfma) {
-            // measured: a softmax head on MFMA (grid 64 wgs, 60% padded
-            // cols) loses to the wave-per-row k_fwd3 head — keep softmax
-            // groups on the per-layer kernel
             for (auto& j : jobs) {
+                int in_total_j = j.in1 + j.in2;
+                if (j.out <= 64) {
+                    // narrow head: split-K partials + wave-per-row finish
+                    int ntm = ceil_div(j.B, MT_M);
+                    int ks = 1;
+                    while (ntm * ks < 512 && ks * 2 * MT_K <= in_total_j)
+                        ks *= 2;
+                    hipLaunchKernelGGL(k_mfma_fwd, dim3(ntm * ks),
+                                       dim3(256), 0, stream, j.x1, j.x2,
+                                       j.wt, j.bias, j.y, j.B, j.in1,
+                                       j.in2, j.out, j.act, ks, dw_parts);
+                    if (ks > 1)
+                        hipLaunchKernelGGL(k_head_finish, dim3(256),
+                                           dim3(256), 0, stream, dw_parts,
+                                           j.bias, j.y, j.B, j.out, ks,
+                                           j.act);
+                    continue;
+                }
                 // OPT-IN (D4PG_FWD2=1): measured 7% SLOWER than the
                 // 64x128 tiles (272 vs 294 steps/s same-box) — the weight
                 // operand is L2-resident so the halved "HBM traffic" was
@@ -3828,7 +3931,8 @@ public:
                 int ntm = ceil_div(j.B, MT_M), ntn = ceil_div(j.out, MT_N);
                 hipLaunchKernelGGL(k_mfma_fwd, dim3(ntm * ntn), dim3(256),
                                    0, stream, j.x1, j.x2, j.wt, j.bias, j.y,
-                                   j.B, j.in1, j.in2, j.out, j.act);
+                                   j.B, j.in1, j.in2, j.out, j.act, 1,
+                                   (float*)nullptr);
             }
             return;
         }
@@ -4151,7 +4255,8 @@ public:
         { int wg = 0; launch_fwd({fwd_job(ct_h3, nullptr, p_critic_t,
                                           cnet.l[3], p_t, ACT_SOFTMAX, wg)}); }
         // P9+P10 fused: projection + CE grad + priorities
-        hipLaunchKernelGGL(k_project_ce, dim3(row_wgs), dim3(256),
+        hipLaunchKernelGGL(k_project_ce, dim3(min(row_wgs, 256)),
+                           dim3(256),
                            waves_per_wg * 64 * sizeof(float), stream,
                            p_t, br, bd, q, bw, m_proj, dlog, pri, cnt,
                            B, K, cfg.v_min, cfg.v_max, cfg.gamma_n,
@@ -4197,7 +4302,8 @@ public:
         { int wg = 0; launch_fwd({fwd_job(pc_h3, nullptr, p_critic,
                                           cnet.l[3], pq, ACT_SOFTMAX, wg)}); }
         // P23: policy head gradient
-        hipLaunchKernelGGL(k_policy_grad, dim3(row_wgs), dim3(256), 0, stream,
+        hipLaunchKernelGGL(k_policy_grad, dim3(min(row_wgs, 256)),
+                           dim3(256), 0, stream,
                            pq, pd3, cnt, B, K, cfg.v_min, cfg.v_max);
         // P24-26: dX back through critic' (no dW), ending at da
         launch_bwd(pd3, pc_h3, nullptr, p_critic, nullptr, cnet.l[3],
@@ -4437,7 +4543,8 @@ public:
                 hipLaunchKernelGGL(k_mfma_fwd, dim3(ntm * ntn), dim3(256),
                                    0, stream, x, (const float*)nullptr,
                                    p_actor + l.w_off, p_actor + l.b_off, y,
-                                   M, l.in1, l.in2, l.out, actk);
+                                   M, l.in1, l.in2, l.out, actk, 1,
+                                   (float*)nullptr);
             } else {
                 FwdJob j{};
                 j.x1 = x; j.x2 = nullptr;

@@ -49,7 +49,7 @@ int main() {
         auto launch = [&]() {
             hipLaunchKernelGGL(k_mfma_fwd, dim3(ntm * ntn), dim3(256), 0, 0,
                                x1, x2, wt, bias, y, s.B, s.in1, s.in2,
-                               s.out, ACT_RELU);
+                               s.out, ACT_RELU, 1, (float*)nullptr);
         };
         launch();
         CHECK(hipDeviceSynchronize());
