@@ -1781,6 +1781,37 @@ k_mfma_fwd(const float* __restrict__ x1, const float* __restrict__ x2,
     }
 }
 
+// Narrow dX (span <= 8, the concat action slice): wave per batch row,
+// lane-parallel over `out` with shfl reduce.  The MFMA path for this
+// shape (one N-tile -> 64-wg grid) left 3/4 of the chip idle and its
+// staging fully exposed (~76 us for 50 MFLOP); this VALU form runs the
+// whole thing in ~10 us.
+__global__ void k_dx_narrow(const float* __restrict__ dz,
+                            const float* __restrict__ wt,
+                            const float* __restrict__ ymask,
+                            float* __restrict__ dx, int B, int in_lo,
+                            int in_hi, int out, int prev_act) {
+    int lane = threadIdx.x & 63, wrow = threadIdx.x / 64;
+    int wpb = blockDim.x / 64;
+    int span = in_hi - in_lo;
+    for (int row = blockIdx.x * wpb + wrow; row < B;
+         row += gridDim.x * wpb) {
+        for (int j = 0; j < span; ++j) {
+            const float* wrowp = wt + (long)(in_lo + j) * out;
+            float acc = 0.f;
+            for (int o = lane; o < out; o += 64)
+                acc += dz[(long)row * out + o] * wrowp[o];
+            for (int s = 32; s > 0; s >>= 1)
+                acc += __shfl_xor(acc, s, 64);
+            if (lane == 0) {
+                float m_ = ymask
+                    ? act_mask(prev_act, ymask[(long)row * span + j]) : 1.f;
+                dx[(long)row * span + j] = acc * m_;
+            }
+        }
+    }
+}
+
 // Head finish: y[row] = act(sum_seg parts[seg][row] + bias), wave == row
 // (out <= 64); softmax reduces across the wave like k_fwd3's epilogue.
 __global__ void k_head_finish(const float* __restrict__ parts,
@@ -3831,13 +3862,17 @@ public:
         // launch_bwd's and guarantees every segment is non-empty
         long dwp = 0;
         if (B >= 512) {
-            auto need = [&](const LayerDesc& l) {
-                int it = l.in1 + l.in2;
-                int ntm = ceil_div(it, MT_M), ntn = ceil_div(l.out, MT_N);
+            auto dw_need = [&](int in_n, int out) {
+                int ntm = ceil_div(in_n, MT_M), ntn = ceil_div(out, MT_N);
                 int ks = 1;
                 while (ntm * ntn * ks < 256 && ks * 2 * MT_K <= B) ks *= 2;
-                long v = (long)ks * ((long)it * l.out + l.out);
+                long v = (long)ks * ((long)in_n * out + out);
                 if (v > dwp) dwp = v;
+            };
+            auto need = [&](const LayerDesc& l) {
+                // dW runs per input block (x1 / x2 separately)
+                dw_need(l.in1, l.out);
+                if (l.in2 > 0) dw_need(l.in2, l.out);
                 if (l.out <= 64) {      // split-K head fwd partials
                     int ntmh = ceil_div(B, MT_M), ksh = 1;
                     while (ntmh * ksh < 512 && ksh * 2 * MT_K <= it)
@@ -3988,25 +4023,38 @@ public:
         int in_total = l.in1 + l.in2;
         if (cfg.batch >= 512) {
             if (want_dw) {
-                int ntm = ceil_div(in_total, MT_M), ntn = ceil_div(l.out, MT_N);
                 // split K (= batch) until the grid covers the chip;
                 // segments write disjoint partials, k_dw_reduce sums
-                // (atomic epilogue retired — see k_dw_reduce comment)
-                int ksplit = 1;
-                while (ntm * ntn * ksplit < 256 &&
-                       ksplit * 2 * MT_K <= cfg.batch)
-                    ksplit *= 2;
-                hipLaunchKernelGGL(k_mfma_dw, dim3(ntm * ntn * ksplit),
-                                   dim3(256), 0, stream, dz, x1, x2,
-                                   gslab + l.w_off, gslab + l.b_off,
-                                   cfg.batch, l.in1, l.in2, l.out, ksplit,
-                                   dw_parts);
-                if (ksplit > 1)
-                    hipLaunchKernelGGL(k_dw_reduce, dim3(1024), dim3(256),
-                                       0, stream, dw_parts,
-                                       gslab + l.w_off, gslab + l.b_off,
-                                       (long)in_total * l.out, l.out,
-                                       ksplit);
+                // (atomic epilogue retired — see k_dw_reduce comment).
+                // Concat layers (in2 > 0) run as TWO GEMMs: the x1 block
+                // with fully vectorized staging, and the tiny x2 (action)
+                // block separately — one mixed GEMM left its edge tiles
+                // on the guarded scalar path, and those straggler wgs
+                // held the whole 200-us dispatch (profiles/wide_dissect2)
+                auto dw_gemm = [&](const float* xop, int in_n, float* dwt,
+                                   float* db) {
+                    int ntm = ceil_div(in_n, MT_M);
+                    int ntn = ceil_div(l.out, MT_N);
+                    int ksplit = 1;
+                    while (ntm * ntn * ksplit < 256 &&
+                           ksplit * 2 * MT_K <= cfg.batch)
+                        ksplit *= 2;
+                    hipLaunchKernelGGL(k_mfma_dw, dim3(ntm * ntn * ksplit),
+                                       dim3(256), 0, stream, dz, xop,
+                                       (const float*)nullptr, dwt, db,
+                                       cfg.batch, in_n, 0, l.out, ksplit,
+                                       dw_parts);
+                    if (ksplit > 1)
+                        hipLaunchKernelGGL(k_dw_reduce, dim3(1024),
+                                           dim3(256), 0, stream, dw_parts,
+                                           dwt, db, (long)in_n * l.out,
+                                           l.out, ksplit);
+                };
+                dw_gemm(x1, l.in1, gslab + l.w_off, gslab + l.b_off);
+                if (l.in2 > 0)
+                    dw_gemm(x2, l.in2,
+                            gslab + l.w_off + (long)l.in1 * l.out,
+                            (float*)nullptr);
             }
             if (dx1) {
                 int ntm = ceil_div(cfg.batch, MT_M);
@@ -4016,13 +4064,24 @@ public:
                                    cfg.batch, 0, l.in1, l.out, prev_act);
             }
             if (dx2) {
-                int ntm = ceil_div(cfg.batch, MT_M);
-                int ntn = ceil_div(l.in2, MT_N);
-                hipLaunchKernelGGL(k_mfma_dx, dim3(ntm * ntn), dim3(256),
-                                   0, stream, dz, slab + l.w_off,
-                                   (const float*)nullptr, dx2,
-                                   cfg.batch, l.in1, in_total, l.out,
-                                   ACT_NONE);
+                if (l.in2 <= 8) {
+                    // action slice: wave-per-row VALU beats a 1-N-tile
+                    // MFMA grid by ~7x here (see k_dx_narrow comment)
+                    hipLaunchKernelGGL(k_dx_narrow, dim3(256), dim3(256),
+                                       0, stream, dz, slab + l.w_off,
+                                       (const float*)nullptr, dx2,
+                                       cfg.batch, l.in1, in_total, l.out,
+                                       ACT_NONE);
+                } else {
+                    int ntm = ceil_div(cfg.batch, MT_M);
+                    int ntn = ceil_div(l.in2, MT_N);
+                    hipLaunchKernelGGL(k_mfma_dx, dim3(ntm * ntn),
+                                       dim3(256), 0, stream, dz,
+                                       slab + l.w_off,
+                                       (const float*)nullptr, dx2,
+                                       cfg.batch, l.in1, in_total, l.out,
+                                       ACT_NONE);
+                }
             }
             return;
         }
