@@ -3874,6 +3874,7 @@ public:
                 dw_need(l.in1, l.out);
                 if (l.in2 > 0) dw_need(l.in2, l.out);
                 if (l.out <= 64) {      // split-K head fwd partials
+                    int it = l.in1 + l.in2;
                     int ntmh = ceil_div(B, MT_M), ksh = 1;
                     while (ntmh * ksh < 512 && ksh * 2 * MT_K <= it)
                         ksh *= 2;
