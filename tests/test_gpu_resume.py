@@ -11,9 +11,9 @@ import torch
 pytestmark = pytest.mark.gpu
 
 
-def _mk_agent(seed=0):
+def _mk_agent(seed=0, batch_size=64):
     from d4pg_amd.algo.d4pg import DDPG
-    return DDPG(3, 1, memory_size=8192, batch_size=64,
+    return DDPG(3, 1, memory_size=8192, batch_size=batch_size,
                 critic_dist_info={"type": "categorical", "v_min": -300.0,
                                   "v_max": 0.0, "n_atoms": 51},
                 n_steps=5, gamma=0.99, prioritized_replay=True,
@@ -69,3 +69,28 @@ def test_engine_resume_bitwise():
     # the loss scalar accumulates via cross-workgroup fp32 atomics, whose
     # order is nondeterministic — approx, not bitwise
     assert c1["loss_critic"] == pytest.approx(c2["loss_critic"], rel=1e-5)
+
+
+def test_engine_resume_wide_batch_graph_path():
+    """the wide/graph path (B >= 512, hipGraph
+    replay) bakes the philox seed into captured kernel ARGUMENTS, so a
+    restored seed must invalidate the graph — otherwise a resumed run
+    silently keeps the stale seed.  Verifies the post-restore trajectory
+    matches the original bitwise THROUGH graph-replayed steps."""
+    a1 = _mk_agent(seed=3, batch_size=512)
+    _fill(a1, n=3000, seed=11)
+    a1.train()                       # builds the bridge
+    eng1 = a1.engine.engine
+    eng1.train_steps(8, steps_per_graph=4)   # capture + replay
+    st = a1.state_dict()
+    eng1.train_steps(8, steps_per_graph=4)
+    ref = _flat_params(a1)
+
+    a2 = _mk_agent(seed=77, batch_size=512)
+    a2.load_state_dict(st)           # builds bridge + restores (incl seed)
+    eng2 = a2.engine.engine
+    assert eng2._captured == 0       # set_seed invalidated any capture
+    eng2.train_steps(8, steps_per_graph=4)   # recapture with restored seed
+    a2.engine.ddpg.train_steps_done += 8
+    got = _flat_params(a2)
+    np.testing.assert_array_equal(got.numpy(), ref.numpy())
