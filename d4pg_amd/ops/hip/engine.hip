@@ -2193,7 +2193,7 @@ k_mfma_dw(const float* __restrict__ dz, const float* __restrict__ x1,
 #define PWOFF (PROWS * PXMAX)
 // pool sized for the chain variant's layout (2 activation row-buffers +
 // 2 weight chunks), which is the largest user
-#define PLDS_FLOATS (2 * 4 * PXMAX + 2 * 64 * 65 + 64)
+#define PLDS_FLOATS (2 * 4 * PXMAX + 2 * 64 * 68 + 64)
 
 struct PStepArgs {
     int B, O, A, H, K;
@@ -2329,6 +2329,9 @@ __device__ __noinline__ void p_fwd(float* lds, const float* x1, const float* x2,
                              int B, int in1, int in2, int out, int act_kind,
                              int wg_rel, int nwg) {
     int in_total = in1 + in2;
+    // x rows staged at a 4-float-aligned stride so the k-loop can read
+    // them with ds_read_b128 (16 B alignment)
+    int in_pad = (in_total + 3) & ~3;
     int nrt = (B + PROWS - 1) / PROWS;
     int nct = (out + 63) / 64;
     int ntiles = nrt * nct;
@@ -2351,7 +2354,7 @@ __device__ __noinline__ void p_fwd(float* lds, const float* x1, const float* x2,
         if (nfull > 0) preload(0);
         // x-stage, register-batched (all loads in flight, then ds_writes)
         {
-            int tot = PROWS * in_total;
+            int tot = PROWS * in_pad;
             for (int base = 0; base < tot; base += 256 * 8) {
                 float tmp[8];
 #pragma unroll
@@ -2359,9 +2362,9 @@ __device__ __noinline__ void p_fwd(float* lds, const float* x1, const float* x2,
                     int e = base + u * 256 + tid;
                     float v = 0.f;
                     if (e < tot) {
-                        int rr_ = e / in_total, kk = e % in_total;
+                        int rr_ = e / in_pad, kk = e % in_pad;
                         int gb = r0 + rr_;
-                        if (gb < B)
+                        if (gb < B && kk < in_total)
                             v = (kk < in1)
                                 ? x1[(long)gb * in1 + kk]
                                 : x2[(long)gb * in2 + (kk - in1)];
@@ -2380,10 +2383,14 @@ __device__ __noinline__ void p_fwd(float* lds, const float* x1, const float* x2,
         int o = c0 + c;
         int r = r0 + rq;
         float acc = 0.f;
-        // weight slice K-chunked through LDS, register-batched + double-
-        // buffered (chunk k+1's loads fly during chunk k's FMAs)
-        float* ws = lds + PWOFF;                 // [64][65]
-        const float* xr = lds + rq * in_total;
+        // weight slice K-chunked through LDS in [col][k] layout (row
+        // stride 68 floats: 16 B-aligned and 4-bank-strided across
+        // lanes), so the k-loop fetches 4 k at a time with ONE
+        // ds_read_b128 per operand — the b32 form issued 3 LDS reads
+        // per k and was issue-bound, not bandwidth-bound.  Summation
+        // stays ascending-k: bitwise identical to the eager oracle.
+        float* ws = lds + PWOFF;                 // [64 cols][68 k]
+        const float* xr = lds + rq * in_pad;
         for (int ch = 0; ch < nfull; ++ch) {
             int kc = ch << 6;
             float wb[16];
@@ -2392,12 +2399,20 @@ __device__ __noinline__ void p_fwd(float* lds, const float* x1, const float* x2,
             if (kc + 64 < (nfull << 6)) preload(kc + 64);
 #pragma unroll
             for (int u = 0; u < 16; ++u)
-                ws[(kk16 + 4 * u) * 65 + cc16] = wb[u];
+                ws[cc16 * 68 + kk16 + 4 * u] = wb[u];
             __syncthreads();
             if (o < out) {
-#pragma unroll 8
-                for (int k = 0; k < 64; ++k)
-                    acc += xr[kc + k] * ws[k * 65 + c];
+                const float* wcol = ws + c * 68;
+#pragma unroll
+                for (int k = 0; k < 64; k += 4) {
+                    float4 wv = *reinterpret_cast<const float4*>(wcol + k);
+                    float4 xv = *reinterpret_cast<const float4*>(
+                        xr + kc + k);
+                    acc += xv.x * wv.x;
+                    acc += xv.y * wv.y;
+                    acc += xv.z * wv.z;
+                    acc += xv.w * wv.w;
+                }
             }
             __syncthreads();
         }
@@ -2406,13 +2421,13 @@ __device__ __noinline__ void p_fwd(float* lds, const float* x1, const float* x2,
             for (int e = tid; e < (klen << 6); e += 256) {
                 int kk = e >> 6, cc = e & 63;
                 int gc = c0 + cc;
-                ws[kk * 65 + cc] = (gc < out)
+                ws[cc * 68 + kk] = (gc < out)
                     ? wt[(long)(kc + kk) * out + gc] : 0.f;
             }
             __syncthreads();
             if (o < out) {
                 for (int k = 0; k < klen; ++k)
-                    acc += xr[kc + k] * ws[k * 65 + c];
+                    acc += xr[kc + k] * ws[c * 68 + k];
             }
             __syncthreads();
         }
@@ -2464,9 +2479,11 @@ __device__ inline void p_bwd_dx(float* lds, const float* dz, const float* wt,
             }
         };
         if (nfull > 0) preload(0);
-        // dz-stage, register-batched
+        // dz-stage, register-batched; rows padded to 4-float stride so
+        // the o-loop reads them with ds_read_b128
+        int out_pad = (out + 3) & ~3;
         {
-            int tot = PROWS * out;
+            int tot = PROWS * out_pad;
             for (int base = 0; base < tot; base += 256 * 8) {
                 float tmp[8];
 #pragma unroll
@@ -2474,9 +2491,10 @@ __device__ inline void p_bwd_dx(float* lds, const float* dz, const float* wt,
                     int e = base + u * 256 + tid;
                     float v = 0.f;
                     if (e < tot) {
-                        int rr_ = e / out, oo = e % out;
+                        int rr_ = e / out_pad, oo = e % out_pad;
                         int gb = r0 + rr_;
-                        if (gb < B) v = dz[(long)gb * out + oo];
+                        if (gb < B && oo < out)
+                            v = dz[(long)gb * out + oo];
                     }
                     tmp[u] = v;
                 }
@@ -2492,10 +2510,12 @@ __device__ inline void p_bwd_dx(float* lds, const float* dz, const float* wt,
         int i = i0 + c;
         int r = r0 + rq;
         float acc = 0.f;
-        // weight tile O-chunked through LDS, transposed write, double-
-        // buffered register-batched staging
-        float* ws = lds + PWOFF;                 // [64][65] = [o][i]
-        const float* zr = lds + rq * out;
+        // weight tile O-chunked through LDS in [i][o] layout (stride 68:
+        // 16 B-aligned rows), so the o-loop fetches 4 o per ds_read_b128
+        // for both operands — see p_fwd's identical rework; summation
+        // stays ascending-o, bitwise identical
+        float* ws = lds + PWOFF;                 // [64 i][68 o]
+        const float* zr = lds + rq * out_pad;
         for (int ch = 0; ch < nfull; ++ch) {
             int oc = ch << 6;
             float wb[16];
@@ -2504,12 +2524,20 @@ __device__ inline void p_bwd_dx(float* lds, const float* dz, const float* wt,
             if (oc + 64 < (nfull << 6)) preload(oc + 64);
 #pragma unroll
             for (int u = 0; u < 16; ++u)
-                ws[(oo16 + 4 * u) * 65 + ii16] = wb[u];
+                ws[ii16 * 68 + oo16 + 4 * u] = wb[u];
             __syncthreads();
             if (i < in_hi) {
-#pragma unroll 8
-                for (int o = 0; o < 64; ++o)
-                    acc += zr[oc + o] * ws[o * 65 + c];
+                const float* wrow = ws + c * 68;
+#pragma unroll
+                for (int o = 0; o < 64; o += 4) {
+                    float4 wv = *reinterpret_cast<const float4*>(wrow + o);
+                    float4 zv = *reinterpret_cast<const float4*>(
+                        zr + oc + o);
+                    acc += zv.x * wv.x;
+                    acc += zv.y * wv.y;
+                    acc += zv.z * wv.z;
+                    acc += zv.w * wv.w;
+                }
             }
             __syncthreads();
         }
@@ -2518,13 +2546,13 @@ __device__ inline void p_bwd_dx(float* lds, const float* dz, const float* wt,
             for (int e = tid; e < 4096; e += 256) {
                 int ii = e >> 6, oo = e & 63;
                 int gi = i0 + ii;
-                ws[oo * 65 + ii] = (gi < in_hi && oo < olen)
+                ws[ii * 68 + oo] = (gi < in_hi && oo < olen)
                     ? wt[(long)gi * out + (oc + oo)] : 0.f;
             }
             __syncthreads();
             if (i < in_hi) {
                 for (int o = 0; o < olen; ++o)
-                    acc += zr[oc + o] * ws[o * 65 + c];
+                    acc += zr[oc + o] * ws[c * 68 + o];
             }
             __syncthreads();
         }
