@@ -2620,8 +2620,10 @@ __device__ __noinline__ void p_dw2(float* lds, const float* dz,
     int in_total = in1 + in2;
     int nti = (in_total + 63) >> 6, nto = (out + 63) >> 6;
     int ntiles = nti * nto;
-    float* xs = lds;                       // [64 b][65] i-slice
-    float* zs = lds + 64 * 65;             // [64 b][65] o-slice
+    float* xs = lds;                       // [64 b][68] i-slice
+    float* zs = lds + 64 * 68;             // [64 b][68] o-slice
+    // row stride 68: 16 B-aligned so the b-loop reads each 4-wide
+    // register block with ONE ds_read_b128 (see p_fwd's rework)
     int tid = threadIdx.x;
     int ig = tid >> 4, og = tid & 15;
     for (int t = wg_rel; t < ntiles; t += nwg) {
@@ -2649,20 +2651,18 @@ __device__ __noinline__ void p_dw2(float* lds, const float* dz,
             for (int u = 0; u < 16; ++u) {
                 int e = u * 256 + tid;
                 int bb = e >> 6, ii = e & 63;
-                xs[bb * 65 + ii] = tx[u];
-                zs[bb * 65 + ii] = tz[u];
+                xs[bb * 68 + ii] = tx[u];
+                zs[bb * 68 + ii] = tz[u];
             }
             __syncthreads();
             int blim = min(64, B - bc);
             for (int b = 0; b < blim; ++b) {
-                float xv0 = xs[b * 65 + ig * 4 + 0];
-                float xv1 = xs[b * 65 + ig * 4 + 1];
-                float xv2 = xs[b * 65 + ig * 4 + 2];
-                float xv3 = xs[b * 65 + ig * 4 + 3];
-                float zv0 = zs[b * 65 + og * 4 + 0];
-                float zv1 = zs[b * 65 + og * 4 + 1];
-                float zv2 = zs[b * 65 + og * 4 + 2];
-                float zv3 = zs[b * 65 + og * 4 + 3];
+                float4 xq = *reinterpret_cast<const float4*>(
+                    xs + b * 68 + ig * 4);
+                float4 zq = *reinterpret_cast<const float4*>(
+                    zs + b * 68 + og * 4);
+                float xv0 = xq.x, xv1 = xq.y, xv2 = xq.z, xv3 = xq.w;
+                float zv0 = zq.x, zv1 = zq.y, zv2 = zq.z, zv3 = zq.w;
                 acc[0][0] += xv0 * zv0; acc[0][1] += xv0 * zv1;
                 acc[0][2] += xv0 * zv2; acc[0][3] += xv0 * zv3;
                 acc[1][0] += xv1 * zv0; acc[1][1] += xv1 * zv1;
