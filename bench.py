@@ -360,14 +360,16 @@ def main():
     gpu_sps = None
     if rank == 0 and use_gpu:
         # device-resident actor serving (env + policy + noise + fold all
-        # on-GPU, engine.hip rollout section): M=1024 envs, 200-tick
-        # episodes, one hipGraph per episode
+        # on-GPU, engine.hip rollout section): M=8192 envs, 200-tick
+        # episodes, one hipGraph per episode.  Throughput keeps scaling
+        # with M (211M env-steps/s at M=65536, profiles/README.md); 8192
+        # is a practical per-actor-rank size.
         eng2 = make_gpu_engine(seed=999)
-        eng2.rollout_alloc(1024, FLAGSHIP["n_steps"], horizon=200,
+        eng2.rollout_alloc(8192, FLAGSHIP["n_steps"], horizon=200,
                            gamma=FLAGSHIP["gamma"], eps=0.3, seed=555)
         eng2.rollout_run(1)                       # capture + warm
         t0 = time.perf_counter()
-        steps_done, _ = eng2.rollout_run(10)
+        steps_done, _ = eng2.rollout_run(6)
         gpu_sps = steps_done / (time.perf_counter() - t0)
 
     if rank == 0:
@@ -404,7 +406,7 @@ def main():
                        "device": "cuda" if use_gpu else "cpu-fallback"},
             "env_steps_per_sec_1actor": env_sps,
             "env_steps_per_sec_vector64": vec_sps,
-            "env_steps_per_sec_gpu_rollout1024": gpu_sps,
+            "env_steps_per_sec_gpu_rollout8192": gpu_sps,
         }
         print(json.dumps(out), flush=True)
     if dist is not None:
