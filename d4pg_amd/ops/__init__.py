@@ -223,8 +223,21 @@ class FusedEngine:
     def synth_fill(self, n, seed=1234):
         self.ext.synth_fill(self.h, int(n), int(seed))
 
+    # device staging buffer rows per ingest call (engine.hip ing_cap)
+    INGEST_CHUNK = 65536
+
     def ingest(self, s, a, r, s2, d):
-        self.ext.ingest(self.h, s, a, r, s2, d)
+        """Append transitions; chunks transparently at the engine's
+        staging-buffer capacity (large actor pushes — e.g. GPU-rollout
+        ranks at hundreds of envs — exceed one staging buffer)."""
+        n = int(r.numel() if hasattr(r, "numel") else len(r))
+        if n <= self.INGEST_CHUNK:
+            self.ext.ingest(self.h, s, a, r, s2, d)
+            return
+        for lo in range(0, n, self.INGEST_CHUNK):
+            hi = min(n, lo + self.INGEST_CHUNK)
+            self.ext.ingest(self.h, s[lo:hi], a[lo:hi], r[lo:hi],
+                            s2[lo:hi], d[lo:hi])
 
     def device_slab(self, name) -> torch.Tensor:
         """Zero-copy torch CUDA view of an engine slab — feed directly to

@@ -260,3 +260,21 @@ def test_synth_fill_and_throughput_sanity():
     assert float((q.sum(1) - 1).abs().max()) < 1e-4
     sa = eng.store_slab("actor")
     assert torch.isfinite(sa).all()
+
+
+def test_ingest_chunks_past_staging_cap():
+    """Regression: one ingest() call larger than the 65536-row device
+    staging buffer must chunk transparently (GPU-rollout actor ranks at
+    hundreds of envs push >100k transitions per exchange)."""
+    eng = make_engine(capacity=262144)
+    n = 70000
+    rng = np.random.default_rng(0)
+    eng.ingest(torch.from_numpy(rng.standard_normal((n, O)).astype("f")),
+               torch.from_numpy(rng.uniform(-1, 1, (n, A)).astype("f")),
+               torch.from_numpy(rng.uniform(-30, 0, n).astype("f")),
+               torch.from_numpy(rng.standard_normal((n, O)).astype("f")),
+               torch.from_numpy(np.zeros(n, np.float32)))
+    c = eng.counters()
+    assert c["size"] == n and c["pos"] == n
+    tree = eng.read("sum_tree").numpy()
+    assert tree[1] == pytest.approx(float(n), rel=1e-9)
