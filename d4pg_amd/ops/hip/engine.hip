@@ -1174,6 +1174,47 @@ __global__ void k_replay_add(float* __restrict__ rs, float* __restrict__ ra,
     }
 }
 
+// Bulk variant of the row-copy + leaf-init half of k_replay_add: grid-
+// stride over T rows (the one-workgroup form serializes at large T —
+// GPU-rollout actor ranks push 100k+ transitions per exchange).  The
+// caller repairs the tree with the k_tree_build_level sweep (bandwidth-
+// cheap) and bumps the counters host-side.
+__global__ void k_replay_copy(float* __restrict__ rs, float* __restrict__ ra,
+                              float* __restrict__ rr,
+                              float* __restrict__ rs2,
+                              float* __restrict__ rd,
+                              double* __restrict__ sum_tree,
+                              double* __restrict__ min_tree,
+                              long tree_cap, long capacity, long pos0,
+                              const float* __restrict__ ts,
+                              const float* __restrict__ ta,
+                              const float* __restrict__ tr,
+                              const float* __restrict__ ts2,
+                              const float* __restrict__ td,
+                              int T, int obs, int act, double pa) {
+    for (long e = (long)blockIdx.x * blockDim.x + threadIdx.x;
+         e < (long)T * obs; e += (long)gridDim.x * blockDim.x) {
+        long t = e / obs, k = e % obs;
+        long slot = (pos0 + t) % capacity;
+        rs[slot * obs + k] = ts[t * obs + k];
+        rs2[slot * obs + k] = ts2[t * obs + k];
+    }
+    for (long e = (long)blockIdx.x * blockDim.x + threadIdx.x;
+         e < (long)T * act; e += (long)gridDim.x * blockDim.x) {
+        long t = e / act, k = e % act;
+        long slot = (pos0 + t) % capacity;
+        ra[slot * act + k] = ta[t * act + k];
+    }
+    for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < T;
+         t += (long)gridDim.x * blockDim.x) {
+        long slot = (pos0 + t) % capacity;
+        rr[slot] = tr[t];
+        rd[slot] = td[t];
+        sum_tree[tree_cap + slot] = pa;
+        min_tree[tree_cap + slot] = pa;
+    }
+}
+
 // ---- synthetic replay fill (bench path: no H2D needed) ----------------------
 __global__ void k_synth_fill(float* rs, float* ra, float* rr, float* rs2,
                              float* rd, double* sum_tree, double* min_tree,
@@ -4573,10 +4614,34 @@ public:
                                  hipMemcpyHostToDevice, stream));
         HIP_CHECK(hipMemcpyAsync(ing_d, hd, (long)T * 4,
                                  hipMemcpyHostToDevice, stream));
-        hipLaunchKernelGGL(k_replay_add, dim3(1), dim3(1024), 0, stream,
-                           rs, ra, rr, rs2, rd, sum_tree, min_tree, tree_cap,
-                           cfg.capacity, ing_s, ing_a, ing_r, ing_s2, ing_d,
-                           T, O, A, cfg.per_alpha, cnt);
+        if (T <= 4096) {
+            hipLaunchKernelGGL(k_replay_add, dim3(1), dim3(1024), 0, stream,
+                               rs, ra, rr, rs2, rd, sum_tree, min_tree,
+                               tree_cap, cfg.capacity, ing_s, ing_a, ing_r,
+                               ing_s2, ing_d, T, O, A, cfg.per_alpha, cnt);
+        } else {
+            // bulk path: grid-wide copy + leaf init, then a full
+            // bandwidth-bound tree rebuild (far cheaper than one wg
+            // chasing 20-level paths for 100k rows)
+            Counters h{};
+            HIP_CHECK(hipMemcpyAsync(&h, cnt, sizeof(h),
+                                     hipMemcpyDeviceToHost, stream));
+            HIP_CHECK(hipStreamSynchronize(stream));
+            double pa = pow((double)h.max_priority, (double)cfg.per_alpha);
+            hipLaunchKernelGGL(k_replay_copy, dim3(1024), dim3(256), 0,
+                               stream, rs, ra, rr, rs2, rd, sum_tree,
+                               min_tree, tree_cap, cfg.capacity, h.pos,
+                               ing_s, ing_a, ing_r, ing_s2, ing_d, T, O, A,
+                               pa);
+            for (long lo = tree_cap / 2; lo >= 1; lo >>= 1)
+                hipLaunchKernelGGL(k_tree_build_level, dim3(1024),
+                                   dim3(256), 0, stream, sum_tree,
+                                   min_tree, lo, 2 * lo);
+            h.pos = (h.pos + T) % cfg.capacity;
+            h.size = h.size + T > cfg.capacity ? cfg.capacity : h.size + T;
+            HIP_CHECK(hipMemcpyAsync(cnt, &h, sizeof(h),
+                                     hipMemcpyHostToDevice, stream));
+        }
         HIP_CHECK(hipStreamSynchronize(stream));
     }
 
