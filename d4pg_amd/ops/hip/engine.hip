@@ -2225,7 +2225,11 @@ k_mfma_dw(const float* __restrict__ dz, const float* __restrict__ x1,
 // parallelism.  nsteps loop inside the kernel => zero host involvement
 // between steps (the hipGraph path is only needed for the wide-batch
 // configs that use the per-layer kernels above).
-#define PNWG 64
+#define PNWG 96
+// derived persistent-grid split points (fractions of PNWG)
+#define PW4 (PNWG / 4)
+#define PW3A (PNWG / 3)
+#define PW3B (PNWG - 2 * (PNWG / 3))
 // 4-row x 64-col tiles: a 64x256x256 GEMM phase becomes 64 tiles — the
 // whole grid — with ~1 us of FMA per tile (16-row tiles measured ~6 us of
 // per-wave compute and left 48 workgroups idle; scripts/tile_bench.hip)
@@ -2274,7 +2278,8 @@ struct PStepArgs {
         (g).tstamp[i] = __builtin_amdgcn_s_memrealtime(); } while (0)
 
 // software grid barrier — tree arrival + write-once go-flag.  Safe because
-// all PNWG workgroups are co-resident (64 wgs of 256 threads on 256 CUs).
+// all PNWG workgroups are co-resident (256 wgs of 256 threads, one per CU
+// — validated by the ctor's occupancy gate).
 //
 // Layout in gbar[]: 8 group counters at [g*16] (separate 128 B lines),
 // root counter at [128], go-flag at [144], cross-launch round base at
@@ -3501,59 +3506,60 @@ k_step_persistent(PStepArgs g, int nsteps) {
     for (int s = 0; s < nsteps; ++s) {
                PTIME(g, s, 0);
         // PH0: PER sample + batch gather (only on a launch's first step —
-        // later steps were pre-sampled under the previous actor-Adam)
-        if (s == 0 || B > 64) {      // 16 overlap wgs cover 64 probes
+        // later steps were pre-sampled under the previous actor-Adam,
+        // whose 64 overlap wgs cover 256 probes)
+        if (s == 0 || B > (PNWG - PNWG * 3 / 4) * 4) {
             p_sample(g, 0, 0, ep0 + s, bt0 + s);
             p_bar(ctr, tgt);
         }
         PTIME(g, s, 1);
         // PH1: four independent L1s (16 wgs each)
-        if (wg < 16)
+        if (wg < PW4)
             p_fwd(lds, g.bs2, nullptr, at.w1, at.b1, g.at_h1, B, O, 0, H,
-                  ACT_RELU, wg, 16);
-        else if (wg < 32)
+                  ACT_RELU, wg, PW4);
+        else if (wg < 2 * PW4)
             p_fwd(lds, g.bs2, nullptr, ct.w1, ct.b1, g.ct_h1, B, O, 0, H,
-                  ACT_RELU, wg - 16, 16);
-        else if (wg < 48)
+                  ACT_RELU, wg - PW4, PW4);
+        else if (wg < 3 * PW4)
             p_fwd(lds, g.bs, nullptr, c.w1, c.b1, g.c_h1, B, O, 0, H,
-                  ACT_RELU, wg - 32, 16);
+                  ACT_RELU, wg - 2 * PW4, PW4);
         else
             p_fwd(lds, g.bs, nullptr, a.w1, a.b1, g.pa_h1, B, O, 0, H,
-                  ACT_RELU, wg - 48, 16);
+                  ACT_RELU, wg - 3 * PW4, PW4);
         p_bar(ctr, tgt); PTIME(g, s, 2);
         // PH2: actor_t.L2 | critic.L2(cat h1, a) | actor.L2
-        if (wg < 21)
+        if (wg < PW3A)
             p_fwd(lds, g.at_h1, nullptr, at.w2, at.b2, g.at_h2, B, H, 0, H,
-                  ACT_NONE, wg, 21);
-        else if (wg < 42)
+                  ACT_NONE, wg, PW3A);
+        else if (wg < 2 * PW3A)
             p_fwd(lds, g.c_h1, g.ba, c.w2, c.b2, g.c_h2, B, H, A, H,
-                  ACT_RELU, wg - 21, 21);
+                  ACT_RELU, wg - PW3A, PW3A);
         else
             p_fwd(lds, g.pa_h1, nullptr, a.w2, a.b2, g.pa_h2, B, H, 0, H,
-                  ACT_NONE, wg - 42, 22);
+                  ACT_NONE, wg - 2 * PW3A, PW3B);
         p_bar(ctr, tgt); PTIME(g, s, 3);
         // PH3: L3s
-        if (wg < 21)
+        if (wg < PW3A)
             p_fwd(lds, g.at_h2, nullptr, at.w3, at.b3, g.at_h3, B, H, 0, H,
-                  ACT_RELU, wg, 21);
-        else if (wg < 42)
+                  ACT_RELU, wg, PW3A);
+        else if (wg < 2 * PW3A)
             p_fwd(lds, g.c_h2, nullptr, c.w3, c.b3, g.c_h3, B, H, 0, H,
-                  ACT_RELU, wg - 21, 21);
+                  ACT_RELU, wg - PW3A, PW3A);
         else
             p_fwd(lds, g.pa_h2, nullptr, a.w3, a.b3, g.pa_h3, B, H, 0, H,
-                  ACT_RELU, wg - 42, 22);
+                  ACT_RELU, wg - 2 * PW3A, PW3B);
         p_bar(ctr, tgt); PTIME(g, s, 4);
         // PH4: heads — actor_t tanh -> a2 | critic softmax -> q |
         //      actor tanh -> a_out
-        if (wg < 21)
+        if (wg < PW3A)
             p_fwd(lds, g.at_h3, nullptr, at.w4, at.b4, g.a2, B, H, 0, A,
-                  ACT_TANH, wg, 21);
-        else if (wg < 42)
+                  ACT_TANH, wg, PW3A);
+        else if (wg < 2 * PW3A)
             p_fwd(lds, g.c_h3, nullptr, c.w4, c.b4, g.q, B, H, 0, K,
-                  ACT_SOFTMAX, wg - 21, 21);
+                  ACT_SOFTMAX, wg - PW3A, PW3A);
         else
             p_fwd(lds, g.pa_h3, nullptr, a.w4, a.b4, g.a_out, B, H, 0, A,
-                  ACT_TANH, wg - 42, 22);
+                  ACT_TANH, wg - 2 * PW3A, PW3B);
         // per-step loss accumulators zeroed here (grid barrier below
         // orders this before the quad-chained proj-CE / pgrad adds)
         if (wg == PNWG - 1 && threadIdx.x == 0) {
@@ -3610,18 +3616,21 @@ k_step_persistent(PStepArgs g, int nsteps) {
         p_bar(ctr, tgt);
         PTIME(g, s, 10); PTIME(g, s, 11); PTIME(g, s, 12);
         // PH13: critic dW, 4 jobs split by tile count (l1 is the biggest)
-        if (wg < 4)
+        if (wg < PNWG / 16)
             p_dw2(lds, g.d1, g.bs, nullptr, g.g_critic + g.cl[0].w_off,
-                 g.g_critic + g.cl[0].b_off, B, O, 0, H, wg, 4);
-        else if (wg < 32)
+                 g.g_critic + g.cl[0].b_off, B, O, 0, H, wg, PNWG / 16);
+        else if (wg < PNWG / 2)
             p_dw2(lds, g.d2, g.c_h1, g.ba, g.g_critic + g.cl[1].w_off,
-                 g.g_critic + g.cl[1].b_off, B, H, A, H, wg - 4, 28);
-        else if (wg < 56)
+                 g.g_critic + g.cl[1].b_off, B, H, A, H, wg - PNWG / 16,
+                 PNWG / 2 - PNWG / 16);
+        else if (wg < PNWG * 7 / 8)
             p_dw2(lds, g.d3, g.c_h2, nullptr, g.g_critic + g.cl[2].w_off,
-                 g.g_critic + g.cl[2].b_off, B, H, 0, H, wg - 32, 24);
+                 g.g_critic + g.cl[2].b_off, B, H, 0, H, wg - PNWG / 2,
+                 PNWG * 7 / 8 - PNWG / 2);
         else
             p_dw2(lds, g.dlog, g.c_h3, nullptr, g.g_critic + g.cl[3].w_off,
-                 g.g_critic + g.cl[3].b_off, B, H, 0, K, wg - 56, 8);
+                 g.g_critic + g.cl[3].b_off, B, H, 0, K, wg - PNWG * 7 / 8,
+                 PNWG - PNWG * 7 / 8);
         p_bar(ctr, tgt); PTIME(g, s, 13);
         // PH14: Adam + target soft-update, critic
         p_adam_lerp(g.p_critic, g.g_critic, g.m_critic, g.v_critic,
@@ -3689,30 +3698,33 @@ k_step_persistent(PStepArgs g, int nsteps) {
         // last workgroup (priorities have been final since the proj+CE
         // phase; the counter tick moves to the final phase so both Adams
         // still read this step's t)
-        if (wg < 4)
+        if (wg < PNWG / 16)
             p_dw2(lds, g.az1, g.bs, nullptr, g.g_actor + g.al[0].w_off,
-                 g.g_actor + g.al[0].b_off, B, O, 0, H, wg, 4);
-        else if (wg < 30)
+                 g.g_actor + g.al[0].b_off, B, O, 0, H, wg, PNWG / 16);
+        else if (wg < PNWG * 15 / 32)
             p_dw2(lds, g.az2, g.pa_h1, nullptr, g.g_actor + g.al[1].w_off,
-                 g.g_actor + g.al[1].b_off, B, H, 0, H, wg - 4, 26);
-        else if (wg < 56)
+                 g.g_actor + g.al[1].b_off, B, H, 0, H, wg - PNWG / 16,
+                 PNWG * 15 / 32 - PNWG / 16);
+        else if (wg < PNWG * 7 / 8)
             p_dw2(lds, g.az3, g.pa_h2, nullptr, g.g_actor + g.al[2].w_off,
-                 g.g_actor + g.al[2].b_off, B, H, 0, H, wg - 30, 26);
-        else if (wg < 63)
+                 g.g_actor + g.al[2].b_off, B, H, 0, H, wg - PNWG * 15 / 32,
+                 PNWG * 7 / 8 - PNWG * 15 / 32);
+        else if (wg < PNWG - 2)
             p_dw2(lds, g.adz, g.pa_h3, nullptr, g.g_actor + g.al[3].w_off,
-                 g.g_actor + g.al[3].b_off, B, H, 0, A, wg - 56, 7);
-        else
+                 g.g_actor + g.al[3].b_off, B, H, 0, A, wg - PNWG * 7 / 8,
+                 PNWG - 2 - PNWG * 7 / 8);
+        else if (wg == PNWG - 1)
             p_per_update(g, false, PNWG - 1);
         p_bar(ctr, tgt); PTIME(g, s, 26);
         // PH27: Adam + soft-update, actor (wgs 0-47) overlapped with the
         // NEXT step's PER sample + gather (wgs 48-63; tree was repaired in
         // PH26, schedule counters offset by the pending tick)
-        if (wg < 48)
+        if (wg < PNWG * 3 / 4)
             p_adam_lerp(g.p_actor, g.g_actor, g.m_actor, g.v_actor,
                         g.p_actor_t, g.n_actor, g.lr_actor, g.tau,
-                        ta0 + s, 48);
-        else if (s + 1 < nsteps && B <= 64)
-            p_sample(g, 48, 1, ep0 + s + 1, bt0 + s + 1);
+                        ta0 + s, PNWG * 3 / 4);
+        else if (s + 1 < nsteps && B <= (PNWG - PNWG * 3 / 4) * 4)
+            p_sample(g, PNWG * 3 / 4, 1, ep0 + s + 1, bt0 + s + 1);
         p_bar(ctr, tgt); PTIME(g, s, 27); PTIME(g, s, 28);
         // (the per-step schedule-counter tick phase is gone: counters were
         // hoisted to base+s registers; final values land below)
@@ -3959,7 +3971,7 @@ public:
             for (int i = 0; i < 4; ++i) { need(anet.l[i]); need(cnet.l[i]); }
         }
         dw_parts = carve<float>(dwp, off);
-        gbar = carve<unsigned long long>(512, off);
+        gbar = carve<unsigned long long>(1536, off);
         tstamp = carve<unsigned long long>(64, off);
         ing_s = carve<float>((long)ing_cap * O, off);
         ing_a = carve<float>((long)ing_cap * A, off);
@@ -4192,7 +4204,7 @@ public:
         // deadlocks otherwise — persistent_fits_ is the ctor's occupancy
         // check; on an occupancy-limited device we fall back to the
         // row-block + hipGraph path instead of corrupting state)
-        return persistent_fits_ && cfg.batch <= PNWG * 4 &&
+        return persistent_fits_ && cfg.batch <= 256 &&
                cfg.obs <= PXMAX && cfg.hidden + cfg.act <= PXMAX &&
                cfg.hidden <= PXMAX && cfg.atoms <= 64;
     }
