@@ -2255,8 +2255,10 @@ struct PStepArgs {
     // replay + trees
     const float *rs, *ra, *rr, *rs2, *rd;
     double *sum_tree, *min_tree;
-    // batch
-    float *bs, *ba, *br, *bs2, *bd, *bw, *pri;
+    // batch (bs is double-buffered: the policy-block tail pre-samples
+    // the NEXT step's batch while this step's bs is still read by the
+    // policy forward and the actor dW)
+    float *bs, *bs_b, *ba, *br, *bs2, *bd, *bw, *pri;
     long *bidx;
     // activations / deltas workspace
     float *at_h1, *at_h2, *at_h3, *a2;
@@ -2363,6 +2365,26 @@ __device__ inline void q_bar(unsigned long long* qc,
 // (A per-member slot-store variant — release stores + one-line relaxed
 // polls instead of the shared atomic counter — measured SLOWER: 3674 vs
 // 3773 steps/s; the four poll loads don't coalesce into one L2 access.)
+
+// group barrier for the policy-block tail crew (q_bar with a caller-set
+// member count; own line at gbar[1280])
+__device__ inline void t_bar(unsigned long long* tc,
+                             unsigned long long& trnd, int members) {
+    __syncthreads();
+    trnd += members;
+    if (threadIdx.x == 0) {
+        __threadfence();
+        atomicAdd(tc, 1ull);
+        volatile unsigned long long* f = tc;
+        long spins = 0;
+        while (*f < trnd) {
+            if (++spins > (1L << 26)) break;
+            __builtin_amdgcn_s_sleep(1);
+        }
+        __threadfence();
+    }
+    __syncthreads();
+}
 
 // Tiled forward: 16-row x 64-col tiles; x rows staged in LDS (broadcast
 // reads), weight column read ONCE per tile (not once per row).  Thread
@@ -2793,7 +2815,10 @@ __device__ inline void p_adam_lerp(float* __restrict__ p,
 __device__ inline void p_sample(const PStepArgs& g, int wg0 = 0,
                                 int sched_off = 0,
                                 long long epoch_ovr = -1,
-                                long long beta_ovr = -1) {
+                                long long beta_ovr = -1,
+                                float* bs_out = nullptr) {
+    // bs_out: which state buffer to gather into (double-buffered in the
+    // persistent kernel; defaults to g.bs)
     // sched_off = +1 when pre-sampling the NEXT step's batch before the
     // counter tick (overlapped under the actor-Adam phase).  epoch_ovr /
     // beta_ovr: explicit schedule values from the hoisted-counter path
@@ -2863,10 +2888,11 @@ __device__ inline void p_sample(const PStepArgs& g, int wg0 = 0,
         g.bd[probe] = g.rd[idx];
     }
     idx = __shfl(idx, 0, 64);
+    float* bs_dst = bs_out ? bs_out : g.bs;
     // replay rows are random and never re-read — keep them out of L2
     // (the weights need it; SURVEY hot-loop note)
     for (int k = lane; k < g.O; k += 64) {
-        g.bs[(long)probe * g.O + k] =
+        bs_dst[(long)probe * g.O + k] =
             __builtin_nontemporal_load(&g.rs[idx * g.O + k]);
         g.bs2[(long)probe * g.O + k] =
             __builtin_nontemporal_load(&g.rs2[idx * g.O + k]);
@@ -3502,14 +3528,23 @@ k_step_persistent(PStepArgs g, int nsteps) {
     const long long ep0 = g.cnt->rng_epoch;
     const long long ta0 = g.cnt->adam_t_actor;
     const long long tc0 = g.cnt->adam_t_critic;
+    // bs is double-buffered by step parity so the policy-block tail crew
+    // can pre-sample the NEXT step's batch while this step's bs is still
+    // read by the policy forward and the actor dW.  Parity derives from
+    // the beta counter, so resume/relaunch stay deterministic.
+    const int QW = ((B + 3) / 4) * 4;        // wgs used by the quad chains
+    const bool otail = (QW + 32 <= PNWG) && (B <= 128);
+    unsigned long long trnd = ctr[1280];
 
     for (int s = 0; s < nsteps; ++s) {
+        float* bs_cur = ((bt0 + s) & 1) ? g.bs_b : g.bs;
+        float* bs_nxt = ((bt0 + s) & 1) ? g.bs : g.bs_b;
                PTIME(g, s, 0);
         // PH0: PER sample + batch gather (only on a launch's first step —
         // later steps were pre-sampled under the previous actor-Adam,
         // whose 64 overlap wgs cover 256 probes)
         if (s == 0 || B > (PNWG - PNWG * 3 / 4) * 4) {
-            p_sample(g, 0, 0, ep0 + s, bt0 + s);
+            p_sample(g, 0, 0, ep0 + s, bt0 + s, bs_cur);
             p_bar(ctr, tgt);
         }
         PTIME(g, s, 1);
@@ -3517,14 +3552,16 @@ k_step_persistent(PStepArgs g, int nsteps) {
         if (wg < PW4)
             p_fwd(lds, g.bs2, nullptr, at.w1, at.b1, g.at_h1, B, O, 0, H,
                   ACT_RELU, wg, PW4);
-        else if (wg < 2 * PW4)
-            p_fwd(lds, g.bs2, nullptr, ct.w1, ct.b1, g.ct_h1, B, O, 0, H,
-                  ACT_RELU, wg - PW4, PW4);
-        else if (wg < 3 * PW4)
-            p_fwd(lds, g.bs, nullptr, c.w1, c.b1, g.c_h1, B, O, 0, H,
-                  ACT_RELU, wg - 2 * PW4, PW4);
-        else
-            p_fwd(lds, g.bs, nullptr, a.w1, a.b1, g.pa_h1, B, O, 0, H,
+        else if (wg < 2 * PW4) {
+            if (!otail || s == 0)     // else precomputed by the tail crew
+                p_fwd(lds, g.bs2, nullptr, ct.w1, ct.b1, g.ct_h1, B, O, 0,
+                      H, ACT_RELU, wg - PW4, PW4);
+        } else if (wg < 3 * PW4) {
+            if (!otail || s == 0)
+                p_fwd(lds, bs_cur, nullptr, c.w1, c.b1, g.c_h1, B, O, 0, H,
+                      ACT_RELU, wg - 2 * PW4, PW4);
+        } else
+            p_fwd(lds, bs_cur, nullptr, a.w1, a.b1, g.pa_h1, B, O, 0, H,
                   ACT_RELU, wg - 3 * PW4, PW4);
         p_bar(ctr, tgt); PTIME(g, s, 2);
         // PH2: actor_t.L2 | critic.L2(cat h1, a) | actor.L2
@@ -3617,7 +3654,7 @@ k_step_persistent(PStepArgs g, int nsteps) {
         PTIME(g, s, 10); PTIME(g, s, 11); PTIME(g, s, 12);
         // PH13: critic dW, 4 jobs split by tile count (l1 is the biggest)
         if (wg < PNWG / 16)
-            p_dw2(lds, g.d1, g.bs, nullptr, g.g_critic + g.cl[0].w_off,
+            p_dw2(lds, g.d1, bs_cur, nullptr, g.g_critic + g.cl[0].w_off,
                  g.g_critic + g.cl[0].b_off, B, O, 0, H, wg, PNWG / 16);
         else if (wg < PNWG / 2)
             p_dw2(lds, g.d2, g.c_h1, g.ba, g.g_critic + g.cl[1].w_off,
@@ -3649,11 +3686,28 @@ k_step_persistent(PStepArgs g, int nsteps) {
         {
             int q = wg >> 2, m = wg & 3;
             unsigned long long* qc = ctr + 192 + (q << 4);
+            if (otail && wg >= QW && wg < QW + 32) {
+                // tail crew (32 spare wgs): this step's PER write-back,
+                // then the NEXT step's sample and its ct/c L1 forwards —
+                // all hidden under the ~70 us policy megachain.  Safe:
+                // bs2/ba/br/bd/bw/bidx of step s were fully consumed
+                // before this block; bs_nxt is the other bs buffer.
+                p_per_update(g, false, QW);
+                t_bar(ctr + 1280, trnd, 32);
+                if (s + 1 < nsteps) {
+                    p_sample(g, QW, 1, ep0 + s + 1, bt0 + s + 1, bs_nxt);
+                    t_bar(ctr + 1280, trnd, 32);
+                    p_fwd(lds, g.bs2, nullptr, ct.w1, ct.b1, g.ct_h1, B,
+                          O, 0, H, ACT_RELU, wg - QW, 32);
+                    p_fwd(lds, bs_nxt, nullptr, c.w1, c.b1, g.c_h1, B, O,
+                          0, H, ACT_RELU, wg - QW, 32);
+                }
+            }
             for (int rg = q; rg < (B + 3) / 4; rg += PNWG / 4) {
                 long r0 = (long)rg * 4;
                 int nb = min(4, B - (int)r0);
                 PTIME(g, s, 40);
-                p_fwd(lds, g.bs + r0 * O, nullptr, c.w1, c.b1,
+                p_fwd(lds, bs_cur + r0 * O, nullptr, c.w1, c.b1,
                       g.pc_h1 + r0 * H, nb, O, 0, H, ACT_RELU, m, 4);
                 q_bar(qc, qrnd); PTIME(g, s, 41);
                 p_fwd(lds, g.pc_h1 + r0 * H, g.a_out + r0 * A, c.w2, c.b2,
@@ -3699,7 +3753,7 @@ k_step_persistent(PStepArgs g, int nsteps) {
         // phase; the counter tick moves to the final phase so both Adams
         // still read this step's t)
         if (wg < PNWG / 16)
-            p_dw2(lds, g.az1, g.bs, nullptr, g.g_actor + g.al[0].w_off,
+            p_dw2(lds, g.az1, bs_cur, nullptr, g.g_actor + g.al[0].w_off,
                  g.g_actor + g.al[0].b_off, B, O, 0, H, wg, PNWG / 16);
         else if (wg < PNWG * 15 / 32)
             p_dw2(lds, g.az2, g.pa_h1, nullptr, g.g_actor + g.al[1].w_off,
@@ -3713,18 +3767,22 @@ k_step_persistent(PStepArgs g, int nsteps) {
             p_dw2(lds, g.adz, g.pa_h3, nullptr, g.g_actor + g.al[3].w_off,
                  g.g_actor + g.al[3].b_off, B, H, 0, A, wg - PNWG * 7 / 8,
                  PNWG - 2 - PNWG * 7 / 8);
-        else if (wg == PNWG - 1)
-            p_per_update(g, false, PNWG - 1);
+        else if (!otail && wg == PNWG - 1)
+            p_per_update(g, false, PNWG - 1);   // tail crew did it when otail
         p_bar(ctr, tgt); PTIME(g, s, 26);
         // PH27: Adam + soft-update, actor (wgs 0-47) overlapped with the
         // NEXT step's PER sample + gather (wgs 48-63; tree was repaired in
         // PH26, schedule counters offset by the pending tick)
-        if (wg < PNWG * 3 / 4)
-            p_adam_lerp(g.p_actor, g.g_actor, g.m_actor, g.v_actor,
-                        g.p_actor_t, g.n_actor, g.lr_actor, g.tau,
-                        ta0 + s, PNWG * 3 / 4);
-        else if (s + 1 < nsteps && B <= (PNWG - PNWG * 3 / 4) * 4)
-            p_sample(g, PNWG * 3 / 4, 1, ep0 + s + 1, bt0 + s + 1);
+        {
+            int anw = otail ? PNWG : PNWG * 3 / 4;
+            if (wg < anw)
+                p_adam_lerp(g.p_actor, g.g_actor, g.m_actor, g.v_actor,
+                            g.p_actor_t, g.n_actor, g.lr_actor, g.tau,
+                            ta0 + s, anw);
+            else if (s + 1 < nsteps && B <= (PNWG - PNWG * 3 / 4) * 4)
+                p_sample(g, PNWG * 3 / 4, 1, ep0 + s + 1, bt0 + s + 1,
+                         bs_nxt);
+        }
         p_bar(ctr, tgt); PTIME(g, s, 27); PTIME(g, s, 28);
         // (the per-step schedule-counter tick phase is gone: counters were
         // hoisted to base+s registers; final values land below)
@@ -3800,7 +3858,7 @@ public:
     double *sum_tree, *min_tree;
     Counters* cnt;
     // batch + activations workspace
-    float *bs, *ba, *br, *bs2, *bd, *bw, *pri;
+    float *bs, *bs_b, *ba, *br, *bs2, *bd, *bw, *pri;
     long *bidx;
     float *at_h1, *at_h2, *at_h3, *a2;              // actor_target path
     float *ct_h1, *ct_h2, *ct_h3, *p_t, *m_proj;    // critic_target path
@@ -3902,7 +3960,9 @@ public:
         sum_tree = carve<double>(2 * tree_cap, off);
         min_tree = carve<double>(2 * tree_cap, off);
         cnt = carve<Counters>(1, off);
-        bs = carve<float>((long)B * O, off); ba = carve<float>((long)B * A, off);
+        bs = carve<float>((long)B * O, off);
+        bs_b = carve<float>((long)B * O, off);
+        ba = carve<float>((long)B * A, off);
         br = carve<float>(B, off); bs2 = carve<float>((long)B * O, off);
         bd = carve<float>(B, off); bw = carve<float>(B, off);
         pri = carve<float>(B, off);
@@ -4229,7 +4289,8 @@ public:
         g.m_critic = m_critic; g.v_critic = v_critic;
         g.rs = rs; g.ra = ra; g.rr = rr; g.rs2 = rs2; g.rd = rd;
         g.sum_tree = sum_tree; g.min_tree = min_tree;
-        g.bs = bs; g.ba = ba; g.br = br; g.bs2 = bs2; g.bd = bd; g.bw = bw;
+        g.bs = bs; g.bs_b = bs_b;
+        g.ba = ba; g.br = br; g.bs2 = bs2; g.bd = bd; g.bw = bw;
         g.pri = pri; g.bidx = bidx;
         g.at_h1 = at_h1; g.at_h2 = at_h2; g.at_h3 = at_h3; g.a2 = a2;
         g.ct_h1 = ct_h1; g.ct_h2 = ct_h2; g.ct_h3 = ct_h3; g.p_t = p_t;

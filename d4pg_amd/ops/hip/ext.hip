@@ -304,9 +304,17 @@ static torch::Tensor read_buffer(int64_t h, std::string name) {
     Engine& e = get(h);
     const int B = e.cfg.batch, O = e.cfg.obs, A = e.cfg.act,
               H = e.cfg.hidden, K = e.cfg.atoms;
+    // the persistent path double-buffers bs by step parity; resolve the
+    // buffer the LAST completed step actually used
+    if (name == "bs" && e.use_persistent()) {
+        Counters c = e.read_counters();
+        if (c.beta_t > 0 && ((c.beta_t - 1) & 1))
+            name = "bs_b";
+    }
     struct Ent { const void* p; std::vector<long> shape; bool is_long; bool is_double; };
     std::unordered_map<std::string, Ent> m = {
         {"bs", {e.bs, {B, O}, false, false}},
+        {"bs_b", {e.bs_b, {B, O}, false, false}},
         {"ba", {e.ba, {B, A}, false, false}},
         {"br", {e.br, {B}, false, false}},
         {"bs2", {e.bs2, {B, O}, false, false}},
