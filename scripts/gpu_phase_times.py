@@ -8,13 +8,16 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 from scripts.gpu_microbench import make_engine  # noqa: E402
 
+# stamp map after the round-2 quad restructure: fused regions stamp all
+# their legacy indices back-to-back, so collapsed rows read ~0 us and the
+# preceding row carries the whole block
 PHASES = [
     "PH0 sample", "PH1 L1x4", "PH2 L2x3", "PH3 L3x3", "PH4 headsx3",
-    "PH5 ct.L2", "PH6 ct.L3", "PH7 ct.L4sm", "PH8 proj+ce",
-    "PH9 c.dX4", "PH10 c.dX3", "PH11 c.dX2", "PH12 c.dW", "PH13 c.adam",
-    "PH14 pc.L1", "PH15 pc.L2", "PH16 pc.L3", "PH17 pc.L4sm", "PH18 pgrad",
-    "PH19 p.dX4", "PH20 p.dX3", "PH21 p.dX2a", "PH22 a.dX4", "PH23 a.dX3",
-    "PH24 a.dX2", "PH25 a.dW", "PH26 a.adam+nx", "PH27 (fused)",
+    "quadA ct+proj", "(fused)", "(fused)", "(fused)",
+    "quadB c.dX", "(fused)", "(fused)", "c.dW", "c.adam",
+    "quadC policy+tail", "(fused)", "(fused)", "(fused)", "(fused)",
+    "(fused)", "(fused)", "(fused)", "(fused)", "(fused)",
+    "(fused)", "a.dW", "a.adam", "(fused)",
 ]
 
 eng = make_engine()
