@@ -132,3 +132,25 @@ def test_staleness_tag_is_version_pulled():
     actor.round()
     learner.drain_mail()
     assert learner.staleness[-1] == 2
+
+
+def test_learner_handles_empty_and_out_of_order_actors():
+    """Mail from several actors with interleaved publish order; drain
+    ingests everything exactly once."""
+    from d4pg_amd.parallel.elastic import (ElasticActor, ElasticLearner,
+                                           make_store)
+    master = make_store(port=PORT + 4, is_master=True)
+    agent = _mk_agent()
+    learner = ElasticLearner(agent, master, 3, 1)
+    actors = [ElasticActor(i, _mk_agent(), master, 3, 1,
+                           lambda ag, i=i: _collect(ag, n=7, seed=i))
+              for i in (3, 9, 17)]
+    for _ in range(3):
+        for a in actors:
+            a.round()
+    got = learner.drain_mail()
+    assert got == 3 * 3 * 7
+    assert sorted(learner.drained) == [3, 9, 17]
+    assert all(learner.drained[a.aid] == 3 for a in actors)
+    # second drain with no new mail is a no-op
+    assert learner.drain_mail() == 0

@@ -96,3 +96,25 @@ def test_load_state_dict_replay_format_guard():
     # load_replay=False skips the replay blob and succeeds
     fresh.load_state_dict(st, load_replay=False)
     assert fresh.train_steps_done == agent.train_steps_done
+
+
+def test_config_dataclass_roundtrip():
+    """D4PGConfig.from_args keeps every CLI flag (known fields + extras)."""
+    from d4pg_amd.config import D4PGConfig, make_parser
+    args = make_parser().parse_args(
+        ["--noise", "ou", "--vector_envs", "16", "--gpu_actors", "1"])
+    cfg = D4PGConfig.from_args(args)
+    assert cfg.noise == "ou" and cfg.vector_envs == 16
+    assert cfg.gpu_actors == 1
+    # no flag silently dropped
+    known = set(D4PGConfig.__dataclass_fields__) - {"extra"}
+    assert set(vars(args)) <= known | set(cfg.extra)
+
+
+def test_ingest_chunk_constant_matches_engine():
+    """FusedEngine.INGEST_CHUNK must match engine.hip's ing_cap."""
+    import re
+    from d4pg_amd.ops import FusedEngine
+    src = open("d4pg_amd/ops/hip/engine.hip").read()
+    m = re.search(r"ing_cap = (\d+);", src)
+    assert m and int(m.group(1)) == FusedEngine.INGEST_CHUNK
