@@ -210,3 +210,37 @@ def test_bench_multirank_gloo_end_to_end():
     assert out["n_gpus"] == 2
     assert out["config"]["parallelism"].startswith("hogwild-rccl x2")
     assert out["value"] > 0
+
+
+def test_split_step_matches_full_step_wide_mfma_path():
+    """Split-step (DP insertion points) parity on the WIDE path too —
+    B >= 512 takes the per-layer MFMA kernels, a different code path from
+    the row-block form covered above."""
+    from d4pg_amd.models import actor, critic
+    from d4pg_amd.ops import FusedEngine
+
+    def mk(seed):
+        eng = FusedEngine(obs_dim=17, act_dim=6, hidden=256, n_atoms=51,
+                          batch=1024, capacity=8192, v_min=-300.0,
+                          v_max=0.0, gamma_n=0.99 ** 5, tau=0.001,
+                          lr_actor=1e-4, lr_critic=1e-4, seed=seed)
+        torch.manual_seed(seed)
+        a = actor(17, 6, hidden=256)
+        c = critic(17, 6, {"type": "categorical", "v_min": -300.0,
+                           "v_max": 0.0, "n_atoms": 51}, hidden=256)
+        eng.load_from_modules(a, a, c, c)
+        eng.synth_fill(8192, seed=seed + 3)
+        return eng
+
+    e1, e2 = mk(31), mk(31)
+    e1.step(1)
+    E = e2
+    E.step_part(E.PH_CRITIC_GRADS)
+    E.step_part(E.PH_CRITIC_APPLY)
+    E.step_part(E.PH_ACTOR_GRADS)
+    E.step_part(E.PH_ACTOR_APPLY)
+    for s in ("actor", "critic", "actor_target", "critic_target"):
+        np.testing.assert_allclose(e1.store_slab(s).numpy(),
+                                   e2.store_slab(s).numpy(),
+                                   rtol=1e-6, atol=1e-7,
+                                   err_msg=f"wide slab {s} diverged")
